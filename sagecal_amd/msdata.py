@@ -1,0 +1,200 @@
+"""Measurement-set data layer: synthetic MS generation + tile iteration.
+
+Plays the role of /root/reference/src/MS/data.cpp (Data::readAuxData,
+Data::loadData, minibatch loaders) behind a backend-neutral interface so the
+solvers and benchmarks never depend on casacore (SURVEY.md §7 "casacore
+coupling"). The synthetic backend generates a LOFAR-like array, exact uvw
+tracks from earth rotation, and visibilities simulated through the same
+predict op the calibrators use, corrupted by smooth per-station Jones gains
+and Gaussian (or Student's-t) noise.
+
+Layouts (time-major, matching data.cpp:604 row order):
+  row = t * Nbase + b,  b enumerating pairs (p<q) in nested order.
+  u,v,w [rows] in SECONDS (divided by c as fullbatch_mode.cpp:320 does).
+  x (channel-averaged DATA) complex [rows, 2, 2]; xo full channels
+  [Nchan, rows, 2, 2].
+"""
+import math
+import numpy as np
+import torch
+
+from .constants import C_LIGHT
+from . import coords
+
+
+def baseline_pairs(N):
+    """All station pairs p<q in nested-loop order (generate_baselines,
+    baseline_utils.c)."""
+    pairs = [(p, q) for p in range(N) for q in range(p + 1, N)]
+    return np.asarray(pairs, dtype=np.int64)
+
+
+def lofar_like_array(N, seed=7, core_frac=0.5, core_radius=1500.0,
+                     max_radius=40000.0):
+    """Synthetic LOFAR-like station layout: a dense core plus log-spaced
+    arms. Returns ITRF-like ENU positions [N,3] metres (z small)."""
+    rng = np.random.default_rng(seed)
+    ncore = max(2, int(N * core_frac))
+    pos = np.zeros((N, 3))
+    r = core_radius * np.sqrt(rng.uniform(0.01, 1.0, ncore))
+    th = rng.uniform(0, 2 * np.pi, ncore)
+    pos[:ncore, 0] = r * np.cos(th)
+    pos[:ncore, 1] = r * np.sin(th)
+    nrem = N - ncore
+    if nrem > 0:
+        rr = np.exp(rng.uniform(np.log(core_radius), np.log(max_radius), nrem))
+        tt = rng.uniform(0, 2 * np.pi, nrem)
+        pos[ncore:, 0] = rr * np.cos(tt)
+        pos[ncore:, 1] = rr * np.sin(tt)
+    pos[:, 2] = rng.uniform(-5.0, 5.0, N)
+    return pos
+
+
+def enu_uvw(pos_enu, lat, ha, dec):
+    """uvw [m] for given hour angle(s) and declination from ENU station
+    positions. Returns per-station uvw; baseline uvw = uvw[p] - uvw[q].
+
+    ENU -> (XYZ equatorial local) -> uvw rotation; standard interferometry
+    geometry (reference gets uvw from the MS UVW column; we synthesize)."""
+    e, n, u_ = pos_enu[:, 0], pos_enu[:, 1], pos_enu[:, 2]
+    # local equatorial coordinates
+    x = -np.sin(lat) * n + np.cos(lat) * u_
+    y = e
+    z = np.cos(lat) * n + np.sin(lat) * u_
+    ha = np.atleast_1d(ha)[:, None]
+    uu = np.sin(ha) * x + np.cos(ha) * y
+    vv = (-np.sin(dec) * np.cos(ha) * x + np.sin(dec) * np.sin(ha) * y
+          + np.cos(dec) * z)
+    ww = (np.cos(dec) * np.cos(ha) * x - np.cos(dec) * np.sin(ha) * y
+          + np.sin(dec) * z)
+    return uu, vv, ww  # each [T, N]
+
+
+class TileData:
+    """One solution interval of data (one -t tile)."""
+
+    def __init__(self, u, v, w, x, xo, flags, freqs, freq0, fdelta, tdelta,
+                 tilesz, Nbase, dec0=0.0):
+        self.u, self.v, self.w = u, v, w        # [rows] float64 seconds
+        self.x = x                              # [rows,2,2] complex avg
+        self.xo = xo                            # [F,rows,2,2] complex
+        self.flags = flags                      # [rows] bool (True=flagged)
+        self.freqs = freqs                      # [F]
+        self.freq0 = freq0                      # mean freq
+        self.fdelta = fdelta                    # total bandwidth
+        self.tdelta = tdelta
+        self.tilesz = tilesz
+        self.Nbase = Nbase
+        self.dec0 = dec0
+
+
+class SyntheticMS:
+    """Synthetic measurement set with ground-truth gains.
+
+    Mirrors Data:: iteration semantics: construct once (aux metadata), then
+    iterate tiles of `tilesz` timeslots via tiles().
+    """
+
+    def __init__(self, N=62, tilesz=10, Ntime=10, Nchan=4, freq0=150e6,
+                 bandwidth=4e6, tdelta=10.0, ra0=0.0, dec0=np.pi / 4,
+                 lat=0.92, pack=None, nchunks=None, seed=11,
+                 noise_sigma=0.01, gain_amp=0.3, robust_noise=None,
+                 device='cpu', dtype=torch.float64):
+        self.N = N
+        self.Nbase = N * (N - 1) // 2
+        self.tilesz = tilesz
+        self.Ntime = Ntime
+        self.Nchan = Nchan
+        self.freqs = freq0 + bandwidth * (np.arange(Nchan) / max(Nchan - 1, 1)
+                                          - 0.5) if Nchan > 1 else np.array([freq0])
+        self.freq0 = float(np.mean(self.freqs))
+        self.fdelta = bandwidth
+        self.tdelta = tdelta
+        self.ra0, self.dec0, self.lat = ra0, dec0, lat
+        self.pairs = baseline_pairs(N)
+        self.pos = lofar_like_array(N, seed=seed)
+        self.device = device
+        self.dtype = dtype
+        self.rng = np.random.default_rng(seed + 1)
+        self.pack = pack
+        self.noise_sigma = noise_sigma
+        self.gain_amp = gain_amp
+        self.robust_noise = robust_noise
+        self.nchunks = nchunks
+        self.J_true = None   # [M, N, 2, 2] ground truth, generated per tile
+
+    def bb_tensor(self, rows=None, device=None):
+        """Station-pair index tensor [rows,2] for a tile (time-major)."""
+        dev = device or self.device
+        T = self.tilesz
+        bb = np.tile(self.pairs, (T, 1))
+        return torch.tensor(bb, dtype=torch.long, device=dev)
+
+    def uvw_for(self, tile_idx):
+        """u,v,w [rows] in seconds for the tile's timeslots."""
+        T = self.tilesz
+        t0 = tile_idx * T
+        times = (np.arange(t0, t0 + T) + 0.5) * self.tdelta
+        ha = times * 7.2921150e-5 - 0.2  # sidereal track
+        uu, vv, ww = enu_uvw(self.pos, self.lat, ha, self.dec0)
+        p, q = self.pairs[:, 0], self.pairs[:, 1]
+        ub = (uu[:, p] - uu[:, q]).reshape(-1) / C_LIGHT
+        vb = (vv[:, p] - vv[:, q]).reshape(-1) / C_LIGHT
+        wb = (ww[:, p] - ww[:, q]).reshape(-1) / C_LIGHT
+        return ub, vb, wb
+
+    def true_jones(self, M, tile_idx):
+        """Smooth random ground-truth Jones per cluster per station:
+        J = I + gain_amp * (random complex), deterministic per tile."""
+        rng = np.random.default_rng(1000 + tile_idx)
+        J = np.zeros((M, self.N, 2, 2), dtype=np.complex128)
+        for ci in range(M):
+            g = (rng.standard_normal((self.N, 2, 2))
+                 + 1j * rng.standard_normal((self.N, 2, 2)))
+            J[ci] = np.eye(2)[None] + self.gain_amp * g
+        return torch.tensor(J, device=self.device)
+
+    def tiles(self):
+        for ti in range(max(1, self.Ntime // self.tilesz)):
+            yield self.load_tile(ti)
+
+    def load_tile(self, tile_idx):
+        from .ops import reference as R
+        T = self.tilesz
+        rows = self.Nbase * T
+        ub, vb, wb = self.uvw_for(tile_idx)
+        u = torch.tensor(ub, dtype=self.dtype, device=self.device)
+        v = torch.tensor(vb, dtype=self.dtype, device=self.device)
+        w = torch.tensor(wb, dtype=self.dtype, device=self.device)
+        bb = self.bb_tensor()
+        flags = torch.zeros(rows, dtype=torch.bool, device=self.device)
+        cdtype = torch.complex128 if self.dtype == torch.float64 else torch.complex64
+        xo = torch.zeros(self.Nchan, rows, 2, 2, dtype=cdtype,
+                         device=self.device)
+        if self.pack is not None:
+            M = self.pack.M
+            self.J_true = self.true_jones(M, tile_idx)
+            fdelta_ch = self.fdelta / self.Nchan
+            for fi, f in enumerate(self.freqs):
+                coh = R.predict_coh(self.pack, u, v, w, float(f), self.freq0,
+                                    fdelta_ch, self.tdelta, self.dec0)
+                for ci in range(M):
+                    xo[fi] += R.apply_jones(coh[ci],
+                                            self.J_true[ci:ci + 1], bb)
+        # noise
+        if self.noise_sigma > 0:
+            sig = self.noise_sigma
+            nre = self.rng.standard_normal(xo.shape + (2,))
+            if self.robust_noise is not None:
+                # Student's-t noise via scaled inverse-gamma mixture
+                nu = self.robust_noise
+                lam = self.rng.chisquare(nu, size=(self.Nchan, rows)) / nu
+                scl = 1.0 / np.sqrt(lam)
+                nre = nre * scl[:, :, None, None, None]
+            noise = torch.tensor(nre[..., 0] + 1j * nre[..., 1],
+                                 device=self.device) * sig
+            xo = xo + noise.to(xo.dtype)
+        x = xo.mean(dim=0)
+        return TileData(u, v, w, x, xo, flags, self.freqs, self.freq0,
+                        self.fdelta, self.tdelta, T, self.Nbase,
+                        dec0=self.dec0)

@@ -1,0 +1,120 @@
+"""Solution-file I/O, byte-compatible with the reference format.
+
+Format (reference README.md §6 "Solution format"; writer
+fullbatch_mode.cpp:285-289,596-605):
+  - '#' comment lines;
+  - first non-comment line: freq(MHz) bandwidth(MHz) time_interval(min)
+    stations clusters effective_clusters;
+  - per solution interval, 8N rows: counter column then one column per
+    effective cluster chunk, clusters written in REVERSE order
+    (for ci=M-1..0), chunks in forward order (fullbatch_mode.cpp:598).
+  - per station, the 8 values S0..S7 encode the 2x2 Jones COLUMN-major:
+    J = [S0+jS1, S4+jS5; S2+jS3, S6+jS7].
+
+Internally sagecal_amd stores Jones row-major ([J00,J01,J10,J11]
+interleaved re/im); this module permutes on read/write.
+"""
+import numpy as np
+import torch
+
+# internal row-major interleaved index of each reference row s0..s7:
+# ref order [00r,00i,10r,10i,01r,01i,11r,11i] -> internal positions
+_REF_TO_INT = np.array([0, 1, 4, 5, 2, 3, 6, 7])
+
+
+def jones_to_ref_vec(J):
+    """J: [..., N, 2, 2] complex -> [..., N*8] float64 in reference per
+    station order."""
+    v = torch.view_as_real(J).reshape(*J.shape[:-2], 8)
+    return v[..., _REF_TO_INT].reshape(*J.shape[:-3], -1)
+
+
+def ref_vec_to_jones(vec, N):
+    """[.., N*8] reference-order reals -> [.., N, 2, 2] complex."""
+    v = torch.as_tensor(vec).reshape(-1, N, 8)
+    inv = np.argsort(_REF_TO_INT)
+    v = v[..., inv].contiguous()
+    return torch.view_as_complex(v.reshape(-1, N, 2, 2, 2).contiguous())
+
+
+class SolutionWriter:
+    def __init__(self, path, freq0, bandwidth, tile_minutes, N, M, Mt):
+        self.f = open(path, 'w')
+        self.N = N
+        self.f.write("# solution file created by SAGECal\n")
+        self.f.write("# freq(MHz) bandwidth(MHz) time_interval(min) stations"
+                     " clusters effective_clusters\n")
+        self.f.write("%lf %lf %lf %d %d %d\n" % (
+            freq0 * 1e-6, bandwidth * 1e-6, tile_minutes, N, M, Mt))
+
+    def write_tile(self, state):
+        """state: CalState with J [Mt, N, 2, 2]; clusters in reverse order,
+        chunks forward (fullbatch_mode.cpp:598)."""
+        N = self.N
+        cols = []
+        M = state.M
+        for ci in range(M - 1, -1, -1):
+            o = state.chunk_off[ci]
+            for ck in range(state.nchunks[ci]):
+                cols.append(jones_to_ref_vec(
+                    state.J[o + ck].cpu().to(torch.complex128)).numpy())
+        cols = np.stack(cols, axis=1)  # [8N, Mt]
+        for cj in range(8 * N):
+            self.f.write("%d " % cj)
+            self.f.write(''.join(" %e" % val for val in cols[cj]))
+            self.f.write("\n")
+        self.f.flush()
+
+    def close(self):
+        self.f.close()
+
+
+def read_solutions(path):
+    """Parse a solution file (reference read_solutions, readsky.c API
+    Dirac_radio.h:102-110). Returns (header dict, list of tiles; each tile
+    is a [Mt, N, 2, 2] complex tensor with columns mapped back to forward
+    cluster order NOT applied — caller maps via cluster chunk counts)."""
+    header = None
+    rows = []
+    tiles = []
+    with open(path) as f:
+        for line in f:
+            line = line.strip()
+            if not line or line.startswith('#'):
+                continue
+            toks = line.split()
+            if header is None:
+                header = {
+                    'freq_mhz': float(toks[0]), 'bw_mhz': float(toks[1]),
+                    'interval_min': float(toks[2]), 'N': int(toks[3]),
+                    'M': int(toks[4]), 'Mt': int(toks[5])}
+                continue
+            rows.append([float(t) for t in toks[1:]])
+            if len(rows) == 8 * header['N']:
+                arr = np.asarray(rows)  # [8N, Mt]
+                rows = []
+                Mt = arr.shape[1]
+                cols = []
+                for k in range(Mt):
+                    cols.append(ref_vec_to_jones(
+                        torch.tensor(arr[:, k]), header['N']))
+                tiles.append(torch.cat(cols, dim=0))
+    return header, tiles
+
+
+def reorder_read_tile(tile_J, nchunks):
+    """Columns in the file are reverse-cluster-order; map a read tile
+    [Mt, N, 2, 2] back to forward cluster order given per-cluster chunk
+    counts."""
+    M = len(nchunks)
+    # file order: cluster M-1..0, chunks forward
+    out = []
+    pos = 0
+    file_spans = {}
+    for ci in range(M - 1, -1, -1):
+        file_spans[ci] = (pos, pos + nchunks[ci])
+        pos += nchunks[ci]
+    for ci in range(M):
+        s, e = file_spans[ci]
+        out.append(tile_J[s:e])
+    return torch.cat(out, dim=0)

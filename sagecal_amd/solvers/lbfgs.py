@@ -1,0 +1,179 @@
+"""LBFGS: full-batch and minibatch (stochastic, persistent-memory).
+
+Re-implements /root/reference/src/lib/Dirac/lbfgs.c: two-loop recursion
+(mult_hessian, lbfgs.c:33), line searches (Fletcher w/ cubic interpolation
+lbfgs.c:115-441; backtracking for stochastic mode lbfgs.c:443), and the
+cyclic y/s curvature memory persisted across minibatches
+(persistent_data_t, Dirac.h:84-110). Works on a generic cost/grad closure,
+so it is radio-agnostic like the reference library (test/Dirac/demo.c
+minimizes Rosenbrock with it — mirrored in tests/test_lbfgs.py).
+"""
+import torch
+
+
+class LBFGSMemory:
+    """Cyclic s/y curvature store (persistent_data_t analog)."""
+
+    def __init__(self, m, n, dtype=torch.float64, device='cpu'):
+        self.m = m
+        self.S = torch.zeros(m, n, dtype=dtype, device=device)
+        self.Y = torch.zeros(m, n, dtype=dtype, device=device)
+        self.rho = torch.zeros(m, dtype=dtype, device=device)
+        self.count = 0          # total updates pushed
+        self.running_avg = None  # gradient-variance tracking (online mode)
+        self.running_avg_sq = None
+        self.niter = 0
+
+    def reset(self):
+        self.S.zero_(); self.Y.zero_(); self.rho.zero_()
+        self.count = 0
+        self.running_avg = None
+        self.running_avg_sq = None
+
+    def push(self, s, y):
+        sy = torch.dot(s, y)
+        if float(sy) <= 1e-300:
+            return  # curvature condition failed; skip update
+        idx = self.count % self.m
+        self.S[idx] = s
+        self.Y[idx] = y
+        self.rho[idx] = 1.0 / sy
+        self.count += 1
+
+    def two_loop(self, g):
+        """H0-scaled two-loop recursion (mult_hessian lbfgs.c:33)."""
+        k = min(self.count, self.m)
+        if k == 0:
+            return g.clone()
+        order = [(self.count - 1 - i) % self.m for i in range(k)]
+        q = g.clone()
+        alphas = []
+        for idx in order:
+            a = self.rho[idx] * torch.dot(self.S[idx], q)
+            alphas.append(a)
+            q = q - a * self.Y[idx]
+        last = order[0]
+        gamma = torch.dot(self.S[last], self.Y[last]) / \
+            torch.dot(self.Y[last], self.Y[last]).clamp_min(1e-300)
+        q = q * gamma
+        for idx, a in zip(reversed(order), reversed(alphas)):
+            b = self.rho[idx] * torch.dot(self.Y[idx], q)
+            q = q + (a - b) * self.S[idx]
+        return q
+
+
+def _cubic_linesearch(fg, p, f0, g0, d, alpha0=1.0, c1=1e-4, c2=0.9,
+                      max_ls=20):
+    """Strong-Wolfe line search with cubic interpolation (the role of
+    lbfgs.c:115-441's Fletcher search). fg(p) -> (f, g)."""
+    dg0 = torch.dot(g0, d)
+    if float(dg0) >= 0:
+        return None, None, None, 0.0
+    alpha = alpha0
+    alpha_lo, alpha_hi = 0.0, float('inf')
+    f_lo = f0
+    best = None
+    for _ in range(max_ls):
+        f1, g1 = fg(p + alpha * d)
+        if float(f1) > float(f0 + c1 * alpha * dg0) or \
+           (best is not None and float(f1) >= float(f_lo)):
+            alpha_hi = alpha
+        else:
+            dg1 = torch.dot(g1, d)
+            best = (f1, g1, alpha)
+            if abs(float(dg1)) <= -c2 * float(dg0):
+                return f1, g1, p + alpha * d, alpha
+            if float(dg1) >= 0:
+                alpha_hi = alpha
+            else:
+                alpha_lo, f_lo = alpha, f1
+        if alpha_hi < float('inf'):
+            alpha = 0.5 * (alpha_lo + alpha_hi)
+        else:
+            alpha *= 2.0
+    if best is not None:
+        f1, g1, a = best
+        return f1, g1, p + a * d, a
+    return None, None, None, 0.0
+
+
+def _backtracking(fg_cost, p, f0, g0, d, alpha0=1.0, c1=1e-4, beta=0.5,
+                  max_ls=25):
+    """Armijo backtracking (stochastic mode, lbfgs.c:443)."""
+    dg0 = torch.dot(g0, d)
+    alpha = alpha0
+    for _ in range(max_ls):
+        f1 = fg_cost(p + alpha * d)
+        if float(f1) <= float(f0 + c1 * alpha * dg0):
+            return alpha
+        alpha *= beta
+    return 0.0
+
+
+def lbfgs_fit(fg, p0, maxiter=50, m=7, gtol=1e-9, mem=None,
+              stochastic=False, alpha0=1.0):
+    """Minimize via LBFGS. fg(p) -> (cost, grad). Returns (p, mem, info).
+
+    With `mem` given, curvature is warm-started and persisted (minibatch
+    mode: lbfgs_fit_minibatch lbfgs.c:717). stochastic=True switches to
+    Armijo backtracking with a decaying initial step."""
+    p = p0.clone()
+    if mem is None:
+        mem = LBFGSMemory(m, p.numel(), dtype=p.dtype, device=p.device)
+    f, g = fg(p)
+    info = {'f0': float(f), 'niter': 0}
+    for it in range(maxiter):
+        if float(g.abs().max()) < gtol:
+            break
+        d = -mem.two_loop(g)
+        if stochastic:
+            a0 = alpha0 / (1.0 + 0.1 * mem.niter)
+            alpha = _backtracking(lambda q: fg(q)[0], p, f, g, d, alpha0=a0)
+            if alpha == 0.0:
+                break
+            pn = p + alpha * d
+            fn, gn = fg(pn)
+        else:
+            fn, gn, pn, alpha = _cubic_linesearch(fg, p, f, g, d,
+                                                  alpha0=1.0 if mem.count else min(1.0, 1.0 / max(float(g.norm()), 1e-12)))
+            if pn is None:
+                break
+        mem.push(pn - p, gn - g)
+        mem.niter += 1
+        p, f, g = pn, fn, gn
+        info['niter'] = it + 1
+    info['f1'] = float(f)
+    return p, mem, info
+
+
+# ---------------------------------------------------------------------------
+# Calibration wrappers (robust_lbfgs.c lbfgs_fit_wrapper:738 analogs)
+# ---------------------------------------------------------------------------
+
+def _pack_params(J):
+    return torch.view_as_real(J).reshape(-1).clone()
+
+
+def _unpack_params(v, Mt, N, cdtype):
+    return torch.view_as_complex(v.reshape(Mt, N, 2, 2, 2).contiguous())
+
+
+def polish(state, cohs, tile, bb, opts):
+    """Final joint LBFGS over the full 8NMt parameter vector
+    (lmfit.c:1019-1037), Gaussian or Student's-t cost per solver mode."""
+    from ..ops import dispatch as ops
+    Mt, N = state.Mt, state.N
+    T, Nbase = tile.tilesz, tile.Nbase
+    chunk_off = state.chunk_off
+    nchunks = state.nchunks
+    nu = float(state.nu.mean()) if opts.robust else None
+
+    def fg(v):
+        J = _unpack_params(v, Mt, N, state.J.dtype)
+        c, g = ops.lbfgs_cost_grad(tile.x, cohs, J, chunk_off, nchunks, bb,
+                                   T, Nbase, robust_nu=nu)
+        return c, g
+
+    v0 = _pack_params(state.J)
+    v1, _, _ = lbfgs_fit(fg, v0, maxiter=opts.lbfgs_iters, m=7)
+    state.J = _unpack_params(v1, Mt, N, state.J.dtype)

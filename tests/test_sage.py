@@ -1,0 +1,122 @@
+"""End-to-end SAGE calibration tests on synthetic data."""
+import numpy as np
+import pytest
+import torch
+
+from sagecal_amd import sky, msdata
+from sagecal_amd.ops.reference import SourcePack
+from sagecal_amd.solvers import sage
+from sagecal_amd.constants import SM_LM_LBFGS, SM_RLM_RLBFGS
+
+
+def setup_ms(N=8, M=2, tilesz=4, Nchan=1, noise=1e-4, robust_noise=None,
+             nchunks=None, seed=0, bandwidth=180e3):
+    """Narrow sub-band (LOFAR-like 180 kHz) so the channel-averaged data is
+    representable by the centre-frequency model (same physics constraint as
+    the reference's per-band calibration)."""
+    srcs, clist = sky.make_synthetic_sky(M=M, nsrc_per_cluster=3, seed=seed)
+    if nchunks:
+        clist = [(cid, nchunks[i], names)
+                 for i, (cid, _, names) in enumerate(clist)]
+    clusters = sky.build_clusters(srcs, clist, 0.0, np.pi / 4, 150e6)
+    pack = SourcePack(clusters)
+    ms = msdata.SyntheticMS(N=N, tilesz=tilesz, Ntime=tilesz, Nchan=Nchan,
+                            pack=pack, noise_sigma=noise, seed=seed,
+                            robust_noise=robust_noise, bandwidth=bandwidth)
+    return ms, pack
+
+
+@pytest.mark.parametrize("mode", ['sequential', 'batched'])
+def test_sagefit_reduces_residual(mode):
+    ms, pack = setup_ms(M=3, noise=1e-3)
+    tile = ms.load_tile(0)
+    bb = ms.bb_tensor()
+    state = sage.CalState(pack, ms.N)
+    cohs = sage.precalc_coherencies(pack, tile)
+    opts = sage.SageSolveOptions(max_emiter=8, max_iter=15,
+                                 solver_mode=SM_LM_LBFGS, mode=mode)
+    res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
+    assert res1 < 0.05 * res0
+
+
+def test_sagefit_recovers_model():
+    """Noiseless: the solved model reproduces the data to high accuracy."""
+    ms, pack = setup_ms(noise=0.0)
+    tile = ms.load_tile(0)
+    bb = ms.bb_tensor()
+    state = sage.CalState(pack, ms.N)
+    cohs = sage.precalc_coherencies(pack, tile)
+    opts = sage.SageSolveOptions(max_emiter=3, max_iter=15,
+                                 solver_mode=SM_LM_LBFGS, mode='batched',
+                                 joint_iters=10)
+    res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
+    assert res1 < 1e-5 * res0
+
+
+def test_sagefit_hybrid_chunks():
+    ms, pack = setup_ms(nchunks=[1, 2], tilesz=4, noise=1e-4)
+    tile = ms.load_tile(0)
+    bb = ms.bb_tensor()
+    state = sage.CalState(pack, ms.N)
+    assert state.Mt == 3
+    cohs = sage.precalc_coherencies(pack, tile)
+    opts = sage.SageSolveOptions(max_emiter=3, max_iter=15,
+                                 solver_mode=SM_LM_LBFGS, mode='batched',
+                                 joint_iters=6)
+    res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
+    assert res1 < 0.05 * res0
+
+
+def test_sagefit_robust_with_outliers():
+    """Student's-t noise: robust mode beats plain LM on outlier data."""
+    ms, pack = setup_ms(noise=5e-3, robust_noise=3.0, seed=2)
+    tile = ms.load_tile(0)
+    bb = ms.bb_tensor()
+    cohs = sage.precalc_coherencies(pack, tile)
+
+    state_r = sage.CalState(pack, ms.N)
+    opts_r = sage.SageSolveOptions(max_emiter=3, max_iter=15,
+                                   solver_mode=SM_RLM_RLBFGS, mode='batched',
+                                   robust_outer=2, joint_iters=6)
+    res0_r, res1_r = sage.sagefit(state_r, cohs, tile, bb, opts_r)
+    assert res1_r < res0_r
+    # recovered gains closer to truth than identity: compare model vis
+    from sagecal_amd.ops import reference as R
+    Vtrue = torch.zeros_like(tile.x)
+    for ci in range(pack.M):
+        Vtrue += R.apply_jones(cohs[ci], ms.J_true[ci:ci + 1], bb)
+    Vest = sage.total_model(state_r, cohs, bb, tile.tilesz, tile.Nbase)
+    err_est = float((Vest - Vtrue).abs().mean())
+    err_id = float((sage.total_model(sage.CalState(pack, ms.N), cohs, bb,
+                                     tile.tilesz, tile.Nbase)
+                    - Vtrue).abs().mean())
+    assert err_est < 0.2 * err_id
+
+
+def test_lbfgs_polish_improves():
+    ms, pack = setup_ms(noise=0.0, seed=4)
+    tile = ms.load_tile(0)
+    bb = ms.bb_tensor()
+    state = sage.CalState(pack, ms.N)
+    cohs = sage.precalc_coherencies(pack, tile)
+    opts = sage.SageSolveOptions(max_emiter=1, max_iter=4,
+                                 solver_mode=SM_LM_LBFGS, mode='batched',
+                                 lbfgs_iters=20)
+    res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
+    assert res1 < res0
+
+
+def test_residuals_multifreq():
+    ms, pack = setup_ms(noise=0.0, Nchan=2, bandwidth=4e3)
+    tile = ms.load_tile(0)
+    bb = ms.bb_tensor()
+    state = sage.CalState(pack, ms.N)
+    cohs = sage.precalc_coherencies(pack, tile)
+    opts = sage.SageSolveOptions(max_emiter=3, max_iter=15,
+                                 solver_mode=SM_LM_LBFGS, mode='batched',
+                                 joint_iters=8)
+    sage.sagefit(state, cohs, tile, bb, opts)
+    xres = sage.calculate_residuals_multifreq(state, pack, tile, bb)
+    assert xres.shape == tile.xo.shape
+    # residual much smaller than data
+    assert float(xres.abs().mean()) < 0.05 * float(tile.xo.abs().mean())

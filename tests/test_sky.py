@@ -1,0 +1,79 @@
+import numpy as np
+import pytest
+
+from sagecal_amd import sky, coords
+
+
+SKY = """\
+# name h m s d m s I Q U V si RM eX eY eP f0
+P1C1 0 12 42.996 85 43 21.514 0.030498 0 0 0 -5.713060 0 0 0 0 115039062.0
+G0  5 34 31.75 22 0 52.86 100 0 0 0 0.0 0 0.0012  0.0008 -2.329615801 130.0e6
+"""
+
+CLUSTER = """\
+# comment
+1 1 P1C1
+-2 2 G0
+"""
+
+
+def _write(tmp_path, name, text):
+    p = tmp_path / name
+    p.write_text(text)
+    return str(p)
+
+
+def test_parse_sky(tmp_path):
+    srcs = sky.read_sky_model(_write(tmp_path, 's.txt', SKY))
+    assert len(srcs) == 2
+    s = srcs['P1C1']
+    assert s.sI == pytest.approx(0.030498)
+    assert s.spec_idx == pytest.approx(-5.713060)
+    assert s.f0 == pytest.approx(115039062.0)
+    assert s.stype == 0
+    g = srcs['G0']
+    assert g.stype == 1
+    assert g.eX == pytest.approx(0.0012)
+
+
+def test_parse_cluster(tmp_path):
+    cl = sky.read_cluster_file(_write(tmp_path, 'c.txt', CLUSTER))
+    assert cl == [(1, 1, ['P1C1']), (-2, 2, ['G0'])]
+
+
+def test_build_clusters(tmp_path):
+    srcs = sky.read_sky_model(_write(tmp_path, 's.txt', SKY))
+    cl = sky.read_cluster_file(_write(tmp_path, 'c.txt', CLUSTER))
+    ra0 = coords.hms_to_rad(0, 12, 42.996)
+    dec0 = coords.dms_to_rad(85, 43, 21.514)
+    clusters = sky.build_clusters(srcs, cl, ra0, dec0, 150e6)
+    assert len(clusters) == 2
+    c0 = clusters[0]
+    # P1C1 is at the phase centre: l=m=0, n=1
+    assert abs(c0.ll[0]) < 1e-12 and abs(c0.mm[0]) < 1e-12
+    assert abs(c0.nn1[0]) < 1e-12
+    # flux scaled to 150 MHz with spectral index
+    expect = np.exp(np.log(0.030498) - 5.713060 * np.log(150e6 / 115039062.0))
+    assert c0.sI[0] == pytest.approx(expect)
+    # gaussian cluster: fwhm->sigma conversion
+    c1 = clusters[1]
+    assert c1.nchunk == 2
+    assert c1.eX[0] == pytest.approx(0.0012 / (2 * np.sqrt(2 * np.log(2))))
+
+
+def test_ignore_and_arho(tmp_path):
+    srcs = sky.read_sky_model(_write(tmp_path, 's.txt', SKY))
+    cl = sky.read_cluster_file(_write(tmp_path, 'c.txt', CLUSTER))
+    clusters = sky.build_clusters(srcs, cl, 0.0, 1.2, 150e6, ignore_ids=[-2])
+    assert len(clusters) == 1
+    arho_txt = "# c h s sp\n1 1 5.0 3.0\n"
+    arho, arho_s = sky.read_arho_file(_write(tmp_path, 'g.txt', arho_txt),
+                                      clusters)
+    assert arho[0] == 5.0 and arho_s[0] == 3.0
+
+
+def test_synthetic_sky():
+    srcs, clist = sky.make_synthetic_sky(M=3, nsrc_per_cluster=2)
+    clusters = sky.build_clusters(srcs, clist, 0.0, np.pi / 4, 150e6)
+    assert len(clusters) == 3
+    assert clusters[0].nsrc == 2
