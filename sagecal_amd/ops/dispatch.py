@@ -56,19 +56,39 @@ def predict_coh(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0, **kw):
                          **kw)
 
 
-def jtj_jtr(x, coh, J, bb, N, weights=None, chunk_rows=None, nchunk=1):
+def jtj_jtr(x, coh, J, bb, N, weights=None, chunk_rows=None, nchunk=1,
+            layout=None):
     if _use_hip(x):
         from . import hip_host
         return hip_host.jtj_jtr(x, coh, J, bb, N, weights, chunk_rows,
-                                nchunk)
+                                nchunk, layout)
     return R.jtj_jtr(x, coh, J, bb, N, weights, chunk_rows, nchunk)
 
 
-def apply_jones(coh, J, bb, chunk_rows=None):
+def apply_jones(coh, J, bb, chunk_rows=None, layout=None):
     if _use_hip(coh):
         from . import hip_host
-        return hip_host.apply_jones(coh, J, bb, chunk_rows)
+        return hip_host.apply_jones(coh, J, bb, chunk_rows, layout)
     return R.apply_jones(coh, J, bb, chunk_rows)
+
+
+def model_cost_per_chunk(x, coh, J, bb, N, weights=None, chunk_rows=None,
+                         nchunk=1, layout=None):
+    """Per-chunk weighted cost [nchunk] (kernel_fcost analog)."""
+    if _use_hip(x) and layout is not None:
+        from . import hip_host
+        return hip_host.model_cost_per_chunk(x, coh, J, bb, N, weights,
+                                             chunk_rows, nchunk, layout)
+    V = R.apply_jones(coh, J, bb, chunk_rows)
+    r = x - V
+    e2 = (r.abs() ** 2).sum(dim=(-1, -2))
+    if weights is not None:
+        e2 = e2 * weights
+    if chunk_rows is None:
+        return e2.sum().unsqueeze(0)
+    out = torch.zeros(nchunk, dtype=e2.dtype, device=e2.device)
+    out.index_add_(0, chunk_rows, e2)
+    return out
 
 
 def model_and_cost(x, coh, J, bb, weights=None, chunk_rows=None):

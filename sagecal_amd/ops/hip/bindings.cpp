@@ -1,0 +1,124 @@
+// PyTorch bindings for the sagecal_amd gfx950 kernel library.
+// Host-compiled: all kernel launches live in launchers.hip (hipcc).
+#include <torch/extension.h>
+#include <hip/hip_runtime.h>
+#include <c10/hip/HIPStream.h>
+
+extern "C" {
+hipError_t launch_predict_coh(const double*, const double*, const double*,
+    const double*, const double*, const double*, const float*, const float*,
+    const float*, const float*, const float*, const float*, const float*,
+    const float*, const float*, const float*, const float*, const float*,
+    const int*, const int*, int, int, double, double, double, float2*,
+    hipStream_t);
+hipError_t launch_jtj_accum(const float2*, const float2*, const float2*,
+    const int*, const int*, const float*, int, int, int, int, float2*,
+    float2*, float2*, float*, int, hipStream_t);
+hipError_t launch_jtj_expand(const float2*, const float2*, const float2*,
+    const int*, int, int, int, float*, float*, hipStream_t);
+hipError_t launch_model_cost(const float2*, const float2*, const float2*,
+    const int*, const int*, const float*, int, int, int, int, float*,
+    hipStream_t);
+hipError_t launch_apply_jones(const float2*, const float2*, const float2*,
+    const int*, const int*, int, int, int, int, int, int, float2*,
+    hipStream_t);
+}
+
+#define CHECK_HIP(x) do { hipError_t e = (x); TORCH_CHECK(e == hipSuccess, \
+    "HIP error: ", hipGetErrorString(e)); } while (0)
+
+static hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+static float2* cptr(torch::Tensor& t) {
+  return reinterpret_cast<float2*>(t.data_ptr());
+}
+static const float2* ccptr(const torch::Tensor& t) {
+  return reinterpret_cast<const float2*>(t.data_ptr());
+}
+
+torch::Tensor predict_coh(
+    torch::Tensor u, torch::Tensor v, torch::Tensor w,
+    torch::Tensor ll, torch::Tensor mm, torch::Tensor nn1,
+    torch::Tensor sI, torch::Tensor sQ, torch::Tensor sU, torch::Tensor sV,
+    torch::Tensor eX, torch::Tensor eY, torch::Tensor eP,
+    torch::Tensor cxi, torch::Tensor sxi, torch::Tensor cphi,
+    torch::Tensor sphi, torch::Tensor r1, torch::Tensor stype,
+    torch::Tensor cluster_off, double freq, double fdelta2, double tdelta) {
+  const int R = u.size(0);
+  const int M = cluster_off.size(0) - 1;
+  auto out = torch::empty({M, R, 4},
+      torch::dtype(torch::kComplexFloat).device(u.device()));
+  CHECK_HIP(launch_predict_coh(
+      u.data_ptr<double>(), v.data_ptr<double>(), w.data_ptr<double>(),
+      ll.data_ptr<double>(), mm.data_ptr<double>(), nn1.data_ptr<double>(),
+      sI.data_ptr<float>(), sQ.data_ptr<float>(), sU.data_ptr<float>(),
+      sV.data_ptr<float>(), eX.data_ptr<float>(), eY.data_ptr<float>(),
+      eP.data_ptr<float>(), cxi.data_ptr<float>(), sxi.data_ptr<float>(),
+      cphi.data_ptr<float>(), sphi.data_ptr<float>(), r1.data_ptr<float>(),
+      stype.data_ptr<int>(), cluster_off.data_ptr<int>(), M, R, freq,
+      fdelta2, tdelta, cptr(out), cur_stream()));
+  return out;
+}
+
+std::vector<torch::Tensor> jtj_jtr(
+    torch::Tensor x, torch::Tensor coh, torch::Tensor J,
+    torch::Tensor pairs, torch::Tensor chunk_tab, torch::Tensor pidx,
+    c10::optional<torch::Tensor> wts, int64_t Nbase, int64_t T, int64_t N,
+    int64_t nseg, int64_t Mt) {
+  auto fopts = torch::dtype(torch::kFloat).device(x.device());
+  auto copts = torch::dtype(torch::kComplexFloat).device(x.device());
+  const int64_t npair = Nbase;
+  auto D = torch::zeros({Mt * N, 4}, copts);
+  auto g = torch::zeros({Mt * N, 4}, copts);
+  auto Cx = torch::zeros({Mt * npair, 16}, copts);
+  auto cost = torch::zeros({Mt}, fopts);
+  const float* wp = wts.has_value() ? wts->data_ptr<float>() : nullptr;
+  CHECK_HIP(launch_jtj_accum(ccptr(x), ccptr(coh), ccptr(J),
+      pairs.data_ptr<int>(), chunk_tab.data_ptr<int>(), wp,
+      (int)Nbase, (int)T, (int)N, (int)nseg, cptr(D), cptr(g), cptr(Cx),
+      cost.data_ptr<float>(), (int)npair, cur_stream()));
+  auto JtJ = torch::empty({Mt, 8 * N, 8 * N}, fopts);
+  auto Jtr = torch::empty({Mt, 8 * N}, fopts);
+  CHECK_HIP(launch_jtj_expand(ccptr(D), ccptr(g), ccptr(Cx),
+      pidx.data_ptr<int>(), (int)N, (int)npair, (int)Mt,
+      JtJ.data_ptr<float>(), Jtr.data_ptr<float>(), cur_stream()));
+  return {JtJ, Jtr, cost};
+}
+
+torch::Tensor model_cost(
+    torch::Tensor x, torch::Tensor coh, torch::Tensor J,
+    torch::Tensor pairs, torch::Tensor chunk_tab,
+    c10::optional<torch::Tensor> wts, int64_t Nbase, int64_t T, int64_t N,
+    int64_t nseg, int64_t Mt) {
+  auto cost = torch::zeros({Mt},
+      torch::dtype(torch::kFloat).device(x.device()));
+  const float* wp = wts.has_value() ? wts->data_ptr<float>() : nullptr;
+  CHECK_HIP(launch_model_cost(ccptr(x), ccptr(coh), ccptr(J),
+      pairs.data_ptr<int>(), chunk_tab.data_ptr<int>(), wp,
+      (int)Nbase, (int)T, (int)N, (int)nseg, cost.data_ptr<float>(),
+      cur_stream()));
+  return cost;
+}
+
+torch::Tensor apply_jones(
+    c10::optional<torch::Tensor> x, torch::Tensor cohs, torch::Tensor J,
+    torch::Tensor pairs, torch::Tensor chunk_tab, int64_t Nbase, int64_t T,
+    int64_t N, int64_t nseg, int64_t M, int64_t sub) {
+  const int64_t B = nseg * T * Nbase;
+  auto out = torch::empty({B, 4},
+      torch::dtype(torch::kComplexFloat).device(cohs.device()));
+  const float2* xp = x.has_value() ? ccptr(*x) : nullptr;
+  CHECK_HIP(launch_apply_jones(xp, ccptr(cohs), ccptr(J),
+      pairs.data_ptr<int>(), chunk_tab.data_ptr<int>(), (int)Nbase, (int)T,
+      (int)N, (int)nseg, (int)M, (int)sub, cptr(out), cur_stream()));
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("predict_coh", &predict_coh, "coherency predict (gfx950)");
+  m.def("jtj_jtr", &jtj_jtr, "fused JtJ/Jtr assembly (gfx950)");
+  m.def("model_cost", &model_cost, "per-chunk model cost (gfx950)");
+  m.def("apply_jones", &apply_jones, "model apply / residual (gfx950)");
+}

@@ -1,0 +1,196 @@
+"""Host-side adapters between the dispatch API and the gfx950 kernels.
+
+Handles layout preparation: per-frequency flux scaling (done with torch on
+device, keeping the predict kernel free of spectral-index math), baking the
+projection identity into unused rotation terms, chunk tables, station-pair
+index tables, and complex64 views.
+"""
+import math
+import torch
+
+from . import dispatch
+
+_pidx_cache = {}
+
+
+def _ext():
+    e = dispatch._load_ext()
+    if e is None:
+        raise RuntimeError(
+            f"dirac_hip extension unavailable: {dispatch._ext_err}")
+    return e
+
+
+class GPUPack:
+    """Device-resident SourcePack in kernel layout (float32 props, float64
+    direction cosines, projection identity baked for use_proj=False)."""
+
+    def __init__(self, pack, device):
+        f32 = lambda t: t.to(device=device, dtype=torch.float32).contiguous()
+        f64 = lambda t: t.to(device=device, dtype=torch.float64).contiguous()
+        self.M = pack.M
+        self.ll, self.mm, self.nn1 = f64(pack.ll), f64(pack.mm), f64(pack.nn1)
+        for k in ('sI', 'sQ', 'sU', 'sV', 'sI0', 'sQ0', 'sU0', 'sV0',
+                  'spec_idx', 'spec_idx1', 'spec_idx2', 'eX', 'eY', 'eP'):
+            setattr(self, k, f32(getattr(pack, k)))
+        self.f0 = pack.f0.to(device=device, dtype=torch.float64)
+        up = pack.use_proj.to(device=device).bool()
+        one = torch.ones_like(f32(pack.cxi))
+        zero = torch.zeros_like(one)
+        self.cxi = torch.where(up, f32(pack.cxi), one).contiguous()
+        self.sxi = torch.where(up, f32(pack.sxi), zero).contiguous()
+        self.cphi = torch.where(up, f32(pack.cphi), one).contiguous()
+        self.sphi = torch.where(up, f32(pack.sphi), zero).contiguous()
+        self.stype = pack.stype.to(device=device,
+                                   dtype=torch.int32).contiguous()
+        # zero shape angles for point sources keep the envelope branch out
+        self.cluster_off = pack.cluster_off.to(device=device,
+                                               dtype=torch.int32).contiguous()
+        self.freq0 = None
+        self._flux_cache = {}
+
+    def fluxes_at(self, freq, freq0):
+        """Per-source fluxes at channel `freq`; pack fluxes are given at
+        freq0 (sI..) and catalogue f0 (sI0..)."""
+        key = round(float(freq), 3)
+        if key in self._flux_cache:
+            return self._flux_cache[key]
+        if abs(freq - freq0) < 1.0:
+            out = (self.sI, self.sQ, self.sU, self.sV)
+        else:
+            lf = torch.log(torch.tensor(float(freq), dtype=torch.float64,
+                                        device=self.ll.device) / self.f0)
+            lf = lf.to(torch.float32)
+            flog = (self.spec_idx * lf + self.spec_idx1 * lf ** 2
+                    + self.spec_idx2 * lf ** 3)
+
+            def scale(s0):
+                mag = torch.exp(torch.log(s0.abs().clamp_min(1e-30)) + flog)
+                return torch.where(s0 == 0, torch.zeros_like(s0),
+                                   torch.sign(s0) * mag)
+            out = (scale(self.sI0), scale(self.sQ0), scale(self.sU0),
+                   scale(self.sV0))
+        self._flux_cache[key] = out
+        return out
+
+    def r1_for(self, dec0):
+        """Time-smear source-distance term sqrt(ll^2+(sin(dec0) mm)^2)
+        (predict.c:98-100), per source."""
+        ds = math.sin(dec0)
+        return torch.sqrt(self.ll ** 2 + (ds * self.mm) ** 2).to(
+            torch.float32).contiguous()
+
+
+_gpu_pack_cache = {}
+
+
+def gpu_pack(pack, device):
+    key = id(pack)
+    if key not in _gpu_pack_cache:
+        _gpu_pack_cache[key] = GPUPack(pack, device)
+    return _gpu_pack_cache[key]
+
+
+def predict_coh(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0, **kw):
+    gp = pack if isinstance(pack, GPUPack) else gpu_pack(pack, u.device)
+    sI, sQ, sU, sV = gp.fluxes_at(freq, freq0)
+    r1 = gp.r1_for(dec0)
+    u64 = u.to(torch.float64).contiguous()
+    v64 = v.to(torch.float64).contiguous()
+    w64 = w.to(torch.float64).contiguous()
+    out = _ext().predict_coh(
+        u64, v64, w64, gp.ll, gp.mm, gp.nn1, sI, sQ, sU, sV,
+        gp.eX, gp.eY, gp.eP, gp.cxi, gp.sxi, gp.cphi, gp.sphi, r1,
+        gp.stype, gp.cluster_off, float(freq), float(fdelta) * 0.5,
+        float(tdelta))
+    return out.view(gp.M, -1, 2, 2)
+
+
+class BaselineLayout:
+    """Row-structure descriptor for the solver kernels: rows are
+    seg*(T*Nbase) + t*Nbase + b with a fixed pair table. Built once per
+    tile; pair-index table cached per (N, Nbase)."""
+
+    def __init__(self, bb, Nbase, T, nseg, N, device):
+        self.Nbase, self.T, self.nseg, self.N = Nbase, T, nseg, N
+        self.pairs = bb[:Nbase].to(device=device,
+                                   dtype=torch.int32).contiguous()
+        key = (N, Nbase, str(device))
+        if key not in _pidx_cache:
+            pidx = torch.full((N, N), -1, dtype=torch.int32)
+            pcpu = self.pairs.cpu()
+            for i in range(Nbase):
+                p, q = int(pcpu[i, 0]), int(pcpu[i, 1])
+                if p >= 0 and q >= 0:
+                    pidx[min(p, q), max(p, q)] = i
+            _pidx_cache[key] = pidx.to(device).contiguous()
+        self.pidx = _pidx_cache[key]
+
+    def chunk_tab(self, chunk_rows):
+        """Per-(seg,t) global chunk id from per-row chunk indices."""
+        if chunk_rows is None:
+            return torch.zeros(self.nseg * self.T, dtype=torch.int32,
+                               device=self.pairs.device)
+        return chunk_rows[::self.Nbase].to(torch.int32).contiguous()
+
+
+def _c64(t):
+    return t.to(torch.complex64).reshape(t.shape[0], 4).contiguous()
+
+
+def jtj_jtr(x, coh, J, bb, N, weights=None, chunk_rows=None, nchunk=1,
+            layout=None):
+    if layout is None:
+        raise RuntimeError("GPU jtj_jtr requires a BaselineLayout "
+                           "(structured rows); got none")
+    ct = layout.chunk_tab(chunk_rows)
+    w32 = weights.to(torch.float32).contiguous() if weights is not None \
+        else None
+    Jc = J.to(torch.complex64).reshape(-1, 4).contiguous()
+    JtJ, Jtr, cost = _ext().jtj_jtr(
+        _c64(x), _c64(coh), Jc, layout.pairs, ct, layout.pidx, w32,
+        layout.Nbase, layout.T, N, layout.nseg, nchunk)
+    return JtJ, Jtr, cost.sum()
+
+
+def model_cost_per_chunk(x, coh, J, bb, N, weights=None, chunk_rows=None,
+                         nchunk=1, layout=None):
+    ct = layout.chunk_tab(chunk_rows)
+    w32 = weights.to(torch.float32).contiguous() if weights is not None \
+        else None
+    Jc = J.to(torch.complex64).reshape(-1, 4).contiguous()
+    return _ext().model_cost(_c64(x), _c64(coh), Jc, layout.pairs, ct, w32,
+                             layout.Nbase, layout.T, N, layout.nseg, nchunk)
+
+
+def apply_jones(coh, J, bb, chunk_rows=None, layout=None):
+    if layout is None:
+        # fall back to torch complex math on GPU (cold path)
+        from . import reference as R
+        return R.apply_jones(coh, J, bb, chunk_rows)
+    ct = layout.chunk_tab(chunk_rows)
+    Jc = J.to(torch.complex64).reshape(-1, 4).contiguous()
+    cohs = _c64(coh).unsqueeze(0)
+    out = _ext().apply_jones(None, cohs.reshape(-1, 4), Jc, layout.pairs,
+                             ct, layout.Nbase, layout.T, J.shape[1],
+                             layout.nseg, 1, 0)
+    return out.view(-1, 2, 2)
+
+
+def residual_total(x, cohs, J, chunk_tabs, layout):
+    """x - sum_ci J C J^H; chunk_tabs: [M, nseg*T] int32."""
+    M = cohs.shape[0]
+    Jc = J.to(torch.complex64).reshape(-1, 4).contiguous()
+    out = _ext().apply_jones(
+        _c64(x), cohs.to(torch.complex64).reshape(-1, 4).contiguous(), Jc,
+        layout.pairs, chunk_tabs.reshape(-1).contiguous(), layout.Nbase,
+        layout.T, J.shape[1], layout.nseg, M, 1)
+    return out.view(-1, 2, 2)
+
+
+def lbfgs_cost_grad(x, cohs, J_packed, chunk_off, nchunks, bb, T, Nbase,
+                    robust_nu=None, weights=None):
+    # torch-on-GPU path (cold op; HIP kernel in a later wave)
+    from . import reference as R
+    return R.lbfgs_cost_grad(x, cohs, J_packed, chunk_off, nchunks, bb, T,
+                             Nbase, robust_nu, weights)

@@ -23,7 +23,7 @@ class LMProblem:
     (or for SAGE-batched mode: all clusters x chunks concatenated)."""
 
     def __init__(self, x, coh, bb, N, nchunk=1, chunk_rows=None,
-                 weights=None):
+                 weights=None, layout=None):
         self.x = x              # [B,2,2] complex (data with own model added)
         self.coh = coh          # [B,2,2] complex cluster coherency
         self.bb = bb            # [B,2] long
@@ -31,6 +31,7 @@ class LMProblem:
         self.nchunk = nchunk
         self.chunk_rows = chunk_rows  # [B] long or None
         self.weights = weights  # [B] float or None
+        self.layout = layout    # BaselineLayout (GPU kernels) or None
 
 
 def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
@@ -54,7 +55,7 @@ def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
     active = torch.ones(nchunk, dtype=torch.bool, device=dev)
 
     JtJ, Jtr, cost0 = ops.jtj_jtr(x, coh, J, bb, N, prob.weights,
-                                  prob.chunk_rows, nchunk)
+                                  prob.chunk_rows, nchunk, prob.layout)
     cost = _per_chunk_cost(x, coh, J, bb, prob)
     init_cost = cost.clone()
     eye = torch.eye(8 * N, dtype=rdt, device=dev).unsqueeze(0)
@@ -96,7 +97,7 @@ def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
             break
         # recompute JtJ/Jtr at (possibly) new J
         JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, prob.weights,
-                                  prob.chunk_rows, nchunk)
+                                  prob.chunk_rows, nchunk, prob.layout)
     info = {'init_cost': init_cost, 'final_cost': cost, 'niter': niter}
     return J, info
 
@@ -130,16 +131,9 @@ def _jones_norm(J, nchunk):
 
 def _per_chunk_cost(x, coh, J, bb, prob):
     """Weighted residual cost per chunk [nchunk]."""
-    V = ops.apply_jones(coh, J, bb, prob.chunk_rows)
-    r = x - V
-    e2 = (r.abs() ** 2).sum(dim=(-1, -2))
-    if prob.weights is not None:
-        e2 = e2 * prob.weights
-    if prob.chunk_rows is None:
-        return e2.sum().unsqueeze(0)
-    out = torch.zeros(prob.nchunk, dtype=e2.dtype, device=e2.device)
-    out.index_add_(0, prob.chunk_rows, e2)
-    return out
+    return ops.model_cost_per_chunk(x, coh, J, bb, prob.N, prob.weights,
+                                    prob.chunk_rows, prob.nchunk,
+                                    prob.layout)
 
 
 def os_lm_solve(prob, J0, maxiter=30, nsubsets=4, seed=0, **kw):
@@ -160,6 +154,7 @@ def os_lm_solve(prob, J0, maxiter=30, nsubsets=4, seed=0, **kw):
             prob.x[sel], prob.coh[sel], prob.bb[sel], prob.N, prob.nchunk,
             prob.chunk_rows[sel] if prob.chunk_rows is not None else None,
             prob.weights[sel] if prob.weights is not None else None)
+        # random-row subsets break the (seg,t,pair) structure: CPU path
         J, _ = lm_solve(sub, J, maxiter=iters_per, **kw)
     J, info = lm_solve(prob, J, maxiter=iters_per, **kw)
     return J, info

@@ -100,9 +100,32 @@ def precalc_coherencies(pack, tile, device=None):
                            tile.fdelta, tile.tdelta, tile.dec0)
 
 
-def _model_cluster(state, ci, coh_ci, bb, T, Nbase, B):
+def _layout_for(state, bb, T, Nbase, nseg, device):
+    """BaselineLayout for GPU kernels (None on CPU). Cached per nseg."""
+    if not (hasattr(bb, 'is_cuda') and bb.is_cuda):
+        return None
+    import os
+    if os.environ.get('SAGECAL_FORCE_REFERENCE') == '1':
+        return None
+    from ..ops.hip_host import BaselineLayout
+    cache = getattr(state, '_lay_cache', None)
+    if cache is None:
+        cache = state._lay_cache = {}
+    key = (nseg, T, Nbase)
+    if key not in cache:
+        cache[key] = BaselineLayout(bb, Nbase, T, nseg, state.N, device)
+    return cache[key]
+
+
+def _model_cluster(state, ci, coh_ci, bb, T, Nbase, B, lay=None):
     rows = R.chunk_rows_for(ci, state.nchunks, T, Nbase, B, coh_ci.device)
-    return ops.apply_jones(coh_ci, state.cluster_J(ci), bb, rows), rows
+    if lay is None:
+        lay = _layout_for(state, bb, T, Nbase, 1, coh_ci.device)
+    # per-cluster chunk rows are LOCAL here; offset to the cluster's global
+    # chunk range for the packed-J kernels
+    rows_g = rows
+    Jc = state.cluster_J(ci)
+    return ops.apply_jones(coh_ci, Jc, bb, rows_g, lay), rows
 
 
 def total_model(state, cohs, bb, T, Nbase):
@@ -174,7 +197,9 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts):
     bbcat = torch.cat(bbs)
     rcat = torch.cat(rows_all)
     J0 = torch.cat([state.cluster_J(ci) for ci in group])
-    prob = lm_mod.LMProblem(xcat, ccat, bbcat, state.N, nch_tot, rcat)
+    lay = _layout_for(state, bb, T, Nbase, len(group), xcat.device)
+    prob = lm_mod.LMProblem(xcat, ccat, bbcat, state.N, nch_tot, rcat,
+                            layout=lay)
     if opts.robust:
         nus = float(torch.stack([state.nu[ci] for ci in group]).mean())
         Jn, nu_new = robust_lm(prob, J0, nus, opts)
