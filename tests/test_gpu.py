@@ -132,7 +132,7 @@ def test_apply_jones_and_cost(problem):
     V_g = apply_jones(cohs[0].to(device=dev, dtype=torch.complex64),
                       J.to(device=dev, dtype=torch.complex64),
                       bb.to(dev), None, lay)
-    err = (V_g.cpu().double() - V_r).abs().max() / float(V_r.abs().max())
+    err = (V_g.cpu().to(torch.complex128) - V_r).abs().max() / float(V_r.abs().max())
     assert float(err) < 5e-6
     cost_r = float(((tile.x - V_r).abs() ** 2).sum())
     cost_g = model_cost_per_chunk(
