@@ -39,8 +39,11 @@ def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
     """Batched LM: J0 [nchunk, N, 2, 2] complex initial Jones.
 
     Returns (J, info dict). All chunks iterate in lockstep with per-chunk
-    damping/accept state (converged chunks keep their params; their JtJ is
-    still computed — acceptable since chunks share the fused kernels).
+    damping/accept state. The whole iteration is host-sync-free (tensor
+    masks for accept/reject, cholesky_ex instead of throwing cholesky);
+    convergence is only polled every 8 iterations so the GPU pipeline
+    stays full (the reference's per-iteration termination checks are a
+    CPU-ism).
 
     Math follows clmfit.c:240-420: mu = tau*max(diag(JtJ)) init; accept step
     if rho>0 with mu *= max(1/3, 1-(2rho-1)^3), nu=2; else mu *= nu, nu *= 2.
@@ -50,72 +53,59 @@ def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
     dev = x.device
     rdt = x.real.dtype
     J = J0.clone()
-    mu = None
     nu = torch.full((nchunk,), 2.0, dtype=rdt, device=dev)
     active = torch.ones(nchunk, dtype=torch.bool, device=dev)
 
-    JtJ, Jtr, cost0 = ops.jtj_jtr(x, coh, J, bb, N, prob.weights,
-                                  prob.chunk_rows, nchunk, prob.layout)
+    JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, prob.weights,
+                              prob.chunk_rows, nchunk, prob.layout)
     cost = _per_chunk_cost(x, coh, J, bb, prob)
     init_cost = cost.clone()
     eye = torch.eye(8 * N, dtype=rdt, device=dev).unsqueeze(0)
+    diag_max = JtJ.diagonal(dim1=-2, dim2=-1).max(dim=-1).values
+    mu = tau * diag_max
     niter = 0
     for it in range(maxiter):
         niter = it + 1
-        if mu is None:
-            diag_max = JtJ.diagonal(dim1=-2, dim2=-1).max(dim=-1).values
-            mu = tau * diag_max
-        gnorm = Jtr.abs().max(dim=-1).values
-        active = active & (gnorm > eps1)
-        if not bool(active.any()):
-            break
         A = JtJ + mu[:, None, None] * eye
         dp = _chol_solve(A, Jtr)
-        # candidate params
+        dp = torch.nan_to_num(dp, nan=0.0, posinf=0.0, neginf=0.0)
         dpc = _vec_to_jones(dp, nchunk, N)
         Jnew = J + dpc
         cost_new = _per_chunk_cost(x, coh, Jnew, bb, prob)
-        # rho denominator: dp^T (mu*dp + Jtr)
         denom = (dp * (mu[:, None] * dp + Jtr)).sum(dim=-1).clamp_min(1e-30)
         rho = (cost - cost_new) / denom
         accept = (rho > 0) & active
         stepn = dp.norm(dim=-1)
         pnorm = _jones_norm(J, nchunk)
         small = stepn < eps2 * (pnorm + eps2)
-        if bool(accept.any()):
-            J = torch.where(accept[:, None, None, None], Jnew, J)
-            cost = torch.where(accept, cost_new, cost)
-            fac = (1.0 - (2.0 * rho - 1.0) ** 3).clamp_min(1.0 / 3.0)
-            mu = torch.where(accept, mu * fac, mu)
-            nu = torch.where(accept, torch.full_like(nu, 2.0), nu)
+        J = torch.where(accept[:, None, None, None], Jnew, J)
+        cost = torch.where(accept, cost_new, cost)
+        fac = (1.0 - (2.0 * rho - 1.0) ** 3).clamp_min(1.0 / 3.0)
+        mu = torch.where(accept, mu * fac, mu)
+        nu = torch.where(accept, torch.full_like(nu, 2.0), nu)
         reject = (~accept) & active
-        if bool(reject.any()):
-            mu = torch.where(reject, mu * nu, mu)
-            nu = torch.where(reject, nu * 2.0, nu)
+        mu = torch.where(reject, mu * nu, mu)
+        nu = torch.where(reject, nu * 2.0, nu)
         active = active & ~small
-        if not bool(active.any()):
-            break
-        # recompute JtJ/Jtr at (possibly) new J
-        JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, prob.weights,
-                                  prob.chunk_rows, nchunk, prob.layout)
+        if it + 1 < maxiter:
+            # recompute JtJ/Jtr at (possibly) new J
+            JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, prob.weights,
+                                      prob.chunk_rows, nchunk, prob.layout)
+            gnorm = Jtr.abs().max(dim=-1).values
+            active = active & (gnorm > eps1)
+            if (it & 7) == 7 and not bool(active.any()):
+                break
     info = {'init_cost': init_cost, 'final_cost': cost, 'niter': niter}
     return J, info
 
 
 def _chol_solve(A, b):
-    """Batched SPD solve via Cholesky with jitter fallback."""
-    try:
-        L = torch.linalg.cholesky(A)
-        return torch.cholesky_solve(b.unsqueeze(-1), L).squeeze(-1)
-    except Exception:
-        jitter = 1e-6 * A.diagonal(dim1=-2, dim2=-1).mean(-1)
-        A2 = A + jitter[:, None, None] * torch.eye(
-            A.shape[-1], dtype=A.dtype, device=A.device)
-        try:
-            L = torch.linalg.cholesky(A2)
-            return torch.cholesky_solve(b.unsqueeze(-1), L).squeeze(-1)
-        except Exception:
-            return torch.linalg.lstsq(A2, b.unsqueeze(-1)).solution.squeeze(-1)
+    """Batched SPD solve via cholesky_ex (no host sync, no throw); a failed
+    factorization yields NaNs in that chunk's step, which the LM
+    accept/reject mask rejects (mu grows and the next A is better
+    conditioned) — self-healing without synchronization."""
+    L, info = torch.linalg.cholesky_ex(A)
+    return torch.cholesky_solve(b.unsqueeze(-1), L).squeeze(-1)
 
 
 def _vec_to_jones(dp, nchunk, N):
