@@ -22,6 +22,8 @@ hipError_t launch_model_cost(const float2*, const float2*, const float2*,
 hipError_t launch_apply_jones(const float2*, const float2*, const float2*,
     const int*, const int*, int, int, int, int, int, int, float2*,
     hipStream_t);
+hipError_t launch_chol_solve(const float*, const float*, const float*, int,
+    int, float*, float*, int*, hipStream_t);
 }
 
 #define CHECK_HIP(x) do { hipError_t e = (x); TORCH_CHECK(e == hipSuccess, \
@@ -116,7 +118,23 @@ torch::Tensor apply_jones(
   return out;
 }
 
+std::vector<torch::Tensor> chol_solve(
+    torch::Tensor JtJ, torch::Tensor Jtr, torch::Tensor mu,
+    torch::Tensor scratch) {
+  const int64_t batch = JtJ.size(0);
+  const int64_t n = JtJ.size(1);
+  auto dp = torch::empty({batch, n},
+      torch::dtype(torch::kFloat).device(JtJ.device()));
+  auto info = torch::zeros({batch},
+      torch::dtype(torch::kInt).device(JtJ.device()));
+  CHECK_HIP(launch_chol_solve(JtJ.data_ptr<float>(), Jtr.data_ptr<float>(),
+      mu.data_ptr<float>(), (int)n, (int)batch, scratch.data_ptr<float>(),
+      dp.data_ptr<float>(), info.data_ptr<int>(), cur_stream()));
+  return {dp, info};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("chol_solve", &chol_solve, "batched damped Cholesky solve (gfx950)");
   m.def("predict_coh", &predict_coh, "coherency predict (gfx950)");
   m.def("jtj_jtr", &jtj_jtr, "fused JtJ/Jtr assembly (gfx950)");
   m.def("model_cost", &model_cost, "per-chunk model cost (gfx950)");

@@ -4,6 +4,7 @@
 #include "common.h"
 #include "predict.hip"
 #include "jtj.hip"
+#include "cholesky.hip"
 
 extern "C" {
 
@@ -62,6 +63,14 @@ hipError_t launch_apply_jones(
   const int nb = (int)((B + tb - 1) / tb);
   hipLaunchKernelGGL(k_apply_jones, dim3(nb), dim3(tb), 0, stream,
       x, cohs, J, pairs, chunk_tab, Nbase, T, N, nseg, M, sub, out);
+  return hipGetLastError();
+}
+
+hipError_t launch_chol_solve(
+    const float* JtJ, const float* Jtr, const float* mu, int n, int batch,
+    float* Lbuf, float* dp, int* info, hipStream_t stream) {
+  hipLaunchKernelGGL(k_chol_solve, dim3(batch), dim3(256), 0, stream,
+      JtJ, Jtr, mu, n, Lbuf, dp, info);
   return hipGetLastError();
 }
 }  // extern "C"

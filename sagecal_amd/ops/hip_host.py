@@ -194,3 +194,23 @@ def lbfgs_cost_grad(x, cohs, J_packed, chunk_off, nchunks, bb, T, Nbase,
     from . import reference as R
     return R.lbfgs_cost_grad(x, cohs, J_packed, chunk_off, nchunks, bb, T,
                              Nbase, robust_nu, weights)
+
+
+_chol_scratch = {}
+
+
+def chol_solve_damped(JtJ, Jtr, mu):
+    """dp = (JtJ + mu I)^-1 Jtr via the fused gfx950 kernel (no A
+    materialization, no rocSOLVER). Failed factorizations return NaN rows
+    so the LM accept mask rejects them."""
+    key = (tuple(JtJ.shape), str(JtJ.device))
+    sc = _chol_scratch.get(key)
+    if sc is None or sc.shape != JtJ.shape:
+        sc = torch.empty_like(JtJ)
+        _chol_scratch[key] = sc
+    dp, info = _ext().chol_solve(JtJ.contiguous(), Jtr.contiguous(),
+                                 mu.to(torch.float32).contiguous(), sc)
+    bad = info != 0
+    if bad.any():
+        dp = torch.where(bad[:, None], torch.full_like(dp, float('nan')), dp)
+    return dp

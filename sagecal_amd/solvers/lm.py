@@ -64,10 +64,16 @@ def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
     diag_max = JtJ.diagonal(dim1=-2, dim2=-1).max(dim=-1).values
     mu = tau * diag_max
     niter = 0
+    use_hip_chol = (prob.layout is not None and x.is_cuda
+                    and 8 * N <= 4096)
     for it in range(maxiter):
         niter = it + 1
-        A = JtJ + mu[:, None, None] * eye
-        dp = _chol_solve(A, Jtr)
+        if use_hip_chol:
+            from ..ops.hip_host import chol_solve_damped
+            dp = chol_solve_damped(JtJ, Jtr, mu)
+        else:
+            A = JtJ + mu[:, None, None] * eye
+            dp = _chol_solve(A, Jtr)
         dp = torch.nan_to_num(dp, nan=0.0, posinf=0.0, neginf=0.0)
         dpc = _vec_to_jones(dp, nchunk, N)
         Jnew = J + dpc

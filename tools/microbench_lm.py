@@ -78,6 +78,14 @@ def main():
     print(f"lm_solve(maxiter=1):    {t:.3f} ms")
     t = timeit(lambda: lm_mod.lm_solve(prob, J, maxiter=8), n=5)
     print(f"lm_solve(maxiter=8):    {t:.3f} ms")
+    from sagecal_amd.ops.hip_host import chol_solve_damped
+    mu2 = torch.ones(2, device=dev) * 0.1
+    t = timeit(lambda: chol_solve_damped(JtJ, Jtr, mu2))
+    print(f"custom chol_solve:      {t:.3f} ms")
+    JtJ10 = JtJ.repeat(5, 1, 1); Jtr10 = Jtr.repeat(5, 1)
+    mu10 = mu2.repeat(5)
+    t = timeit(lambda: chol_solve_damped(JtJ10, Jtr10, mu10))
+    print(f"custom chol batch10:    {t:.3f} ms")
 
 
 if __name__ == '__main__':
