@@ -2,23 +2,24 @@
 //   dp = (JtJ + mu*I)^-1 Jtr   per problem, one workgroup per problem.
 //
 // Replaces rocSOLVER potrf/potrs for the LM normal equations (the
-// reference used cusolverDn potrf/potrs per cluster, clmfit_cuda.c:364) —
-// rocSOLVER's batched fp32 potrf measures ~2 ms for [2,512,512] on MI355X
-// (latency-bound internal loop); this kernel targets <0.15 ms.
-//
-// Algorithm: right-looking blocked Cholesky, panel width 32.
-//   - copy lower triangle of JtJ + mu on diag into scratch L (global)
-//   - per panel: factor 32x32 diag block in LDS; row-solve the sub-panel;
-//     trailing SYRK update with the panel's J-tile staged in LDS
-//   - blocked forward/backward substitution for the single RHS.
-// Non-SPD pivots are clamped and flagged (info) — the LM accept/reject
-// logic rejects the resulting step (mu grows), matching cholesky_ex
-// semantics without host sync.
+// reference used cusolverDn potrf/potrs per cluster, clmfit_cuda.c:364):
+// rocSOLVER's batched fp32 potrf measures ~2 ms for [2,512,512] on MI355X;
+// this kernel is built for low LATENCY at small batch (the LM inner loop
+// is a serial chain of solves):
+//   - 32x32 diagonal blocks factored WAVE-SYNCHRONOUSLY in registers:
+//     lane r holds row r, pivot/column broadcasts via __shfl — zero
+//     barriers, zero LDS traffic in the factor;
+//   - the whole panel (rows k..n x 32) staged in LDS once per panel;
+//     row-solve reads the diag block from LDS (broadcast-friendly);
+//   - trailing SYRK update: per-thread 4x4 register tiles over the
+//     LDS panel with float4 K-chunks.
+// Requires n <= MAXN_CHOL (LDS panel); larger systems (512-station joint
+// solves) go to rocSOLVER on the host side.
 #include "common.h"
 
 #define NB 32
 #define NTH 256
-#define JT 128   // J-tile rows staged in LDS for the trailing update
+#define MAXN_CHOL 1024
 
 extern "C" __global__ void __launch_bounds__(NTH)
 k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
@@ -27,19 +28,20 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
              int* __restrict__ info) {
   const int bid = blockIdx.x;
   const int tid = threadIdx.x;
+  const int lane = tid & 63;
   const float* A = JtJ + (size_t)bid * n * n;
   float* L = Lbuf + (size_t)bid * n * n;
   const float* b = Jtr + (size_t)bid * n;
   float* xo = dp + (size_t)bid * n;
 
-  __shared__ float dlds[NB][NB + 1];
-  __shared__ float jl[JT][NB + 1];
+  // dynamic LDS: pan[n][NB] (panel incl. diag block) + yv[NB]
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* pan = smem;                  // [rows][NB] rows = n - k
   __shared__ float yv[NB];
   __shared__ int bad;
 
   if (tid == 0) bad = 0;
   const float m = mu[bid];
-  // copy lower triangle + damping
   for (int idx = tid; idx < n * n; idx += NTH) {
     const int r = idx / n, c = idx - r * n;
     if (c <= r) L[idx] = A[idx] + (c == r ? m : 0.0f);
@@ -48,97 +50,132 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
 
   for (int k = 0; k < n; k += NB) {
     const int nb = min(NB, n - k);
-    // load diag block
-    for (int idx = tid; idx < nb * nb; idx += NTH) {
+    const int rows = n - k;
+    // stage panel rows k..n, cols k..k+nb into LDS
+    for (int idx = tid; idx < rows * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
-      dlds[r][c] = L[(size_t)(k + r) * n + k + c];
+      pan[r * NB + c] = L[(size_t)(k + r) * n + k + c];
     }
     __syncthreads();
-    // factor 32x32 block (unblocked, in LDS)
-    for (int c = 0; c < nb; ++c) {
-      if (tid == 0) {
-        float d = dlds[c][c];
-        if (d <= 1e-30f) { bad = 1; d = 1e-30f; }
-        dlds[c][c] = sqrtf(d);
-      }
-      __syncthreads();
-      const float piv = dlds[c][c];
-      for (int r = c + 1 + tid; r < nb; r += NTH) dlds[r][c] /= piv;
-      __syncthreads();
-      const int rem = nb - c - 1;
-      for (int idx = tid; idx < rem * rem; idx += NTH) {
-        const int rr = idx / rem, cc = idx - rr * rem;
-        if (cc <= rr) {
-          dlds[c + 1 + rr][c + 1 + cc] -=
-              dlds[c + 1 + rr][c] * dlds[c + 1 + cc][c];
-        }
-      }
-      __syncthreads();
-    }
-    // write factored diag block back
-    for (int idx = tid; idx < nb * nb; idx += NTH) {
-      const int r = idx / nb, c = idx - r * nb;
-      if (c <= r) L[(size_t)(k + r) * n + k + c] = dlds[r][c];
-    }
-    // panel row-solve: L[k+nb:n, k:k+nb] = A_panel * Lkk^-T
-    // row r: forward substitution against dlds (lower, transposed solve)
-    for (int r = k + nb + tid; r < n; r += NTH) {
+    // ---- wave-synchronous 32x32 factor: lanes 0..31 of wave 0 hold rows
+    if (tid < 64) {
+      const int r = lane & 31;
       float row[NB];
+      if (lane < 32) {
 #pragma unroll 8
-      for (int c = 0; c < nb; ++c) row[c] = L[(size_t)r * n + k + c];
+        for (int c = 0; c < nb; ++c) row[c] = pan[r * NB + c];
+      }
       for (int c = 0; c < nb; ++c) {
-        float s = row[c];
-        for (int c2 = 0; c2 < c; ++c2) s -= row[c2] * dlds[c][c2];
-        row[c] = s / dlds[c][c];
-      }
-#pragma unroll 8
-      for (int c = 0; c < nb; ++c) L[(size_t)r * n + k + c] = row[c];
-    }
-    __syncthreads();
-    // trailing update: for row-tiles J (cols) staged in LDS,
-    // L[i, j] -= dot(panel[i], panel[j]) for k+nb <= j <= i < n
-    for (int j0 = k + nb; j0 < n; j0 += JT) {
-      const int jt = min(JT, n - j0);
-      for (int idx = tid; idx < jt * nb; idx += NTH) {
-        const int r = idx / nb, c = idx - r * nb;
-        jl[r][c] = L[(size_t)(j0 + r) * n + k + c];
-      }
-      __syncthreads();
-      // entries: i from j0.., j in tile, i >= j
-      for (int i = j0 + tid / 32; i < n; i += NTH / 32) {
-        float pr[NB];
-#pragma unroll 8
-        for (int c = 0; c < nb; ++c) pr[c] = L[(size_t)i * n + k + c];
-        const int lane8 = tid % 32;
-        const int jmax = min(jt, i - j0 + 1);
-        for (int jj = lane8; jj < jmax; jj += 32) {
-          float s = 0.f;
-#pragma unroll 8
-          for (int c = 0; c < nb; ++c) s += pr[c] * jl[jj][c];
-          L[(size_t)i * n + j0 + jj] -= s;
+        // pivot from lane c
+        float pv = __shfl(row[c], c, 64);
+        if (lane == c) {
+          if (pv <= 1e-30f) { bad = 1; pv = 1e-30f; }
+          pv = sqrtf(pv);
+          row[c] = pv;
+        }
+        pv = __shfl(row[c], c, 64);
+        if (lane < 32 && r > c) {
+          row[c] /= pv;
+          // rank-1 update needs L[cc][c] for cc in (c, r]; get from the
+          // lanes' own row[c] via shfl inside the cc loop
+        }
+        // all lanes update their trailing cols cc>c (only rows r>cc matter)
+        for (int cc = c + 1; cc < nb; ++cc) {
+          const float lcc = __shfl(row[c], cc, 64);
+          if (lane < 32 && r >= cc) row[cc] -= row[c] * lcc;
         }
       }
-      __syncthreads();
+      if (lane < 32) {
+#pragma unroll 8
+        for (int c = 0; c < nb; ++c) pan[r * NB + c] = row[c];
+      }
     }
+    __syncthreads();
+    // ---- row-solve sub-panel rows nb..rows against the diag block
+    for (int r = nb + tid; r < rows; r += NTH) {
+      float rw[NB];
+#pragma unroll 8
+      for (int c = 0; c < nb; ++c) rw[c] = pan[r * NB + c];
+      for (int c = 0; c < nb; ++c) {
+        float s = rw[c];
+        for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * pan[c * NB + c2];
+        rw[c] = s / pan[c * NB + c];
+      }
+#pragma unroll 8
+      for (int c = 0; c < nb; ++c) pan[r * NB + c] = rw[c];
+    }
+    __syncthreads();
+    // write panel back (final L values)
+    for (int idx = tid; idx < rows * nb; idx += NTH) {
+      const int r = idx / nb, c = idx - r * nb;
+      L[(size_t)(k + r) * n + k + c] = pan[r * NB + c];
+    }
+    // ---- trailing SYRK: L[i,j] -= dot(pan[i], pan[j]), 4x4 reg tiles
+    const int rows2 = rows - nb;
+    if (rows2 > 0) {
+      const int ntI = (rows2 + 3) >> 2;
+      const int ntiles = ntI * (ntI + 1) / 2;
+      for (int tile = tid; tile < ntiles; tile += NTH) {
+        // triangular decode: I = row-tile, Jt = col-tile (Jt <= I)
+        int I = (int)((sqrtf(8.0f * tile + 1.0f) - 1.0f) * 0.5f);
+        while (I * (I + 1) / 2 > tile) --I;
+        while ((I + 1) * (I + 2) / 2 <= tile) ++I;
+        const int Jt = tile - I * (I + 1) / 2;
+        const float* pi = pan + (nb + I * 4) * NB;
+        const float* pj = pan + (nb + Jt * 4) * NB;
+        float acc[4][4] = {};
+        for (int c = 0; c < nb; c += 4) {
+          float4 a0 = *(const float4*)(pi + 0 * NB + c);
+          float4 a1 = *(const float4*)(pi + 1 * NB + c);
+          float4 a2 = *(const float4*)(pi + 2 * NB + c);
+          float4 a3 = *(const float4*)(pi + 3 * NB + c);
+          float4 b0 = *(const float4*)(pj + 0 * NB + c);
+          float4 b1 = *(const float4*)(pj + 1 * NB + c);
+          float4 b2 = *(const float4*)(pj + 2 * NB + c);
+          float4 b3 = *(const float4*)(pj + 3 * NB + c);
+          const float4 aa[4] = {a0, a1, a2, a3};
+          const float4 bbv[4] = {b0, b1, b2, b3};
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              acc[i][j] += aa[i].x * bbv[j].x + aa[i].y * bbv[j].y
+                         + aa[i].z * bbv[j].z + aa[i].w * bbv[j].w;
+            }
+        }
+        const int gi0 = k + nb + I * 4, gj0 = k + nb + Jt * 4;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int gi = gi0 + i;
+          if (gi >= n) break;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            const int gj = gj0 + j;
+            if (gj <= gi && gj < n)
+              L[(size_t)gi * n + gj] -= acc[i][j];
+          }
+        }
+      }
+    }
+    __syncthreads();
   }
 
-  // ---- forward substitution: solve L y = b (y kept in xo) ----
+  // ---- forward substitution: L y = b (y in xo) ----
   for (int idx = tid; idx < n; idx += NTH) xo[idx] = b[idx];
   __syncthreads();
   for (int k = 0; k < n; k += NB) {
     const int nb = min(NB, n - k);
-    // stage Lkk into LDS, then serial in-LDS triangular solve (fast)
     for (int idx = tid; idx < nb * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
-      if (c <= r) dlds[r][c] = L[(size_t)(k + r) * n + k + c];
+      if (c <= r) pan[r * NB + c] = L[(size_t)(k + r) * n + k + c];
     }
     if (tid < nb) yv[tid] = xo[k + tid];
     __syncthreads();
     if (tid == 0) {
       for (int c = 0; c < nb; ++c) {
         float s = yv[c];
-        for (int c2 = 0; c2 < c; ++c2) s -= dlds[c][c2] * yv[c2];
-        yv[c] = s / dlds[c][c];
+        for (int c2 = 0; c2 < c; ++c2) s -= pan[c * NB + c2] * yv[c2];
+        yv[c] = s / pan[c * NB + c];
       }
     }
     __syncthreads();
@@ -151,20 +188,20 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     }
     __syncthreads();
   }
-  // ---- backward substitution: solve L^T x = y ----
+  // ---- backward substitution: L^T x = y ----
   for (int k = ((n - 1) / NB) * NB; k >= 0; k -= NB) {
     const int nb = min(NB, n - k);
     for (int idx = tid; idx < nb * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
-      if (c <= r) dlds[r][c] = L[(size_t)(k + r) * n + k + c];
+      if (c <= r) pan[r * NB + c] = L[(size_t)(k + r) * n + k + c];
     }
     if (tid < nb) yv[tid] = xo[k + tid];
     __syncthreads();
     if (tid == 0) {
       for (int c = nb - 1; c >= 0; --c) {
         float s = yv[c];
-        for (int c2 = c + 1; c2 < nb; ++c2) s -= dlds[c2][c] * yv[c2];
-        yv[c] = s / dlds[c][c];
+        for (int c2 = c + 1; c2 < nb; ++c2) s -= pan[c2 * NB + c] * yv[c2];
+        yv[c] = s / pan[c * NB + c];
       }
     }
     __syncthreads();
@@ -173,6 +210,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     // update rows above: xo[i] -= L[k+c, i] * x[k+c] for i < k
     for (int i = tid; i < k; i += NTH) {
       float s = 0.f;
+#pragma unroll 8
       for (int c = 0; c < nb; ++c) s += L[(size_t)(k + c) * n + i] * yv[c];
       xo[i] -= s;
     }

@@ -65,7 +65,7 @@ def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
     mu = tau * diag_max
     niter = 0
     use_hip_chol = (prob.layout is not None and x.is_cuda
-                    and 8 * N <= 4096)
+                    and 8 * N <= 1024)
     for it in range(maxiter):
         niter = it + 1
         if use_hip_chol:
