@@ -18,6 +18,7 @@
 #include "common.h"
 
 #define NB 32
+#define PST 36   // padded panel row stride (floats): bank-conflict-free float4 reads
 #define NTH 256
 #define MAXN_CHOL 1024
 
@@ -54,7 +55,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     // stage panel rows k..n, cols k..k+nb into LDS
     for (int idx = tid; idx < rows * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
-      pan[r * NB + c] = L[(size_t)(k + r) * n + k + c];
+      pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];
     }
     __syncthreads();
     // ---- wave-synchronous 32x32 factor: lanes 0..31 of wave 0 hold rows
@@ -63,7 +64,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
       float row[NB];
       if (lane < 32) {
 #pragma unroll 8
-        for (int c = 0; c < nb; ++c) row[c] = pan[r * NB + c];
+        for (int c = 0; c < nb; ++c) row[c] = pan[r * PST + c];
       }
       for (int c = 0; c < nb; ++c) {
         // pivot from lane c
@@ -87,7 +88,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
       }
       if (lane < 32) {
 #pragma unroll 8
-        for (int c = 0; c < nb; ++c) pan[r * NB + c] = row[c];
+        for (int c = 0; c < nb; ++c) pan[r * PST + c] = row[c];
       }
     }
     __syncthreads();
@@ -95,20 +96,20 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     for (int r = nb + tid; r < rows; r += NTH) {
       float rw[NB];
 #pragma unroll 8
-      for (int c = 0; c < nb; ++c) rw[c] = pan[r * NB + c];
+      for (int c = 0; c < nb; ++c) rw[c] = pan[r * PST + c];
       for (int c = 0; c < nb; ++c) {
         float s = rw[c];
-        for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * pan[c * NB + c2];
-        rw[c] = s / pan[c * NB + c];
+        for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * pan[c * PST + c2];
+        rw[c] = s / pan[c * PST + c];
       }
 #pragma unroll 8
-      for (int c = 0; c < nb; ++c) pan[r * NB + c] = rw[c];
+      for (int c = 0; c < nb; ++c) pan[r * PST + c] = rw[c];
     }
     __syncthreads();
     // write panel back (final L values)
     for (int idx = tid; idx < rows * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
-      L[(size_t)(k + r) * n + k + c] = pan[r * NB + c];
+      L[(size_t)(k + r) * n + k + c] = pan[r * PST + c];
     }
     // ---- trailing SYRK: L[i,j] -= dot(pan[i], pan[j]), 4x4 reg tiles
     const int rows2 = rows - nb;
@@ -121,18 +122,18 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
         while (I * (I + 1) / 2 > tile) --I;
         while ((I + 1) * (I + 2) / 2 <= tile) ++I;
         const int Jt = tile - I * (I + 1) / 2;
-        const float* pi = pan + (nb + I * 4) * NB;
-        const float* pj = pan + (nb + Jt * 4) * NB;
+        const float* pi = pan + (nb + I * 4) * PST;
+        const float* pj = pan + (nb + Jt * 4) * PST;
         float acc[4][4] = {};
         for (int c = 0; c < nb; c += 4) {
-          float4 a0 = *(const float4*)(pi + 0 * NB + c);
-          float4 a1 = *(const float4*)(pi + 1 * NB + c);
-          float4 a2 = *(const float4*)(pi + 2 * NB + c);
-          float4 a3 = *(const float4*)(pi + 3 * NB + c);
-          float4 b0 = *(const float4*)(pj + 0 * NB + c);
-          float4 b1 = *(const float4*)(pj + 1 * NB + c);
-          float4 b2 = *(const float4*)(pj + 2 * NB + c);
-          float4 b3 = *(const float4*)(pj + 3 * NB + c);
+          float4 a0 = *(const float4*)(pi + 0 * PST + c);
+          float4 a1 = *(const float4*)(pi + 1 * PST + c);
+          float4 a2 = *(const float4*)(pi + 2 * PST + c);
+          float4 a3 = *(const float4*)(pi + 3 * PST + c);
+          float4 b0 = *(const float4*)(pj + 0 * PST + c);
+          float4 b1 = *(const float4*)(pj + 1 * PST + c);
+          float4 b2 = *(const float4*)(pj + 2 * PST + c);
+          float4 b3 = *(const float4*)(pj + 3 * PST + c);
           const float4 aa[4] = {a0, a1, a2, a3};
           const float4 bbv[4] = {b0, b1, b2, b3};
 #pragma unroll
@@ -167,15 +168,15 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     const int nb = min(NB, n - k);
     for (int idx = tid; idx < nb * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
-      if (c <= r) pan[r * NB + c] = L[(size_t)(k + r) * n + k + c];
+      if (c <= r) pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];
     }
     if (tid < nb) yv[tid] = xo[k + tid];
     __syncthreads();
     if (tid == 0) {
       for (int c = 0; c < nb; ++c) {
         float s = yv[c];
-        for (int c2 = 0; c2 < c; ++c2) s -= pan[c * NB + c2] * yv[c2];
-        yv[c] = s / pan[c * NB + c];
+        for (int c2 = 0; c2 < c; ++c2) s -= pan[c * PST + c2] * yv[c2];
+        yv[c] = s / pan[c * PST + c];
       }
     }
     __syncthreads();
@@ -193,15 +194,15 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     const int nb = min(NB, n - k);
     for (int idx = tid; idx < nb * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
-      if (c <= r) pan[r * NB + c] = L[(size_t)(k + r) * n + k + c];
+      if (c <= r) pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];
     }
     if (tid < nb) yv[tid] = xo[k + tid];
     __syncthreads();
     if (tid == 0) {
       for (int c = nb - 1; c >= 0; --c) {
         float s = yv[c];
-        for (int c2 = c + 1; c2 < nb; ++c2) s -= pan[c2 * NB + c] * yv[c2];
-        yv[c] = s / pan[c * NB + c];
+        for (int c2 = c + 1; c2 < nb; ++c2) s -= pan[c2 * PST + c] * yv[c2];
+        yv[c] = s / pan[c * PST + c];
       }
     }
     __syncthreads();
@@ -216,5 +217,10 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     }
     __syncthreads();
   }
-  if (tid == 0 && bad) atomicOr(&info[bid], 1);
+  if (bad) {
+    // poison the step so the LM accept mask rejects it (no host sync)
+    const float qn = __int_as_float(0x7fc00000);
+    for (int idx = tid; idx < n; idx += NTH) xo[idx] = qn;
+    if (tid == 0) atomicOr(&info[bid], 1);
+  }
 }

@@ -210,7 +210,5 @@ def chol_solve_damped(JtJ, Jtr, mu):
         _chol_scratch[key] = sc
     dp, info = _ext().chol_solve(JtJ.contiguous(), Jtr.contiguous(),
                                  mu.to(torch.float32).contiguous(), sc)
-    bad = info != 0
-    if bad.any():
-        dp = torch.where(bad[:, None], torch.full_like(dp, float('nan')), dp)
+    # failed factorizations already return NaN rows (kernel poisons dp)
     return dp
