@@ -50,7 +50,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
   __syncthreads();
 
   for (int k = 0; k < n; k += NB) {
-    const int nb = min(NB, n - k);
+    const int nb = NB;            // host pads n to a multiple of NB
     const int rows = n - k;
     // stage panel rows k..n, cols k..k+nb into LDS
     for (int idx = tid; idx < rows * nb; idx += NTH) {
@@ -63,10 +63,11 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
       const int r = lane & 31;
       float row[NB];
       if (lane < 32) {
-#pragma unroll 8
+#pragma unroll
         for (int c = 0; c < nb; ++c) row[c] = pan[r * PST + c];
       }
-      for (int c = 0; c < nb; ++c) {
+#pragma unroll
+      for (int c = 0; c < NB; ++c) {
         // pivot from lane c
         float pv = __shfl(row[c], c, 64);
         if (lane == c) {
@@ -81,13 +82,14 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
           // lanes' own row[c] via shfl inside the cc loop
         }
         // all lanes update their trailing cols cc>c (only rows r>cc matter)
-        for (int cc = c + 1; cc < nb; ++cc) {
+#pragma unroll
+        for (int cc = c + 1; cc < NB; ++cc) {
           const float lcc = __shfl(row[c], cc, 64);
           if (lane < 32 && r >= cc) row[cc] -= row[c] * lcc;
         }
       }
       if (lane < 32) {
-#pragma unroll 8
+#pragma unroll
         for (int c = 0; c < nb; ++c) pan[r * PST + c] = row[c];
       }
     }
@@ -95,15 +97,17 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     // ---- row-solve sub-panel rows nb..rows against the diag block
     for (int r = nb + tid; r < rows; r += NTH) {
       float rw[NB];
-#pragma unroll 8
-      for (int c = 0; c < nb; ++c) rw[c] = pan[r * PST + c];
-      for (int c = 0; c < nb; ++c) {
+#pragma unroll
+      for (int c = 0; c < NB; ++c) rw[c] = pan[r * PST + c];
+#pragma unroll
+      for (int c = 0; c < NB; ++c) {
         float s = rw[c];
+#pragma unroll
         for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * pan[c * PST + c2];
         rw[c] = s / pan[c * PST + c];
       }
-#pragma unroll 8
-      for (int c = 0; c < nb; ++c) pan[r * PST + c] = rw[c];
+#pragma unroll
+      for (int c = 0; c < NB; ++c) pan[r * PST + c] = rw[c];
     }
     __syncthreads();
     // write panel back (final L values)
@@ -125,7 +129,8 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
         const float* pi = pan + (nb + I * 4) * PST;
         const float* pj = pan + (nb + Jt * 4) * PST;
         float acc[4][4] = {};
-        for (int c = 0; c < nb; c += 4) {
+#pragma unroll
+        for (int c = 0; c < NB; c += 4) {
           float4 a0 = *(const float4*)(pi + 0 * PST + c);
           float4 a1 = *(const float4*)(pi + 1 * PST + c);
           float4 a2 = *(const float4*)(pi + 2 * PST + c);
@@ -165,7 +170,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
   for (int idx = tid; idx < n; idx += NTH) xo[idx] = b[idx];
   __syncthreads();
   for (int k = 0; k < n; k += NB) {
-    const int nb = min(NB, n - k);
+    const int nb = NB;
     for (int idx = tid; idx < nb * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
       if (c <= r) pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];
@@ -183,15 +188,15 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     if (tid < nb) xo[k + tid] = yv[tid];
     for (int i = k + nb + tid; i < n; i += NTH) {
       float s = 0.f;
-#pragma unroll 8
-      for (int c = 0; c < nb; ++c) s += L[(size_t)i * n + k + c] * yv[c];
+#pragma unroll
+      for (int c = 0; c < NB; ++c) s += L[(size_t)i * n + k + c] * yv[c];
       xo[i] -= s;
     }
     __syncthreads();
   }
   // ---- backward substitution: L^T x = y ----
   for (int k = ((n - 1) / NB) * NB; k >= 0; k -= NB) {
-    const int nb = min(NB, n - k);
+    const int nb = NB;
     for (int idx = tid; idx < nb * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
       if (c <= r) pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];

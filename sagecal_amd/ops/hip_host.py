@@ -202,7 +202,19 @@ _chol_scratch = {}
 def chol_solve_damped(JtJ, Jtr, mu):
     """dp = (JtJ + mu I)^-1 Jtr via the fused gfx950 kernel (no A
     materialization, no rocSOLVER). Failed factorizations return NaN rows
-    so the LM accept mask rejects them."""
+    so the LM accept mask rejects them. n is padded to a multiple of 32
+    with identity diagonal (kernel requires it; pad solutions are 0)."""
+    n = JtJ.shape[1]
+    npad = (n + 31) // 32 * 32
+    if npad != n:
+        batch = JtJ.shape[0]
+        J2 = torch.zeros(batch, npad, npad, dtype=JtJ.dtype,
+                         device=JtJ.device)
+        J2[:, :n, :n] = JtJ
+        J2[:, range(n, npad), range(n, npad)] = 1.0
+        b2 = torch.zeros(batch, npad, dtype=Jtr.dtype, device=Jtr.device)
+        b2[:, :n] = Jtr
+        JtJ, Jtr = J2, b2
     key = (tuple(JtJ.shape), str(JtJ.device))
     sc = _chol_scratch.get(key)
     if sc is None or sc.shape != JtJ.shape:
@@ -211,4 +223,4 @@ def chol_solve_damped(JtJ, Jtr, mu):
     dp, info = _ext().chol_solve(JtJ.contiguous(), Jtr.contiguous(),
                                  mu.to(torch.float32).contiguous(), sc)
     # failed factorizations already return NaN rows (kernel poisons dp)
-    return dp
+    return dp[:, :n]
