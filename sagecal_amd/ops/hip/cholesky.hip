@@ -26,7 +26,7 @@ extern "C" __global__ void __launch_bounds__(NTH)
 k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
              const float* __restrict__ mu, int n,
              float* __restrict__ Lbuf, float* __restrict__ dp,
-             int* __restrict__ info) {
+             int* __restrict__ info, int stages) {
   const int bid = blockIdx.x;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -48,10 +48,12 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     if (c <= r) L[idx] = A[idx] + (c == r ? m : 0.0f);
   }
   __syncthreads();
+  if (stages == 0) return;       // ablation: copy only
 
   for (int k = 0; k < n; k += NB) {
     const int nb = NB;            // host pads n to a multiple of NB
     const int rows = n - k;
+    if (stages == 1 && k > 0) break;   // ablation: one panel
     // stage panel rows k..n, cols k..k+nb into LDS
     for (int idx = tid; idx < rows * nb; idx += NTH) {
       const int r = idx / nb, c = idx - r * nb;
@@ -165,6 +167,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     }
     __syncthreads();
   }
+  if (stages <= 2) return;       // ablation: factorization only
 
   // ---- forward substitution: L y = b (y in xo) ----
   for (int idx = tid; idx < n; idx += NTH) xo[idx] = b[idx];

@@ -67,10 +67,10 @@ hipError_t launch_apply_jones(
 
 hipError_t launch_chol_solve(
     const float* JtJ, const float* Jtr, const float* mu, int n, int batch,
-    float* Lbuf, float* dp, int* info, hipStream_t stream) {
+    float* Lbuf, float* dp, int* info, int stages, hipStream_t stream) {
   const size_t shmem = (size_t)(n + 4) * PST * sizeof(float);
   hipLaunchKernelGGL(k_chol_solve, dim3(batch), dim3(256), shmem, stream,
-      JtJ, Jtr, mu, n, Lbuf, dp, info);
+      JtJ, Jtr, mu, n, Lbuf, dp, info, stages);
   return hipGetLastError();
 }
 }  // extern "C"

@@ -221,6 +221,6 @@ def chol_solve_damped(JtJ, Jtr, mu):
         sc = torch.empty_like(JtJ)
         _chol_scratch[key] = sc
     dp, info = _ext().chol_solve(JtJ.contiguous(), Jtr.contiguous(),
-                                 mu.to(torch.float32).contiguous(), sc)
+                                 mu.to(torch.float32).contiguous(), sc, 3)
     # failed factorizations already return NaN rows (kernel poisons dp)
     return dp[:, :n]
