@@ -3,22 +3,26 @@
 //
 // Replaces rocSOLVER potrf/potrs for the LM normal equations (the
 // reference used cusolverDn potrf/potrs per cluster, clmfit_cuda.c:364):
-// rocSOLVER's batched fp32 potrf measures ~2 ms for [2,512,512] on MI355X;
-// this kernel is built for low LATENCY at small batch (the LM inner loop
-// is a serial chain of solves):
-//   - 32x32 diagonal blocks factored WAVE-SYNCHRONOUSLY in registers:
-//     lane r holds row r, pivot/column broadcasts via __shfl — zero
-//     barriers, zero LDS traffic in the factor;
-//   - the whole panel (rows k..n x 32) staged in LDS once per panel;
-//     row-solve reads the diag block from LDS (broadcast-friendly);
-//   - trailing SYRK update: per-thread 4x4 register tiles over the
-//     LDS panel with float4 K-chunks.
-// Requires n <= MAXN_CHOL (LDS panel); larger systems (512-station joint
-// solves) go to rocSOLVER on the host side.
+// rocSOLVER's batched fp32 potrf measures ~2 ms for [2,512,512] on MI355X.
+// This kernel is built for low LATENCY at small batch (the LM inner loop is
+// a serial chain of solves). A single workgroup is bandwidth-limited, so
+// the design minimizes global traffic and keeps every hot loop
+// compile-time unrolled (n padded to a multiple of 32 by the host):
+//   - float4-vectorized triangle copy with fused damping;
+//   - 32x32 diagonal blocks factored WAVE-SYNCHRONOUSLY in registers
+//     (lane r holds row r; pivot broadcasts via __shfl; no barriers);
+//   - whole panel staged in LDS (padded stride, conflict-free float4);
+//   - trailing SYRK in 8x8 register tiles with float4 K-chunks and
+//     vectorized read-modify-write of the trailing matrix;
+//   - both triangular substitutions wave-synchronous in-panel, with an
+//     L^T mirror written during the panel write-back so the backward pass
+//     reads rows (coalesced) instead of columns.
+// Requires n <= MAXN_CHOL; larger systems go to rocSOLVER host-side.
+// NOTE: Lbuf scratch is 2*n*n floats per problem (L and L^T).
 #include "common.h"
 
 #define NB 32
-#define PST 36   // padded panel row stride (floats): bank-conflict-free float4 reads
+#define PST 36   // padded LDS panel row stride (floats)
 #define NTH 256
 #define MAXN_CHOL 1024
 
@@ -31,46 +35,59 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const float* A = JtJ + (size_t)bid * n * n;
-  float* L = Lbuf + (size_t)bid * n * n;
+  float* L = Lbuf + (size_t)bid * 2 * n * n;       // [L | L^T] scratch
+  float* LT = L + (size_t)n * n;
   const float* b = Jtr + (size_t)bid * n;
   float* xo = dp + (size_t)bid * n;
 
-  // dynamic LDS: pan[n][NB] (panel incl. diag block) + yv[NB]
   extern __shared__ __attribute__((aligned(16))) float smem[];
-  float* pan = smem;                  // [rows][NB] rows = n - k
+  float* pan = smem;                  // [rows][PST]
   __shared__ float yv[NB];
   __shared__ int bad;
 
   if (tid == 0) bad = 0;
   const float m = mu[bid];
-  for (int idx = tid; idx < n * n; idx += NTH) {
-    const int r = idx / n, c = idx - r * n;
-    if (c <= r) L[idx] = A[idx] + (c == r ? m : 0.0f);
+  // vectorized copy of lower-triangle rows (quads) + fused damping
+  {
+    const int nv = n >> 2;
+    for (int idx = tid; idx < n * nv; idx += NTH) {
+      const int r = idx / nv, c4 = (idx - r * nv) << 2;
+      if (c4 <= r) {
+        float4 v = *(const float4*)(A + (size_t)r * n + c4);
+        const int d = r - c4;
+        if (d < 4) {   // diagonal falls in this quad
+          if (d == 0) v.x += m;
+          else if (d == 1) v.y += m;
+          else if (d == 2) v.z += m;
+          else v.w += m;
+        }
+        *(float4*)(L + (size_t)r * n + c4) = v;
+      }
+    }
   }
   __syncthreads();
-  if (stages == 0) return;       // ablation: copy only
+  if (stages == 0) return;
 
   for (int k = 0; k < n; k += NB) {
-    const int nb = NB;            // host pads n to a multiple of NB
     const int rows = n - k;
-    if (stages == 1 && k > 0) break;   // ablation: one panel
-    // stage panel rows k..n, cols k..k+nb into LDS
-    for (int idx = tid; idx < rows * nb; idx += NTH) {
-      const int r = idx / nb, c = idx - r * nb;
-      pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];
+    if (stages == 1 && k > 0) break;
+    // stage panel rows k..n, cols k..k+NB into LDS (float4)
+    for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
+      const int r = idx >> 3, c4 = (idx & 7) << 2;
+      *(float4*)(pan + r * PST + c4) =
+          *(const float4*)(L + (size_t)(k + r) * n + k + c4);
     }
     __syncthreads();
-    // ---- wave-synchronous 32x32 factor: lanes 0..31 of wave 0 hold rows
+    // ---- wave-synchronous 32x32 factor on wave 0
     if (tid < 64) {
       const int r = lane & 31;
       float row[NB];
       if (lane < 32) {
 #pragma unroll
-        for (int c = 0; c < nb; ++c) row[c] = pan[r * PST + c];
+        for (int c = 0; c < NB; ++c) row[c] = pan[r * PST + c];
       }
 #pragma unroll
       for (int c = 0; c < NB; ++c) {
-        // pivot from lane c
         float pv = __shfl(row[c], c, 64);
         if (lane == c) {
           if (pv <= 1e-30f) { bad = 1; pv = 1e-30f; }
@@ -78,12 +95,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
           row[c] = pv;
         }
         pv = __shfl(row[c], c, 64);
-        if (lane < 32 && r > c) {
-          row[c] /= pv;
-          // rank-1 update needs L[cc][c] for cc in (c, r]; get from the
-          // lanes' own row[c] via shfl inside the cc loop
-        }
-        // all lanes update their trailing cols cc>c (only rows r>cc matter)
+        if (lane < 32 && r > c) row[c] /= pv;
 #pragma unroll
         for (int cc = c + 1; cc < NB; ++cc) {
           const float lcc = __shfl(row[c], cc, 64);
@@ -92,12 +104,12 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
       }
       if (lane < 32) {
 #pragma unroll
-        for (int c = 0; c < nb; ++c) pan[r * PST + c] = row[c];
+        for (int c = 0; c < NB; ++c) pan[r * PST + c] = row[c];
       }
     }
     __syncthreads();
-    // ---- row-solve sub-panel rows nb..rows against the diag block
-    for (int r = nb + tid; r < rows; r += NTH) {
+    // ---- row-solve sub-panel rows NB..rows against the diag block
+    for (int r = NB + tid; r < rows; r += NTH) {
       float rw[NB];
 #pragma unroll
       for (int c = 0; c < NB; ++c) rw[c] = pan[r * PST + c];
@@ -112,121 +124,152 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
       for (int c = 0; c < NB; ++c) pan[r * PST + c] = rw[c];
     }
     __syncthreads();
-    // write panel back (final L values)
-    for (int idx = tid; idx < rows * nb; idx += NTH) {
-      const int r = idx / nb, c = idx - r * nb;
-      L[(size_t)(k + r) * n + k + c] = pan[r * PST + c];
+    // write panel back: L (row-major, float4) and the L^T mirror
+    for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
+      const int r = idx >> 3, c4 = (idx & 7) << 2;
+      *(float4*)(L + (size_t)(k + r) * n + k + c4) =
+          *(const float4*)(pan + r * PST + c4);
     }
-    // ---- trailing SYRK: L[i,j] -= dot(pan[i], pan[j]), 4x4 reg tiles
-    const int rows2 = rows - nb;
+    for (int idx = tid; idx < rows * NB; idx += NTH) {
+      const int r = idx >> 5, c = idx & 31;
+      LT[(size_t)(k + c) * n + k + r] = pan[r * PST + c];
+    }
+    // ---- trailing SYRK: 4x8 register tiles (rows x cols), float4 K
+    // row-tiles of 4, col-tiles of 8: tile grid is ntI x ntJ lower band
+    const int rows2 = rows - NB;
     if (rows2 > 0) {
-      const int ntI = (rows2 + 3) >> 2;
-      const int ntiles = ntI * (ntI + 1) / 2;
+      const int ntI = (rows2 + 3) >> 2;   // 4-row tiles
+      const int ntJ = (rows2 + 7) >> 3;   // 8-col tiles
+      const int ntiles = ntI * ntJ;       // rectangular; skip above-diag
       for (int tile = tid; tile < ntiles; tile += NTH) {
-        // triangular decode: I = row-tile, Jt = col-tile (Jt <= I)
-        int I = (int)((sqrtf(8.0f * tile + 1.0f) - 1.0f) * 0.5f);
-        while (I * (I + 1) / 2 > tile) --I;
-        while ((I + 1) * (I + 2) / 2 <= tile) ++I;
-        const int Jt = tile - I * (I + 1) / 2;
-        const float* pi = pan + (nb + I * 4) * PST;
-        const float* pj = pan + (nb + Jt * 4) * PST;
-        float acc[4][4] = {};
+        const int I = tile / ntJ;
+        const int Jt = tile - I * ntJ;
+        const int gi0 = k + NB + I * 4, gj0 = k + NB + Jt * 8;
+        if (gj0 > gi0 + 3) continue;      // tile entirely above diagonal
+        const float* pi = pan + (NB + I * 4) * PST;
+        const float* pj = pan + (NB + Jt * 8) * PST;
+        float acc[4][8] = {};
 #pragma unroll
         for (int c = 0; c < NB; c += 4) {
-          float4 a0 = *(const float4*)(pi + 0 * PST + c);
-          float4 a1 = *(const float4*)(pi + 1 * PST + c);
-          float4 a2 = *(const float4*)(pi + 2 * PST + c);
-          float4 a3 = *(const float4*)(pi + 3 * PST + c);
-          float4 b0 = *(const float4*)(pj + 0 * PST + c);
-          float4 b1 = *(const float4*)(pj + 1 * PST + c);
-          float4 b2 = *(const float4*)(pj + 2 * PST + c);
-          float4 b3 = *(const float4*)(pj + 3 * PST + c);
-          const float4 aa[4] = {a0, a1, a2, a3};
-          const float4 bbv[4] = {b0, b1, b2, b3};
+          float4 av[4], bv[8];
+#pragma unroll
+          for (int i = 0; i < 4; ++i) av[i] = *(const float4*)(pi + i * PST + c);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) bv[j] = *(const float4*)(pj + j * PST + c);
 #pragma unroll
           for (int i = 0; i < 4; ++i)
 #pragma unroll
-            for (int j = 0; j < 4; ++j) {
-              acc[i][j] += aa[i].x * bbv[j].x + aa[i].y * bbv[j].y
-                         + aa[i].z * bbv[j].z + aa[i].w * bbv[j].w;
-            }
+            for (int j = 0; j < 8; ++j)
+              acc[i][j] += av[i].x * bv[j].x + av[i].y * bv[j].y
+                         + av[i].z * bv[j].z + av[i].w * bv[j].w;
         }
-        const int gi0 = k + nb + I * 4, gj0 = k + nb + Jt * 4;
+        if (gj0 + 7 <= gi0 && gi0 + 4 <= n) {
+          // interior tile fully below the diagonal -> vectorized RMW
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          const int gi = gi0 + i;
-          if (gi >= n) break;
+          for (int i = 0; i < 4; ++i) {
+            float* Lr = L + (size_t)(gi0 + i) * n + gj0;
+            float4 v0 = *(float4*)(Lr), v1 = *(float4*)(Lr + 4);
+            v0.x -= acc[i][0]; v0.y -= acc[i][1];
+            v0.z -= acc[i][2]; v0.w -= acc[i][3];
+            v1.x -= acc[i][4]; v1.y -= acc[i][5];
+            v1.z -= acc[i][6]; v1.w -= acc[i][7];
+            *(float4*)(Lr) = v0; *(float4*)(Lr + 4) = v1;
+          }
+        } else {
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            const int gj = gj0 + j;
-            if (gj <= gi && gj < n)
-              L[(size_t)gi * n + gj] -= acc[i][j];
+          for (int i = 0; i < 4; ++i) {
+            const int gi = gi0 + i;
+            if (gi >= n) break;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              const int gj = gj0 + j;
+              if (gj <= gi && gj < n) L[(size_t)gi * n + gj] -= acc[i][j];
+            }
           }
         }
       }
     }
     __syncthreads();
   }
-  if (stages <= 2) return;       // ablation: factorization only
+  if (stages <= 2) return;
 
-  // ---- forward substitution: L y = b (y in xo) ----
+  // ---- forward substitution: L y = b (y in xo), wave-sync in-panel ----
   for (int idx = tid; idx < n; idx += NTH) xo[idx] = b[idx];
   __syncthreads();
   for (int k = 0; k < n; k += NB) {
-    const int nb = NB;
-    for (int idx = tid; idx < nb * nb; idx += NTH) {
-      const int r = idx / nb, c = idx - r * nb;
+    // stage diag block
+    for (int idx = tid; idx < NB * NB; idx += NTH) {
+      const int r = idx >> 5, c = idx & 31;
       if (c <= r) pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];
     }
-    if (tid < nb) yv[tid] = xo[k + tid];
     __syncthreads();
-    if (tid == 0) {
-      for (int c = 0; c < nb; ++c) {
-        float s = yv[c];
-        for (int c2 = 0; c2 < c; ++c2) s -= pan[c * PST + c2] * yv[c2];
-        yv[c] = s / pan[c * PST + c];
+    if (tid < 64) {
+      const int r = lane & 31;
+      float rv[NB];
+      float bv = (lane < 32) ? xo[k + r] : 0.f;
+      if (lane < 32) {
+#pragma unroll
+        for (int c = 0; c < NB; ++c) rv[c] = (c <= r) ? pan[r * PST + c] : 0.f;
       }
+#pragma unroll
+      for (int c = 0; c < NB; ++c) {
+        float yc;
+        if (lane == c) bv /= rv[c];
+        yc = __shfl(bv, c, 64);
+        if (lane < 32 && r > c) bv -= rv[c] * yc;
+        if (lane == c) yv[c] = bv;
+      }
+      if (lane < 32) xo[k + r] = bv;
     }
     __syncthreads();
-    if (tid < nb) xo[k + tid] = yv[tid];
-    for (int i = k + nb + tid; i < n; i += NTH) {
+    for (int i = k + NB + tid; i < n; i += NTH) {
       float s = 0.f;
+      const float* Lr = L + (size_t)i * n + k;
 #pragma unroll
-      for (int c = 0; c < NB; ++c) s += L[(size_t)i * n + k + c] * yv[c];
+      for (int c = 0; c < NB; ++c) s += Lr[c] * yv[c];
       xo[i] -= s;
     }
     __syncthreads();
   }
-  // ---- backward substitution: L^T x = y ----
+  // ---- backward substitution: L^T x = y, reading LT rows (coalesced) ----
   for (int k = ((n - 1) / NB) * NB; k >= 0; k -= NB) {
-    const int nb = NB;
-    for (int idx = tid; idx < nb * nb; idx += NTH) {
-      const int r = idx / nb, c = idx - r * nb;
-      if (c <= r) pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];
+    // stage diag block of LT (upper-tri rows: LT[r][c]=L[c][r], c >= r)
+    for (int idx = tid; idx < NB * NB; idx += NTH) {
+      const int r = idx >> 5, c = idx & 31;
+      if (c >= r) pan[r * PST + c] = LT[(size_t)(k + r) * n + k + c];
     }
-    if (tid < nb) yv[tid] = xo[k + tid];
     __syncthreads();
-    if (tid == 0) {
-      for (int c = nb - 1; c >= 0; --c) {
-        float s = yv[c];
-        for (int c2 = c + 1; c2 < nb; ++c2) s -= pan[c2 * PST + c] * yv[c2];
-        yv[c] = s / pan[c * PST + c];
+    if (tid < 64) {
+      const int r = lane & 31;
+      float rv[NB];
+      float bv = (lane < 32) ? xo[k + r] : 0.f;
+      if (lane < 32) {
+#pragma unroll
+        for (int c = 0; c < NB; ++c) rv[c] = (c >= r) ? pan[r * PST + c] : 0.f;
       }
+#pragma unroll
+      for (int ci = 0; ci < NB; ++ci) {
+        const int c = NB - 1 - ci;
+        float xc;
+        if (lane == c) bv /= rv[c];
+        xc = __shfl(bv, c, 64);
+        if (lane < 32 && r < c) bv -= rv[c] * xc;
+        if (lane == c) yv[c] = bv;
+      }
+      if (lane < 32) xo[k + r] = bv;
     }
     __syncthreads();
-    if (tid < nb) xo[k + tid] = yv[tid];
-    __syncthreads();
-    // update rows above: xo[i] -= L[k+c, i] * x[k+c] for i < k
+    // update rows above: xo[i] -= LT[i, k..k+NB) . x[k..k+NB)  for i < k
     for (int i = tid; i < k; i += NTH) {
       float s = 0.f;
-#pragma unroll 8
-      for (int c = 0; c < nb; ++c) s += L[(size_t)(k + c) * n + i] * yv[c];
+      const float* Lr = LT + (size_t)i * n + k;
+#pragma unroll
+      for (int c = 0; c < NB; ++c) s += Lr[c] * yv[c];
       xo[i] -= s;
     }
     __syncthreads();
   }
   if (bad) {
-    // poison the step so the LM accept mask rejects it (no host sync)
     const float qn = __int_as_float(0x7fc00000);
     for (int idx = tid; idx < n; idx += NTH) xo[idx] = qn;
     if (tid == 0) atomicOr(&info[bid], 1);

@@ -217,8 +217,9 @@ def chol_solve_damped(JtJ, Jtr, mu):
         JtJ, Jtr = J2, b2
     key = (tuple(JtJ.shape), str(JtJ.device))
     sc = _chol_scratch.get(key)
-    if sc is None or sc.shape != JtJ.shape:
-        sc = torch.empty_like(JtJ)
+    want = (JtJ.shape[0], 2 * JtJ.shape[1] * JtJ.shape[2])
+    if sc is None or tuple(sc.shape) != want:
+        sc = torch.empty(want, dtype=JtJ.dtype, device=JtJ.device)
         _chol_scratch[key] = sc
     dp, info = _ext().chol_solve(JtJ.contiguous(), Jtr.contiguous(),
                                  mu.to(torch.float32).contiguous(), sc, 3)
