@@ -71,13 +71,63 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
   for (int k = 0; k < n; k += NB) {
     const int rows = n - k;
     if (stages == 1 && k > 0) break;
-    // stage panel rows k..n, cols k..k+NB into LDS (float4)
+    // stage S = damped A panel rows k..n, cols k..k+NB into LDS (float4)
     for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
       const int r = idx >> 3, c4 = (idx & 7) << 2;
       *(float4*)(pan + r * PST + c4) =
           *(const float4*)(L + (size_t)(k + r) * n + k + c4);
     }
     __syncthreads();
+    // ---- LEFT-LOOKING update: S -= Lp(rows x k) * Lc(32 x k)^T
+    // K-chunks of 32 columns; Lc chunk staged in LDS; no trailing RMW.
+    if (k > 0) {
+      const int ntI = (rows + 3) >> 2;    // 4-row tiles
+      const int ntJ = NB / 8;             // 4 col-tiles of 8
+      for (int j0 = 0; j0 < k; j0 += NB) {
+        // stage Lc chunk: rows k..k+NB, cols j0..j0+NB -> lc LDS
+        float* lc = pan + (rows + 1) * PST;   // after S (pan area)
+        for (int idx = tid; idx < NB * (NB / 4); idx += NTH) {
+          const int r = idx >> 3, c4 = (idx & 7) << 2;
+          *(float4*)(lc + r * PST + c4) =
+              *(const float4*)(L + (size_t)(k + r) * n + j0 + c4);
+        }
+        __syncthreads();
+        for (int tile = tid; tile < ntI * ntJ; tile += NTH) {
+          const int I = tile >> 2, Jt = tile & 3;
+          float acc[4][8] = {};
+          const float* Lp0 = L + (size_t)(k + I * 4) * n + j0;
+          const float* pj = lc + (Jt * 8) * PST;
+#pragma unroll
+          for (int c = 0; c < NB; c += 4) {
+            float4 av[4], bv[8];
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+              av[i] = (k + I * 4 + i < n)
+                  ? *(const float4*)(Lp0 + (size_t)i * n + c)
+                  : make_float4(0.f, 0.f, 0.f, 0.f);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              bv[j] = *(const float4*)(pj + j * PST + c);
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+              for (int j = 0; j < 8; ++j)
+                acc[i][j] += av[i].x * bv[j].x + av[i].y * bv[j].y
+                           + av[i].z * bv[j].z + av[i].w * bv[j].w;
+          }
+#pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            const int r = I * 4 + i;
+            if (r < rows) {
+#pragma unroll
+              for (int j = 0; j < 8; ++j)
+                pan[r * PST + Jt * 8 + j] -= acc[i][j];
+            }
+          }
+        }
+        __syncthreads();
+      }
+    }
     // ---- wave-synchronous 32x32 factor on wave 0
     if (tid < 64) {
       const int r = lane & 31;
@@ -125,68 +175,16 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     }
     __syncthreads();
     // write panel back: L (row-major, float4) and the L^T mirror
+    // (LT writes coalesced: r varies fastest within a column)
     for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
       const int r = idx >> 3, c4 = (idx & 7) << 2;
       *(float4*)(L + (size_t)(k + r) * n + k + c4) =
           *(const float4*)(pan + r * PST + c4);
     }
-    for (int idx = tid; idx < rows * NB; idx += NTH) {
-      const int r = idx >> 5, c = idx & 31;
-      LT[(size_t)(k + c) * n + k + r] = pan[r * PST + c];
-    }
-    // ---- trailing SYRK: 4x8 register tiles (rows x cols), float4 K
-    // row-tiles of 4, col-tiles of 8: tile grid is ntI x ntJ lower band
-    const int rows2 = rows - NB;
-    if (rows2 > 0) {
-      const int ntI = (rows2 + 3) >> 2;   // 4-row tiles
-      const int ntJ = (rows2 + 7) >> 3;   // 8-col tiles
-      const int ntiles = ntI * ntJ;       // rectangular; skip above-diag
-      for (int tile = tid; tile < ntiles; tile += NTH) {
-        const int I = tile / ntJ;
-        const int Jt = tile - I * ntJ;
-        const int gi0 = k + NB + I * 4, gj0 = k + NB + Jt * 8;
-        if (gj0 > gi0 + 3) continue;      // tile entirely above diagonal
-        const float* pi = pan + (NB + I * 4) * PST;
-        const float* pj = pan + (NB + Jt * 8) * PST;
-        float acc[4][8] = {};
 #pragma unroll
-        for (int c = 0; c < NB; c += 4) {
-          float4 av[4], bv[8];
-#pragma unroll
-          for (int i = 0; i < 4; ++i) av[i] = *(const float4*)(pi + i * PST + c);
-#pragma unroll
-          for (int j = 0; j < 8; ++j) bv[j] = *(const float4*)(pj + j * PST + c);
-#pragma unroll
-          for (int i = 0; i < 4; ++i)
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              acc[i][j] += av[i].x * bv[j].x + av[i].y * bv[j].y
-                         + av[i].z * bv[j].z + av[i].w * bv[j].w;
-        }
-        if (gj0 + 7 <= gi0 && gi0 + 4 <= n) {
-          // interior tile fully below the diagonal -> vectorized RMW
-#pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            float* Lr = L + (size_t)(gi0 + i) * n + gj0;
-            float4 v0 = *(float4*)(Lr), v1 = *(float4*)(Lr + 4);
-            v0.x -= acc[i][0]; v0.y -= acc[i][1];
-            v0.z -= acc[i][2]; v0.w -= acc[i][3];
-            v1.x -= acc[i][4]; v1.y -= acc[i][5];
-            v1.z -= acc[i][6]; v1.w -= acc[i][7];
-            *(float4*)(Lr) = v0; *(float4*)(Lr + 4) = v1;
-          }
-        } else {
-#pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            const int gi = gi0 + i;
-            if (gi >= n) break;
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              const int gj = gj0 + j;
-              if (gj <= gi && gj < n) L[(size_t)gi * n + gj] -= acc[i][j];
-            }
-          }
-        }
+    for (int c = 0; c < NB; ++c) {
+      for (int r = tid; r < rows; r += NTH) {
+        LT[(size_t)(k + c) * n + k + r] = pan[r * PST + c];
       }
     }
     __syncthreads();
