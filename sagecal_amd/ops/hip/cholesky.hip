@@ -79,54 +79,51 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     }
     __syncthreads();
     // ---- LEFT-LOOKING update: S -= Lp(rows x k) * Lc(32 x k)^T
-    // K-chunks of 32 columns; Lc chunk staged in LDS; no trailing RMW.
+    // work flattened over (4-row tile, 8-col tile, 32-K chunk): every
+    // thread busy even on late panels; partials accumulate into the LDS
+    // panel with atomics; operands read from L (L2-resident).
     if (k > 0) {
-      const int ntI = (rows + 3) >> 2;    // 4-row tiles
-      const int ntJ = NB / 8;             // 4 col-tiles of 8
-      for (int j0 = 0; j0 < k; j0 += NB) {
-        // stage Lc chunk: rows k..k+NB, cols j0..j0+NB -> lc LDS
-        float* lc = pan + (rows + 1) * PST;   // after S (pan area)
-        for (int idx = tid; idx < NB * (NB / 4); idx += NTH) {
-          const int r = idx >> 3, c4 = (idx & 7) << 2;
-          *(float4*)(lc + r * PST + c4) =
-              *(const float4*)(L + (size_t)(k + r) * n + j0 + c4);
-        }
-        __syncthreads();
-        for (int tile = tid; tile < ntI * ntJ; tile += NTH) {
-          const int I = tile >> 2, Jt = tile & 3;
-          float acc[4][8] = {};
-          const float* Lp0 = L + (size_t)(k + I * 4) * n + j0;
-          const float* pj = lc + (Jt * 8) * PST;
+      const int ntI = (rows + 3) >> 2;
+      const int KC = k >> 5;                // k is a multiple of 32
+      const int nwork = ntI * 4 * KC;
+      for (int wi = tid; wi < nwork; wi += NTH) {
+        int tmp = wi;
+        const int kc = tmp % KC; tmp /= KC;
+        const int Jt = tmp & 3;
+        const int I = tmp >> 2;
+        const int j0 = kc << 5;
+        float acc[4][8] = {};
+        const float* Lp0 = L + (size_t)(k + I * 4) * n + j0;
+        const float* Lq0 = L + (size_t)k * n + (size_t)(Jt * 8) * n + j0;
 #pragma unroll
-          for (int c = 0; c < NB; c += 4) {
-            float4 av[4], bv[8];
+        for (int c = 0; c < NB; c += 4) {
+          float4 av[4], bv[8];
 #pragma unroll
-            for (int i = 0; i < 4; ++i)
-              av[i] = (k + I * 4 + i < n)
-                  ? *(const float4*)(Lp0 + (size_t)i * n + c)
-                  : make_float4(0.f, 0.f, 0.f, 0.f);
+          for (int i = 0; i < 4; ++i)
+            av[i] = (I * 4 + i < rows)
+                ? *(const float4*)(Lp0 + (size_t)i * n + c)
+                : make_float4(0.f, 0.f, 0.f, 0.f);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            bv[j] = *(const float4*)(Lq0 + (size_t)j * n + c);
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-              bv[j] = *(const float4*)(pj + j * PST + c);
+              acc[i][j] += av[i].x * bv[j].x + av[i].y * bv[j].y
+                         + av[i].z * bv[j].z + av[i].w * bv[j].w;
+        }
 #pragma unroll
-            for (int i = 0; i < 4; ++i)
+        for (int i = 0; i < 4; ++i) {
+          const int r = I * 4 + i;
+          if (r < rows) {
 #pragma unroll
-              for (int j = 0; j < 8; ++j)
-                acc[i][j] += av[i].x * bv[j].x + av[i].y * bv[j].y
-                           + av[i].z * bv[j].z + av[i].w * bv[j].w;
-          }
-#pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            const int r = I * 4 + i;
-            if (r < rows) {
-#pragma unroll
-              for (int j = 0; j < 8; ++j)
-                pan[r * PST + Jt * 8 + j] -= acc[i][j];
-            }
+            for (int j = 0; j < 8; ++j)
+              atomicAdd(&pan[r * PST + Jt * 8 + j], -acc[i][j]);
           }
         }
-        __syncthreads();
       }
+      __syncthreads();
     }
     // ---- wave-synchronous 32x32 factor on wave 0
     if (tid < 64) {
