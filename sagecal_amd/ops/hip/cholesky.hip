@@ -79,39 +79,42 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     }
     __syncthreads();
     // ---- LEFT-LOOKING update: S -= Lp(rows x k) * Lc(32 x k)^T
-    // work flattened over (4-row tile, 8-col tile, 32-K chunk): every
-    // thread busy even on late panels; partials accumulate into the LDS
-    // panel with atomics; operands read from L (L2-resident).
+    // Each (4-row x 8-col) tile owns its FULL K accumulation in registers
+    // (no atomics, no barriers); operands stream from L via L2.
     if (k > 0) {
       const int ntI = (rows + 3) >> 2;
-      const int KC = k >> 5;                // k is a multiple of 32
-      const int nwork = ntI * 4 * KC;
-      for (int wi = tid; wi < nwork; wi += NTH) {
-        int tmp = wi;
-        const int kc = tmp % KC; tmp /= KC;
-        const int Jt = tmp & 3;
-        const int I = tmp >> 2;
-        const int j0 = kc << 5;
+      for (int tile = tid; tile < ntI * 4; tile += NTH) {
+        const int Jt = tile & 3;
+        const int I = tile >> 2;
         float acc[4][8] = {};
-        const float* Lp0 = L + (size_t)(k + I * 4) * n + j0;
-        const float* Lq0 = L + (size_t)k * n + (size_t)(Jt * 8) * n + j0;
+        const float* Lp0 = L + (size_t)(k + I * 4) * n;
+        const float* Lq0 = L + (size_t)(k + Jt * 8) * n;
+        const bool full = (I * 4 + 3) < rows;
+        for (int j0 = 0; j0 < k; j0 += NB) {
 #pragma unroll
-        for (int c = 0; c < NB; c += 4) {
-          float4 av[4], bv[8];
+          for (int c = 0; c < NB; c += 4) {
+            float4 av[4], bv[8];
+            if (full) {
 #pragma unroll
-          for (int i = 0; i < 4; ++i)
-            av[i] = (I * 4 + i < rows)
-                ? *(const float4*)(Lp0 + (size_t)i * n + c)
-                : make_float4(0.f, 0.f, 0.f, 0.f);
+              for (int i = 0; i < 4; ++i)
+                av[i] = *(const float4*)(Lp0 + (size_t)i * n + j0 + c);
+            } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            bv[j] = *(const float4*)(Lq0 + (size_t)j * n + c);
-#pragma unroll
-          for (int i = 0; i < 4; ++i)
+              for (int i = 0; i < 4; ++i)
+                av[i] = (I * 4 + i < rows)
+                    ? *(const float4*)(Lp0 + (size_t)i * n + j0 + c)
+                    : make_float4(0.f, 0.f, 0.f, 0.f);
+            }
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-              acc[i][j] += av[i].x * bv[j].x + av[i].y * bv[j].y
-                         + av[i].z * bv[j].z + av[i].w * bv[j].w;
+              bv[j] = *(const float4*)(Lq0 + (size_t)j * n + j0 + c);
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+              for (int j = 0; j < 8; ++j)
+                acc[i][j] += av[i].x * bv[j].x + av[i].y * bv[j].y
+                           + av[i].z * bv[j].z + av[i].w * bv[j].w;
+          }
         }
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
@@ -119,7 +122,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
           if (r < rows) {
 #pragma unroll
             for (int j = 0; j < 8; ++j)
-              atomicAdd(&pan[r * PST + Jt * 8 + j], -acc[i][j]);
+              pan[r * PST + Jt * 8 + j] -= acc[i][j];
           }
         }
       }
