@@ -81,7 +81,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     // ---- LEFT-LOOKING update: S -= Lp(rows x k) * Lc(32 x k)^T
     // Each (4-row x 8-col) tile owns its FULL K accumulation in registers
     // (no atomics, no barriers); operands stream from L via L2.
-    if (k > 0) {
+    if (k > 0 && stages != 4) {
       const int ntI = (rows + 3) >> 2;
       for (int tile = tid; tile < ntI * 4; tile += NTH) {
         const int Jt = tile & 3;
@@ -159,7 +159,7 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     }
     __syncthreads();
     // ---- row-solve sub-panel rows NB..rows against the diag block
-    for (int r = NB + tid; r < rows; r += NTH) {
+    for (int r = NB + tid; stages != 5 && r < rows; r += NTH) {
       float rw[NB];
 #pragma unroll
       for (int c = 0; c < NB; ++c) rw[c] = pan[r * PST + c];
@@ -181,15 +181,17 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
       *(float4*)(L + (size_t)(k + r) * n + k + c4) =
           *(const float4*)(pan + r * PST + c4);
     }
+    if (stages != 6) {
 #pragma unroll
-    for (int c = 0; c < NB; ++c) {
-      for (int r = tid; r < rows; r += NTH) {
-        LT[(size_t)(k + c) * n + k + r] = pan[r * PST + c];
+      for (int c = 0; c < NB; ++c) {
+        for (int r = tid; r < rows; r += NTH) {
+          LT[(size_t)(k + c) * n + k + r] = pan[r * PST + c];
+        }
       }
     }
     __syncthreads();
   }
-  if (stages <= 2) return;
+  if (stages <= 2 || stages >= 4) return;
 
   // ---- forward substitution: L y = b (y in xo), wave-sync in-panel ----
   for (int idx = tid; idx < n; idx += NTH) xo[idx] = b[idx];

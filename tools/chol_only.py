@@ -12,7 +12,7 @@ mu = torch.full((batch,), 0.1, device=dev)
 sc = torch.empty(batch, 2*n*n, dtype=torch.float32, device=dev)
 def run(st):
     return ext.chol_solve(A, b, mu, sc, st)
-for st in (0,1,2,3):
+for st in (0,1,2,3,4,5,6):
     run(st); torch.cuda.synchronize()
     t0=time.perf_counter()
     for _ in range(20): run(st)
