@@ -79,51 +79,33 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     }
     __syncthreads();
     // ---- LEFT-LOOKING update: S -= Lp(rows x k) * Lc(32 x k)^T
-    // Each (4-row x 8-col) tile owns its FULL K accumulation in registers
-    // (no atomics, no barriers); operands stream from L via L2.
+    // MFMA f32 16x16x4 (exact f32, 157 TF class): both fragments load
+    // COALESCED from the LT mirror (A[i][k'] = LT[k'][k+I16+i],
+    // B[k'][c] = LT[k'][k+c]) — one dword per lane per operand per MFMA,
+    // ~6x fewer load-issue slots than scalar float4 tiles (the previous
+    // version was load-issue-bound on a single CU).
     if (k > 0 && stages != 4) {
-      const int ntI = (rows + 3) >> 2;
-      for (int tile = tid; tile < ntI * 4; tile += NTH) {
-        const int Jt = tile & 3;
-        const int I = tile >> 2;
-        float acc[4][8] = {};
-        const float* Lp0 = L + (size_t)(k + I * 4) * n;
-        const float* Lq0 = L + (size_t)(k + Jt * 8) * n;
-        const bool full = (I * 4 + 3) < rows;
-        for (int j0 = 0; j0 < k; j0 += NB) {
-#pragma unroll
-          for (int c = 0; c < NB; c += 4) {
-            float4 av[4], bv[8];
-            if (full) {
-#pragma unroll
-              for (int i = 0; i < 4; ++i)
-                av[i] = *(const float4*)(Lp0 + (size_t)i * n + j0 + c);
-            } else {
-#pragma unroll
-              for (int i = 0; i < 4; ++i)
-                av[i] = (I * 4 + i < rows)
-                    ? *(const float4*)(Lp0 + (size_t)i * n + j0 + c)
-                    : make_float4(0.f, 0.f, 0.f, 0.f);
-            }
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              bv[j] = *(const float4*)(Lq0 + (size_t)j * n + j0 + c);
-#pragma unroll
-            for (int i = 0; i < 4; ++i)
-#pragma unroll
-              for (int j = 0; j < 8; ++j)
-                acc[i][j] += av[i].x * bv[j].x + av[i].y * bv[j].y
-                           + av[i].z * bv[j].z + av[i].w * bv[j].w;
-          }
+      const int w = tid >> 6;              // wave id (4 waves)
+      const int l15 = lane & 15, l4 = lane >> 4;
+      const int ntI = (rows + 15) >> 4;
+      const int ntiles = ntI * 2;          // 2 col-tiles of 16
+      for (int tile = w; tile < ntiles; tile += 4) {
+        const int I = tile >> 1, Jt = tile & 1;
+        const int arow = k + I * 16 + l15;       // S row this lane loads
+        const int bcol = k + Jt * 16 + l15;
+        const bool aok = (I * 16 + l15) < rows;
+        float4 acc = {0.f, 0.f, 0.f, 0.f};
+        const float* LTb = LT + (size_t)l4 * n;  // + kk*n walks K
+        for (int kk = 0; kk < k; kk += 4) {
+          const float a = aok ? LTb[(size_t)kk * n + arow] : 0.f;
+          const float bfrag = LTb[(size_t)kk * n + bcol];
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bfrag, acc, 0, 0, 0);
         }
+        // C/D layout: col = lane&15, row = (lane>>4)*4 + reg
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          const int r = I * 4 + i;
-          if (r < rows) {
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              pan[r * PST + Jt * 8 + j] -= acc[i][j];
-          }
+        for (int reg = 0; reg < 4; ++reg) {
+          const int r = I * 16 + l4 * 4 + reg;
+          if (r < rows) pan[r * PST + Jt * 16 + l15] -= acc[reg];
         }
       }
       __syncthreads();
