@@ -94,7 +94,8 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
         const int arow = k + I * 16 + l15;       // S row this lane loads
         const int bcol = k + Jt * 16 + l15;
         const bool aok = (I * 16 + l15) < rows;
-        float4 acc = {0.f, 0.f, 0.f, 0.f};
+        using f32x4 = __attribute__((ext_vector_type(4))) float;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
         const float* LTb = LT + (size_t)l4 * n;  // + kk*n walks K
         for (int kk = 0; kk < k; kk += 4) {
           const float a = aok ? LTb[(size_t)kk * n + arow] : 0.f;
