@@ -97,7 +97,27 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
         using f32x4 = __attribute__((ext_vector_type(4))) float;
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
         const float* LTb = LT + (size_t)l4 * n;  // + kk*n walks K
-        for (int kk = 0; kk < k; kk += 4) {
+        // software-pipelined K loop: batch 4 MFMAs' loads ahead so L2
+        // latency hides under the MFMA chain
+        int kk = 0;
+        for (; kk + 16 <= k; kk += 16) {
+          float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+          if (aok) {
+            a0 = LTb[(size_t)kk * n + arow];
+            a1 = LTb[(size_t)(kk + 4) * n + arow];
+            a2 = LTb[(size_t)(kk + 8) * n + arow];
+            a3 = LTb[(size_t)(kk + 12) * n + arow];
+          }
+          const float b0 = LTb[(size_t)kk * n + bcol];
+          const float b1 = LTb[(size_t)(kk + 4) * n + bcol];
+          const float b2 = LTb[(size_t)(kk + 8) * n + bcol];
+          const float b3 = LTb[(size_t)(kk + 12) * n + bcol];
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc, 0, 0, 0);
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc, 0, 0, 0);
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a2, b2, acc, 0, 0, 0);
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a3, b3, acc, 0, 0, 0);
+        }
+        for (; kk < k; kk += 4) {
           const float a = aok ? LTb[(size_t)kk * n + arow] : 0.f;
           const float bfrag = LTb[(size_t)kk * n + bcol];
           acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bfrag, acc, 0, 0, 0);
@@ -111,34 +131,33 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
       }
       __syncthreads();
     }
-    // ---- wave-synchronous 32x32 factor on wave 0
-    if (tid < 64) {
-      const int r = lane & 31;
+    // ---- wave-synchronous 32x32 factor on wave 0: lane r holds row r;
+    // the scaled pivot column is published through LDS (yv) once per step
+    // (in-wave LDS write->read ordering; broadcast reads are 1 cycle) —
+    // ~10x fewer DS ops than a shuffle-based rank-1 update.
+    if (tid < 64 && lane < 32) {
+      const int r = lane;
       float row[NB];
-      if (lane < 32) {
 #pragma unroll
-        for (int c = 0; c < NB; ++c) row[c] = pan[r * PST + c];
-      }
+      for (int c = 0; c < NB; ++c) row[c] = pan[r * PST + c];
 #pragma unroll
       for (int c = 0; c < NB; ++c) {
-        float pv = __shfl(row[c], c, 64);
-        if (lane == c) {
+        if (r == c) {
+          float pv = row[c];
           if (pv <= 1e-30f) { bad = 1; pv = 1e-30f; }
-          pv = sqrtf(pv);
-          row[c] = pv;
+          yv[c] = sqrtf(pv);
         }
-        pv = __shfl(row[c], c, 64);
-        if (lane < 32 && r > c) row[c] /= pv;
+        const float pv = yv[c];      // in-wave LDS broadcast
+        if (r >= c) row[c] /= pv;
+        yv[r] = row[c];              // publish scaled column c
+        const float lrc = row[c];
 #pragma unroll
         for (int cc = c + 1; cc < NB; ++cc) {
-          const float lcc = __shfl(row[c], cc, 64);
-          if (lane < 32 && r >= cc) row[cc] -= row[c] * lcc;
+          if (r >= cc) row[cc] -= lrc * yv[cc];
         }
       }
-      if (lane < 32) {
 #pragma unroll
-        for (int c = 0; c < NB; ++c) pan[r * PST + c] = row[c];
-      }
+      for (int c = 0; c < NB; ++c) pan[r * PST + c] = row[c];
     }
     __syncthreads();
     // ---- row-solve sub-panel rows NB..rows against the diag block
