@@ -142,14 +142,17 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
       for (int c = 0; c < NB; ++c) row[c] = pan[r * PST + c];
 #pragma unroll
       for (int c = 0; c < NB; ++c) {
-        if (r == c) {
-          float pv = row[c];
-          if (pv <= 1e-30f) { bad = 1; pv = 1e-30f; }
-          yv[c] = sqrtf(pv);
+        // pivot broadcast via shfl (a divergent LDS write followed by a
+        // uniform read is NOT ordered: the compiler emits the else-path
+        // read before the taken-path write)
+        float pivraw = __shfl(row[c], c, 64);
+        if (pivraw <= 1e-30f) {
+          if (lane == 0) bad = 1;
+          pivraw = 1e-30f;
         }
-        const float pv = yv[c];      // in-wave LDS broadcast
+        const float pv = sqrtf(pivraw);
         if (r >= c) row[c] /= pv;
-        yv[r] = row[c];              // publish scaled column c
+        yv[r] = row[c];              // non-divergent publish (one ds_write)
         const float lrc = row[c];
 #pragma unroll
         for (int cc = c + 1; cc < NB; ++cc) {
