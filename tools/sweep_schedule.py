@@ -1,0 +1,47 @@
+"""Sweep solver schedules on the bench config: wall time vs residual
+quality, to pick the production schedule (fixed quality target, maximum
+throughput)."""
+import sys, os, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+
+
+def main():
+    dev = 'cuda:0'
+    import bench
+    from sagecal_amd.solvers import sage
+    from sagecal_amd.constants import SM_RTR_OSRLM_RLBFGS
+
+    class A: pass
+    a = A(); a.__dict__.update(stations=64, dirs=10, srcs=5, tilesz=60,
+                               chan=8, freq0=150e6, bandwidth=180e3)
+    pack, ms, tile, bb = bench.build_problem(a, dev, torch.float32)
+
+    configs = [
+        # (emiter, maxiter, robust_outer, em_group, label)
+        (3, 12, 2, 2, 'current'),
+        (3, 8, 1, 2, 'lean'),
+        (2, 10, 1, 2, 'leaner'),
+        (2, 8, 1, 2, 'min'),
+        (3, 12, 2, 3, 'g3'),
+        (4, 8, 1, 2, 'more-em-lean'),
+        (3, 8, 1, 3, 'lean-g3'),
+        (2, 6, 1, 2, 'tiny'),
+    ]
+    for emiter, maxiter, ro, eg, label in configs:
+        opts = sage.SageSolveOptions(
+            max_emiter=emiter, max_iter=maxiter,
+            solver_mode=SM_RTR_OSRLM_RLBFGS, robust_outer=ro, em_group=eg)
+        state = sage.CalState(pack, 64, device=dev, dtype=torch.complex64)
+        cohs = sage.precalc_coherencies(pack, tile).to(torch.complex64)
+        sage.sagefit(state, cohs, tile, bb, opts)   # warm
+        state.reset(); state.nu.fill_(2.0)
+        torch.cuda.synchronize(); t0 = time.perf_counter()
+        res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
+        torch.cuda.synchronize(); t1 = time.perf_counter()
+        print(f"{label:14s} em={emiter} it={maxiter} ro={ro} g={eg}: "
+              f"{1e3*(t1-t0):7.1f} ms  res {res0:.3f}->{res1:.4f}")
+
+
+if __name__ == '__main__':
+    main()
