@@ -79,9 +79,10 @@ def main():
     ap.add_argument('--freq0', type=float, default=150e6)
     ap.add_argument('--bandwidth', type=float, default=180e3)
     ap.add_argument('--emiter', type=int, default=3)
-    ap.add_argument('--maxiter', type=int, default=12)
+    ap.add_argument('--maxiter', type=int, default=8)
+    ap.add_argument('--robust-outer', type=int, default=1)
     ap.add_argument('--joint', type=int, default=0)
-    ap.add_argument('--em-group', type=int, default=2)
+    ap.add_argument('--em-group', type=int, default=3)
     ap.add_argument('--cpu', action='store_true')
     args = ap.parse_args()
 
@@ -111,7 +112,8 @@ def main():
     opts = sage.SageSolveOptions(
         max_emiter=args.emiter, max_iter=args.maxiter,
         solver_mode=SM_RTR_OSRLM_RLBFGS,  # robust default (-j 5 analog)
-        robust_outer=2, em_group=args.em_group, joint_iters=args.joint)
+        robust_outer=args.robust_outer, em_group=args.em_group,
+        joint_iters=args.joint)
 
     def sync():
         if use_gpu:
