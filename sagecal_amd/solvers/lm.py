@@ -13,9 +13,15 @@ Ordered-subsets (OS-LM, oslmfit.c) acceleration: each outer pass runs LM on
 a random contiguous fraction of the baselines then a final full pass.
 """
 import math
+import os
 import torch
 
 from ..ops import dispatch as ops
+
+# hipGraph cache for the fixed-iteration LM body: keyed by problem shape.
+# Capturing the whole inner loop removes all per-iteration host dispatch
+# (the loop is host-sync-free by construction).
+_lm_graph_cache = {}
 
 
 class LMProblem:
@@ -36,6 +42,14 @@ class LMProblem:
 
 def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
              verbose=False):
+    if (prob.layout is not None and prob.x.is_cuda
+            and os.environ.get('SAGECAL_NO_GRAPH') != '1'):
+        return _lm_solve_graphed(prob, J0, maxiter, tau, eps1, eps2)
+    return _lm_solve_eager(prob, J0, maxiter, tau, eps1, eps2)
+
+
+def _lm_solve_eager(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
+                    verbose=False):
     """Batched LM: J0 [nchunk, N, 2, 2] complex initial Jones.
 
     Returns (J, info dict). All chunks iterate in lockstep with per-chunk
@@ -224,3 +238,94 @@ def joint_lm_solve(x, cohs, J_packed, chunk_off, nchunks, bb, T, Nbase,
             if nu > 1e12:
                 break
     return J, cost
+
+
+def _lm_body(x, coh, bb, N, nchunk, chunk_rows, weights, layout, J0,
+             maxiter, tau):
+    """The capture-safe fixed-iteration LM loop (no host reads)."""
+    dev = x.device
+    rdt = x.real.dtype
+    from ..ops.hip_host import chol_solve_damped
+    J = J0.clone()
+    nu = torch.full((nchunk,), 2.0, dtype=rdt, device=dev)
+    JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, weights, chunk_rows,
+                              nchunk, layout)
+    cost = ops.model_cost_per_chunk(x, coh, J, bb, N, weights, chunk_rows,
+                                    nchunk, layout)
+    init_cost = cost.clone()
+    diag_max = JtJ.diagonal(dim1=-2, dim2=-1).max(dim=-1).values
+    mu = tau * diag_max
+    for it in range(maxiter):
+        dp = chol_solve_damped(JtJ, Jtr, mu)
+        dp = torch.nan_to_num(dp, nan=0.0, posinf=0.0, neginf=0.0)
+        Jnew = J + _vec_to_jones(dp, nchunk, N)
+        cost_new = ops.model_cost_per_chunk(x, coh, Jnew, bb, N, weights,
+                                            chunk_rows, nchunk, layout)
+        denom = (dp * (mu[:, None] * dp + Jtr)).sum(dim=-1).clamp_min(1e-30)
+        rho = (cost - cost_new) / denom
+        accept = rho > 0
+        J = torch.where(accept[:, None, None, None], Jnew, J)
+        cost = torch.where(accept, cost_new, cost)
+        fac = (1.0 - (2.0 * rho - 1.0) ** 3).clamp_min(1.0 / 3.0)
+        mu = torch.where(accept, mu * fac, mu * nu)
+        nu = torch.where(accept, torch.full_like(nu, 2.0), nu * 2.0)
+        if it + 1 < maxiter:
+            JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, weights, chunk_rows,
+                                      nchunk, layout)
+    return J, cost, init_cost
+
+
+def _lm_solve_graphed(prob, J0, maxiter, tau, eps1, eps2):
+    """hipGraph-captured LM: one replay per solve, zero per-iteration host
+    dispatch. Static input buffers per (B, nchunk, maxiter, layout) shape."""
+    x, coh, bb, N = prob.x, prob.coh, prob.bb, prob.N
+    nchunk = prob.nchunk
+    dev = x.device
+    B = x.shape[0]
+    key = (B, nchunk, N, maxiter, id(prob.layout), float(tau))
+    ent = _lm_graph_cache.get(key)
+    if ent is None:
+        ent = {}
+        ent['x'] = torch.empty_like(x)
+        ent['coh'] = torch.empty_like(coh)
+        ent['J0'] = torch.empty_like(J0)
+        ent['w'] = torch.ones(B, dtype=torch.float32, device=dev)
+        ent['rows'] = torch.zeros(B, dtype=torch.long, device=dev)
+        ent['bb'] = bb
+        # warmup on a side stream (allocator + kernels)
+        ent['x'].copy_(x); ent['coh'].copy_(coh); ent['J0'].copy_(J0)
+        if prob.weights is not None:
+            ent['w'].copy_(prob.weights.to(torch.float32))
+        if prob.chunk_rows is not None:
+            ent['rows'].copy_(prob.chunk_rows)
+        st = torch.cuda.Stream()
+        st.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(st):
+            for _ in range(2):
+                _lm_body(ent['x'], ent['coh'], ent['bb'], N, nchunk,
+                         ent['rows'], ent['w'], prob.layout, ent['J0'],
+                         maxiter, tau)
+        torch.cuda.current_stream().wait_stream(st)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            outs = _lm_body(ent['x'], ent['coh'], ent['bb'], N, nchunk,
+                            ent['rows'], ent['w'], prob.layout, ent['J0'],
+                            maxiter, tau)
+        ent['graph'] = g
+        ent['outs'] = outs
+        _lm_graph_cache[key] = ent
+    ent['x'].copy_(x)
+    ent['coh'].copy_(coh)
+    ent['J0'].copy_(J0)
+    if prob.weights is not None:
+        ent['w'].copy_(prob.weights.to(torch.float32))
+    else:
+        ent['w'].fill_(1.0)
+    if prob.chunk_rows is not None:
+        ent['rows'].copy_(prob.chunk_rows)
+    else:
+        ent['rows'].zero_()
+    ent['graph'].replay()
+    J, cost, init_cost = ent['outs']
+    return J.clone(), {'init_cost': init_cost.clone(),
+                       'final_cost': cost.clone(), 'niter': maxiter}
