@@ -211,7 +211,7 @@ def chol_solve_damped(JtJ, Jtr, mu):
         J2 = torch.zeros(batch, npad, npad, dtype=JtJ.dtype,
                          device=JtJ.device)
         J2[:, :n, :n] = JtJ
-        J2[:, range(n, npad), range(n, npad)] = 1.0
+        J2.diagonal(dim1=1, dim2=2)[:, n:].fill_(1.0)
         b2 = torch.zeros(batch, npad, dtype=Jtr.dtype, device=Jtr.device)
         b2[:, :n] = Jtr
         JtJ, Jtr = J2, b2
