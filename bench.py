@@ -53,15 +53,28 @@ def build_problem(args, device, dtype, rank=0, world=1):
     return pack, ms, tile, bb
 
 
-def run_step(state, pack, tile, bb, opts, args, device):
-    """One full tile calibration (the benchmark unit of work)."""
+def run_step(state, pack, tile, bb, opts, args, device, adm=None):
+    """One full tile calibration (the benchmark unit of work).
+
+    With consensus (adm): the sagecal-mpi per-tile ADMM loop — n_admm
+    outer iterations, each one EM sweep + fused Z all-reduce over
+    RCCL/xGMI; world=1 runs the identical loop without communication, so
+    the scaling series is apples-to-apples (weak scaling over sub-bands).
+    """
     from sagecal_amd.solvers import sage
     state.reset()
     state.nu.fill_(2.0)
     cohs = sage.precalc_coherencies(pack, tile)
     if device != 'cpu':
         cohs = cohs.to(torch.complex64)
-    res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
+    if adm is not None:
+        adm.Y.zero_()
+        adm.Z.zero_()
+        adm.Yhat_prev = None
+        adm.J_prev = None
+        res0, res1 = adm.run(cohs, tile, bb, opts, n_admm=args.emiter)
+    else:
+        res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
     xres = sage.calculate_residuals_multifreq(state, pack, tile, bb)
     return res0, res1, xres
 
@@ -83,6 +96,9 @@ def main():
     ap.add_argument('--robust-outer', type=int, default=1)
     ap.add_argument('--joint', type=int, default=0)
     ap.add_argument('--em-group', type=int, default=3)
+    ap.add_argument('--npoly', type=int, default=2)
+    ap.add_argument('--admm-rho', type=float, default=5.0)
+    ap.add_argument('--no-consensus', action='store_true')
     ap.add_argument('--cpu', action='store_true')
     args = ap.parse_args()
 
@@ -109,8 +125,16 @@ def main():
     pack, ms, tile, bb = build_problem(args, device, dtype, rank, world)
     cdtype = torch.complex64 if dtype == torch.float32 else torch.complex128
     state = sage.CalState(pack, args.stations, device=device, dtype=cdtype)
+    adm = None
+    if not args.no_consensus:
+        from sagecal_amd.consensus.admm import ConsensusADMM
+        freqs_all = [args.freq0 + r * args.bandwidth for r in range(world)]
+        adm = ConsensusADMM(state, freqs_all, args.freq0, rank, world,
+                            Npoly=min(args.npoly, world),
+                            rho=torch.full((pack.M,), args.admm_rho))
     opts = sage.SageSolveOptions(
-        max_emiter=args.emiter, max_iter=args.maxiter,
+        max_emiter=1 if adm is not None else args.emiter,
+        max_iter=args.maxiter,
         solver_mode=SM_RTR_OSRLM_RLBFGS,  # robust default (-j 5 analog)
         robust_outer=args.robust_outer, em_group=args.em_group,
         joint_iters=args.joint)
@@ -123,12 +147,12 @@ def main():
 
     # warmup
     for _ in range(args.warmup):
-        run_step(state, pack, tile, bb, opts, args, device)
+        run_step(state, pack, tile, bb, opts, args, device, adm)
     sync()
     t0 = time.perf_counter()
     res = None
     for _ in range(args.steps):
-        res = run_step(state, pack, tile, bb, opts, args, device)
+        res = run_step(state, pack, tile, bb, opts, args, device, adm)
     sync()
     t1 = time.perf_counter()
     elapsed = t1 - t0
@@ -163,7 +187,7 @@ def main():
                 'tilesz': args.tilesz, 'channels': args.chan,
                 'global_batch': vis_per_step,
                 'seq_len': args.tilesz,
-                'parallelism': f'freq-band dp{world}',
+                'parallelism': f'consensus-admm freq-band dp{world}',
                 'res0': res[0] if res else None,
                 'res1': res[1] if res else None,
             },

@@ -29,7 +29,7 @@ class LMProblem:
     (or for SAGE-batched mode: all clusters x chunks concatenated)."""
 
     def __init__(self, x, coh, bb, N, nchunk=1, chunk_rows=None,
-                 weights=None, layout=None):
+                 weights=None, layout=None, admm=None):
         self.x = x              # [B,2,2] complex (data with own model added)
         self.coh = coh          # [B,2,2] complex cluster coherency
         self.bb = bb            # [B,2] long
@@ -38,6 +38,10 @@ class LMProblem:
         self.chunk_rows = chunk_rows  # [B] long or None
         self.weights = weights  # [B] float or None
         self.layout = layout    # BaselineLayout (GPU kernels) or None
+        # consensus-ADMM augmentation (rtr_solve_robust_admm.c analog for
+        # LM): admm = (rho [nchunk], Y [nchunk,N,2,2], BZ [nchunk,N,2,2]);
+        # adds  y^T(J - BZ) + rho/2 ||J - BZ||^2  to the cost.
+        self.admm = admm
 
 
 def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
@@ -72,7 +76,8 @@ def _lm_solve_eager(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
 
     JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, prob.weights,
                               prob.chunk_rows, nchunk, prob.layout)
-    cost = _per_chunk_cost(x, coh, J, bb, prob)
+    JtJ, Jtr = _apply_admm_terms(JtJ, Jtr, J, prob, N, nchunk)
+    cost = _total_cost(x, coh, J, bb, prob)
     init_cost = cost.clone()
     eye = torch.eye(8 * N, dtype=rdt, device=dev).unsqueeze(0)
     diag_max = JtJ.diagonal(dim1=-2, dim2=-1).max(dim=-1).values
@@ -91,7 +96,7 @@ def _lm_solve_eager(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
         dp = torch.nan_to_num(dp, nan=0.0, posinf=0.0, neginf=0.0)
         dpc = _vec_to_jones(dp, nchunk, N)
         Jnew = J + dpc
-        cost_new = _per_chunk_cost(x, coh, Jnew, bb, prob)
+        cost_new = _total_cost(x, coh, Jnew, bb, prob)
         denom = (dp * (mu[:, None] * dp + Jtr)).sum(dim=-1).clamp_min(1e-30)
         rho = (cost - cost_new) / denom
         accept = (rho > 0) & active
@@ -111,12 +116,45 @@ def _lm_solve_eager(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
             # recompute JtJ/Jtr at (possibly) new J
             JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, prob.weights,
                                       prob.chunk_rows, nchunk, prob.layout)
+            JtJ, Jtr = _apply_admm_terms(JtJ, Jtr, J, prob, N, nchunk)
             gnorm = Jtr.abs().max(dim=-1).values
             active = active & (gnorm > eps1)
             if (it & 7) == 7 and not bool(active.any()):
                 break
     info = {'init_cost': init_cost, 'final_cost': cost, 'niter': niter}
     return J, info
+
+
+def _apply_admm_terms(JtJ, Jtr, J, prob, N, nchunk):
+    """Add the consensus augmentation to the normal equations:
+    JtJ += rho/2 I;  Jtr -= vecR(Y/2 + rho/2 (J - BZ))."""
+    if prob.admm is None:
+        return JtJ, Jtr
+    rho, Y, BZ = prob.admm
+    rho = rho.to(JtJ.dtype)
+    JtJ = JtJ + (0.5 * rho)[:, None, None] * torch.eye(
+        8 * N, dtype=JtJ.dtype, device=JtJ.device).unsqueeze(0)
+    extra = 0.5 * Y + (0.5 * rho)[:, None, None, None] * (J - BZ)
+    from ..ops.reference import vecR
+    Jtr = Jtr - vecR(extra).reshape(nchunk, 8 * N).to(Jtr.dtype)
+    return JtJ, Jtr
+
+
+def _admm_cost(J, prob):
+    """Per-chunk  Re<Y, J-BZ> + rho/2 ||J-BZ||^2."""
+    rho, Y, BZ = prob.admm
+    d = J - BZ
+    lin = (Y.conj() * d).real.sum(dim=(-1, -2, -3))
+    quad = 0.5 * rho * (d.abs() ** 2).sum(dim=(-1, -2, -3))
+    return (lin + quad).to(torch.float32 if J.dtype == torch.complex64
+                           else torch.float64)
+
+
+def _total_cost(x, coh, J, bb, prob):
+    c = _per_chunk_cost(x, coh, J, bb, prob)
+    if prob.admm is not None:
+        c = c + _admm_cost(J, prob).to(c.dtype)
+    return c
 
 
 def _chol_solve(A, b):
@@ -241,17 +279,25 @@ def joint_lm_solve(x, cohs, J_packed, chunk_off, nchunks, bb, T, Nbase,
 
 
 def _lm_body(x, coh, bb, N, nchunk, chunk_rows, weights, layout, J0,
-             maxiter, tau):
+             maxiter, tau, prob):
     """The capture-safe fixed-iteration LM loop (no host reads)."""
     dev = x.device
     rdt = x.real.dtype
     from ..ops.hip_host import chol_solve_damped
     J = J0.clone()
     nu = torch.full((nchunk,), 2.0, dtype=rdt, device=dev)
+
+    def full_cost(Jc):
+        c = ops.model_cost_per_chunk(x, coh, Jc, bb, N, weights, chunk_rows,
+                                     nchunk, layout)
+        if prob.admm is not None:
+            c = c + _admm_cost(Jc, prob).to(c.dtype)
+        return c
+
     JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, weights, chunk_rows,
                               nchunk, layout)
-    cost = ops.model_cost_per_chunk(x, coh, J, bb, N, weights, chunk_rows,
-                                    nchunk, layout)
+    JtJ, Jtr = _apply_admm_terms(JtJ, Jtr, J, prob, N, nchunk)
+    cost = full_cost(J)
     init_cost = cost.clone()
     diag_max = JtJ.diagonal(dim1=-2, dim2=-1).max(dim=-1).values
     mu = tau * diag_max
@@ -259,8 +305,7 @@ def _lm_body(x, coh, bb, N, nchunk, chunk_rows, weights, layout, J0,
         dp = chol_solve_damped(JtJ, Jtr, mu)
         dp = torch.nan_to_num(dp, nan=0.0, posinf=0.0, neginf=0.0)
         Jnew = J + _vec_to_jones(dp, nchunk, N)
-        cost_new = ops.model_cost_per_chunk(x, coh, Jnew, bb, N, weights,
-                                            chunk_rows, nchunk, layout)
+        cost_new = full_cost(Jnew)
         denom = (dp * (mu[:, None] * dp + Jtr)).sum(dim=-1).clamp_min(1e-30)
         rho = (cost - cost_new) / denom
         accept = rho > 0
@@ -272,6 +317,7 @@ def _lm_body(x, coh, bb, N, nchunk, chunk_rows, weights, layout, J0,
         if it + 1 < maxiter:
             JtJ, Jtr, _ = ops.jtj_jtr(x, coh, J, bb, N, weights, chunk_rows,
                                       nchunk, layout)
+            JtJ, Jtr = _apply_admm_terms(JtJ, Jtr, J, prob, N, nchunk)
     return J, cost, init_cost
 
 
@@ -282,7 +328,8 @@ def _lm_solve_graphed(prob, J0, maxiter, tau, eps1, eps2):
     nchunk = prob.nchunk
     dev = x.device
     B = x.shape[0]
-    key = (B, nchunk, N, maxiter, id(prob.layout), float(tau))
+    key = (B, nchunk, N, maxiter, id(prob.layout), float(tau),
+           prob.admm is not None)
     ent = _lm_graph_cache.get(key)
     if ent is None:
         ent = {}
@@ -292,25 +339,39 @@ def _lm_solve_graphed(prob, J0, maxiter, tau, eps1, eps2):
         ent['w'] = torch.ones(B, dtype=torch.float32, device=dev)
         ent['rows'] = torch.zeros(B, dtype=torch.long, device=dev)
         ent['bb'] = bb
+
+        class _P:  # static stand-in problem carrying the admm statics
+            pass
+        sp = _P()
+        sp.admm = None
+        if prob.admm is not None:
+            rho_a, Y_a, BZ_a = prob.admm
+            ent['admm'] = (torch.empty_like(rho_a), torch.empty_like(Y_a),
+                           torch.empty_like(BZ_a))
+            sp.admm = ent['admm']
+        ent['sp'] = sp
         # warmup on a side stream (allocator + kernels)
         ent['x'].copy_(x); ent['coh'].copy_(coh); ent['J0'].copy_(J0)
         if prob.weights is not None:
             ent['w'].copy_(prob.weights.to(torch.float32))
         if prob.chunk_rows is not None:
             ent['rows'].copy_(prob.chunk_rows)
+        if prob.admm is not None:
+            for d, sr in zip(ent['admm'], prob.admm):
+                d.copy_(sr)
         st = torch.cuda.Stream()
         st.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(st):
             for _ in range(2):
                 _lm_body(ent['x'], ent['coh'], ent['bb'], N, nchunk,
                          ent['rows'], ent['w'], prob.layout, ent['J0'],
-                         maxiter, tau)
+                         maxiter, tau, sp)
         torch.cuda.current_stream().wait_stream(st)
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             outs = _lm_body(ent['x'], ent['coh'], ent['bb'], N, nchunk,
                             ent['rows'], ent['w'], prob.layout, ent['J0'],
-                            maxiter, tau)
+                            maxiter, tau, sp)
         ent['graph'] = g
         ent['outs'] = outs
         _lm_graph_cache[key] = ent
@@ -325,6 +386,9 @@ def _lm_solve_graphed(prob, J0, maxiter, tau, eps1, eps2):
         ent['rows'].copy_(prob.chunk_rows)
     else:
         ent['rows'].zero_()
+    if prob.admm is not None:
+        for d, sr in zip(ent['admm'], prob.admm):
+            d.copy_(sr)
     ent['graph'].replay()
     J, cost, init_cost = ent['outs']
     return J.clone(), {'init_cost': init_cost.clone(),

@@ -174,7 +174,8 @@ def robust_lm(prob, J0, nu0, opts):
     return J, nu
 
 
-def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts):
+def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts,
+                 admm_terms=None):
     """Solve a group of clusters as ONE batched LM problem (block-diagonal
     across clusters via the chunk axis), then update the running residual
     incrementally: res += sum(V_old - V_new) over the group."""
@@ -198,8 +199,20 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts):
     rcat = torch.cat(rows_all)
     J0 = torch.cat([state.cluster_J(ci) for ci in group])
     lay = _layout_for(state, bb, T, Nbase, len(group), xcat.device)
+    admm = None
+    if admm_terms is not None:
+        rho_m, Y, BZ = admm_terms
+        sel = torch.cat([torch.arange(state.chunk_off[ci],
+                                      state.chunk_off[ci]
+                                      + state.nchunks[ci])
+                         for ci in group]).to(Y.device)
+        rho_c = torch.cat([
+            torch.full((state.nchunks[ci],), float(rho_m[ci]))
+            for ci in group]).to(device=Y.device,
+                                 dtype=torch.float32 if Y.dtype == torch.complex64 else torch.float64)
+        admm = (rho_c, Y[sel], BZ[sel])
     prob = lm_mod.LMProblem(xcat, ccat, bbcat, state.N, nch_tot, rcat,
-                            layout=lay)
+                            layout=lay, admm=admm)
     if opts.robust:
         nus = float(torch.stack([state.nu[ci] for ci in group]).mean())
         Jn, nu_new = robust_lm(prob, J0, nus, opts)
@@ -232,12 +245,12 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts):
         # redo the group sequentially
         for ci in group:
             res = _solve_group(state, [ci], res, cohs, bb, T, Nbase, B,
-                               opts)
+                               opts, admm_terms)
         return res
     return res_new
 
 
-def sagefit(state, cohs, tile, bb, opts, flags=None):
+def sagefit(state, cohs, tile, bb, opts, flags=None, admm_terms=None):
     """The SAGE EM loop (lmfit.c:777-1053). Returns (res_0, res_1): initial
     and final residual norms (per-visibility RMS like lmfit.c res_0/res_1).
 
@@ -262,7 +275,7 @@ def sagefit(state, cohs, tile, bb, opts, flags=None):
     for em in range(opts.max_emiter):
         for group in groups:
             res = _solve_group(state, group, res, cohs, bb, T, Nbase, B,
-                               opts)
+                               opts, admm_terms)
         # divergence guard (fullbatch_mode.cpp:622-632 resets on blow-up):
         rn = resnorm(res)
         if not (rn == rn) or rn > 5.0 * res_0:
