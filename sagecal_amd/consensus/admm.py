@@ -70,12 +70,24 @@ class ConsensusADMM:
         """B_f Z -> per-chunk [Mt, N, 2, 2] (each chunk of cluster ci gets
         the same consensus target)."""
         st = self.state
+        if st.Mt == st.M:
+            return Zj.to(self.cdtype)
         out = torch.empty(st.Mt, st.N, 2, 2, dtype=self.cdtype,
                           device=self.dev)
         for ci in range(st.M):
             o = st.chunk_off[ci]
             out[o:o + st.nchunks[ci]] = Zj[ci]
         return out
+
+    def _rho_chunk(self):
+        st = self.state
+        if st.Mt == st.M:
+            return self.rho.to(device=self.dev, dtype=torch.float32
+                               if self.cdtype == torch.complex64
+                               else torch.float64)
+        return torch.cat([
+            torch.full((st.nchunks[ci],), float(self.rho[ci]))
+            for ci in range(st.M)]).to(self.dev)
 
     def _allreduce(self, t):
         if self.world > 1 and dist.is_initialized():
@@ -88,16 +100,21 @@ class ConsensusADMM:
         TAG_CONSENSUS, sagecal_master.cpp:813-877)."""
         st = self.state
         M, N, P = st.M, st.N, self.Npoly
-        acc = torch.zeros(M, P, N, 2, 2, dtype=self.cdtype, device=self.dev)
-        for ci in range(M):
-            o = st.chunk_off[ci]
-            nc = st.nchunks[ci]
-            # average chunks (consensus over time-chunks shares one Z)
-            Jc = st.J[o:o + nc].mean(dim=0)
-            Yc = self.Y[o:o + nc].mean(dim=0)
-            contrib = Yc + float(self.rho[ci]) * Jc
-            for p in range(P):
-                acc[ci, p] = float(self.Bf[p]) * contrib
+        if st.Mt == M:           # common case: one chunk per cluster
+            Jm = st.J
+            Ym = self.Y
+        else:
+            Jm = torch.stack([st.J[st.chunk_off[ci]:st.chunk_off[ci]
+                                   + st.nchunks[ci]].mean(dim=0)
+                              for ci in range(M)])
+            Ym = torch.stack([self.Y[st.chunk_off[ci]:st.chunk_off[ci]
+                                     + st.nchunks[ci]].mean(dim=0)
+                              for ci in range(M)])
+        rho_m = self.rho.to(device=self.dev).to(Jm.real.dtype)
+        contrib = Ym + rho_m[:, None, None, None].to(self.cdtype) * Jm
+        Bfd = self.Bf.to(device=self.dev, dtype=Jm.real.dtype)
+        acc = (Bfd[None, :, None, None, None].to(self.cdtype)
+               * contrib[:, None]).contiguous()
         self._allreduce(torch.view_as_real(acc))
         self.Z = poly.update_global_z(acc, self.Bii)
         return self.Z
@@ -110,9 +127,7 @@ class ConsensusADMM:
     def y_update(self, BZ):
         """Y <- Y + rho (J - BZ) (sagecal_slave.cpp:870-888)."""
         st = self.state
-        rho_chunk = torch.cat([
-            torch.full((st.nchunks[ci],), float(self.rho[ci]))
-            for ci in range(st.M)]).to(self.dev)
+        rho_chunk = self._rho_chunk()
         self.Y = self.Y + rho_chunk[:, None, None, None].to(self.cdtype) * \
             (st.J - BZ)
 
@@ -121,9 +136,7 @@ class ConsensusADMM:
         consensus_poly.c:928): Yhat = Y + rho (J - B Zold); deltas vs the
         previous iteration, averaged across bands by all-reduce."""
         st = self.state
-        rho_chunk = torch.cat([
-            torch.full((st.nchunks[ci],), float(self.rho[ci]))
-            for ci in range(st.M)]).to(self.dev)
+        rho_chunk = self._rho_chunk()
         Yhat = self.Y + rho_chunk[:, None, None, None].to(self.cdtype) * \
             (st.J - BZ_old)
         if self.Yhat_prev is not None:
