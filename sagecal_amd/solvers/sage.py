@@ -20,7 +20,9 @@ import torch
 from ..ops import dispatch as ops
 from ..ops import reference as R
 from . import lm as lm_mod
-from ..constants import ROBUST_MODES, SM_RLM_RLBFGS, NU_LOW, NU_HIGH, NU_GRID
+from ..constants import (ROBUST_MODES, SM_RLM_RLBFGS, SM_RTR_OSLM_LBFGS,
+                         SM_RTR_OSRLM_RLBFGS, SM_NSD_RLBFGS, NU_LOW,
+                         NU_HIGH, NU_GRID, LMCUT)
 
 
 class SageSolveOptions:
@@ -155,14 +157,30 @@ def _solve_cluster(state, ci, xsub, coh_ci, bb, rows, opts):
     return J
 
 
+def _inner_solve(prob, J, opts, maxiter):
+    """Dispatch one deterministic inner solve by solver mode, honoring the
+    reference's LMCUT heuristic (sagecalmain.h:24: N <= 40 -> RTR/NSD
+    replaced by LM)."""
+    mode = opts.solver_mode
+    if prob.N > LMCUT and mode in (SM_RTR_OSLM_LBFGS,
+                                   SM_RTR_OSRLM_RLBFGS):
+        from . import rtr as rtr_mod
+        return rtr_mod.rtr_solve(prob, J, maxiter=maxiter)[0]
+    if prob.N > LMCUT and mode == SM_NSD_RLBFGS:
+        from . import rtr as rtr_mod
+        return rtr_mod.nsd_solve(prob, J, maxiter=2 * maxiter)[0]
+    return lm_mod.lm_solve(prob, J, maxiter=maxiter)[0]
+
+
 def robust_lm(prob, J0, nu0, opts):
-    """IRLS Student's-t LM (robustlm.c rlevmar_der_single_*): alternate
-    weighted LM with weight + nu AECM updates."""
+    """IRLS Student's-t wrapper (robustlm.c rlevmar_der_single_* /
+    rtr_solve_robust.c): alternate the weighted inner solver with weight +
+    nu AECM updates."""
     nu = nu0 if nu0 > 0 else 2.0
     J = J0
     inner = max(3, opts.max_iter // max(1, opts.robust_outer))
     for outer in range(opts.robust_outer):
-        J, _ = lm_mod.lm_solve(prob, J, maxiter=inner)
+        J = _inner_solve(prob, J, opts, inner)
         V = ops.apply_jones(prob.coh, J, prob.bb, prob.chunk_rows)
         r = prob.x - V
         w = ops.update_weights(r, nu, p=8)
@@ -170,7 +188,7 @@ def robust_lm(prob, J0, nu0, opts):
                                 nuhigh=opts.robust_nuhigh, Nd=NU_GRID, p=8)
         prob.weights = w
     # final solve with last weights
-    J, _ = lm_mod.lm_solve(prob, J, maxiter=inner)
+    J = _inner_solve(prob, J, opts, inner)
     return J, nu
 
 
@@ -223,7 +241,7 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts,
             Jn, _ = lm_mod.os_lm_solve(prob, J0, maxiter=opts.max_iter,
                                        nsubsets=opts.nsubsets)
         else:
-            Jn, _ = lm_mod.lm_solve(prob, J0, maxiter=opts.max_iter)
+            Jn = _inner_solve(prob, J0, opts, opts.max_iter)
     Jprev = [state.cluster_J(ci).clone() for ci in group]
     off = 0
     for gi, ci in enumerate(group):
