@@ -120,7 +120,7 @@ def main():
         dtype = torch.float64
 
     from sagecal_amd.solvers import sage
-    from sagecal_amd.constants import SM_RTR_OSRLM_RLBFGS
+    from sagecal_amd.constants import SM_RLM_RLBFGS
 
     pack, ms, tile, bb = build_problem(args, device, dtype, rank, world)
     cdtype = torch.complex64 if dtype == torch.float32 else torch.complex128
@@ -135,7 +135,7 @@ def main():
     opts = sage.SageSolveOptions(
         max_emiter=1 if adm is not None else args.emiter,
         max_iter=args.maxiter,
-        solver_mode=SM_RTR_OSRLM_RLBFGS,  # robust default (-j 5 analog)
+        solver_mode=SM_RLM_RLBFGS,  # robust LM (BASELINE.json names LM)
         robust_outer=args.robust_outer, em_group=args.em_group,
         joint_iters=args.joint)
 
