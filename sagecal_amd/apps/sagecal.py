@@ -1,0 +1,221 @@
+"""sagecal CLI — single-node calibration app.
+
+Mirrors the reference CLI (/root/reference/src/MS/main.cpp:110-308: same
+single-letter flags and the three run modes of fullbatch_mode.cpp /
+minibatch_mode.cpp / minibatch_consensus_mode.cpp). MS input is the NpzMS
+container (this environment has no casacore; the backend interface in
+msdata.py keeps a python-casacore MS backend pluggable).
+
+Usage:
+  sagecal.py -d obs.npz -s sky.txt -c cluster.txt [-p sol.txt] [-t 10] ...
+  sagecal.py -d obs.npz -s sky.txt -c cluster.txt -N 2 -M 2 -w 4   (stochastic)
+  sagecal.py -d obs.npz -s sky.txt -c cluster.txt -a 1 [-p sol.txt] (simulate)
+"""
+import argparse
+import sys
+import time
+
+import numpy as np
+import torch
+
+
+def build_argparser():
+    ap = argparse.ArgumentParser(
+        prog='sagecal', description=__doc__,
+        formatter_class=argparse.RawDescriptionHelpFormatter)
+    ap.add_argument('-d', dest='ms', help='MS (npz container)')
+    ap.add_argument('-f', dest='mslist', help='text file with MS names')
+    ap.add_argument('-s', dest='sky', required=True, help='sky model file')
+    ap.add_argument('-c', dest='cluster', required=True,
+                    help='cluster file')
+    ap.add_argument('-p', dest='solfile',
+                    help='solutions file (write; read when simulating)')
+    ap.add_argument('-F', dest='format', type=int, default=0,
+                    help='sky format: 0 LSM, 1 LSM 3-order spectra')
+    ap.add_argument('-I', dest='incol', default='data')
+    ap.add_argument('-O', dest='outcol', default='residual')
+    ap.add_argument('-e', dest='max_emiter', type=int, default=3)
+    ap.add_argument('-g', dest='max_iter', type=int, default=10)
+    ap.add_argument('-l', dest='max_lbfgs', type=int, default=10)
+    ap.add_argument('-m', dest='lbfgs_m', type=int, default=7)
+    ap.add_argument('-n', dest='nthreads', type=int, default=6)
+    ap.add_argument('-t', dest='tilesz', type=int, default=10)
+    ap.add_argument('-a', dest='dosim', type=int, default=0,
+                    help='0 calibrate; 1 simulate; 2 sim+add; 3 sim+sub')
+    ap.add_argument('-z', dest='ignfile', help='ignore-cluster file')
+    ap.add_argument('-b', dest='dochan', type=int, default=0)
+    ap.add_argument('-B', dest='dobeam', type=int, default=0)
+    ap.add_argument('-E', dest='gpupredict', type=int, default=None,
+                    help='1: use GPU (default: auto)')
+    ap.add_argument('-x', dest='min_uvcut', type=float, default=0.0)
+    ap.add_argument('-y', dest='max_uvcut', type=float, default=1e9)
+    ap.add_argument('-k', dest='ccid', type=int, default=-99999,
+                    help='correct residuals with this cluster id')
+    ap.add_argument('-o', dest='rho_corr', type=float, default=1e-9)
+    ap.add_argument('-j', dest='solver_mode', type=int, default=5,
+                    help='0 OSLM,1 LM,2 OSRLM,3 RLM,4 RTR,5 RRTR,6 NSD')
+    ap.add_argument('-L', dest='nulow', type=float, default=2.0)
+    ap.add_argument('-H', dest='nuhigh', type=float, default=30.0)
+    ap.add_argument('-q', dest='initsol', help='initial solutions file')
+    ap.add_argument('-N', dest='epochs', type=int, default=0,
+                    help='>0: stochastic calibration epochs')
+    ap.add_argument('-M', dest='minibatches', type=int, default=1)
+    ap.add_argument('-w', dest='minibands', type=int, default=1)
+    ap.add_argument('-A', dest='nadmm', type=int, default=1)
+    ap.add_argument('-P', dest='npoly', type=int, default=2)
+    ap.add_argument('-Q', dest='polytype', type=int, default=0)
+    ap.add_argument('-r', dest='admm_rho', type=float, default=5.0)
+    ap.add_argument('-V', dest='verbose', action='store_true')
+    return ap
+
+
+def load_context(args):
+    from .. import sky as skymod, msdata, solutions
+    from ..ops.reference import SourcePack
+    use_gpu = torch.cuda.is_available() if args.gpupredict is None \
+        else bool(args.gpupredict)
+    device = 'cuda:0' if use_gpu else 'cpu'
+    dtype = torch.float32 if use_gpu else torch.float64
+    ms = msdata.NpzMS(args.ms, tilesz=args.tilesz, device=device,
+                      dtype=dtype)
+    ignore = skymod.read_ignore_file(args.ignfile) if args.ignfile else ()
+    clusters = skymod.read_sky_cluster(args.sky, args.cluster, ms.ra0,
+                                       ms.dec0, ms.freq0, fmt=args.format,
+                                       ignore_ids=ignore)
+    pack = SourcePack(clusters)
+    return ms, pack, clusters, device, dtype
+
+
+def uv_flags(tile, args):
+    """Baseline uv-cut flags (predict.c flag=2 semantics for -x/-y)."""
+    uvlen = torch.sqrt(tile.u ** 2 + tile.v ** 2) * tile.freq0
+    return tile.flags | (uvlen < args.min_uvcut) | (uvlen > args.max_uvcut)
+
+
+def run_calibration(args):
+    from ..solvers import sage
+    from .. import solutions
+    ms, pack, clusters, device, dtype = load_context(args)
+    cdtype = torch.complex64 if dtype == torch.float32 else torch.complex128
+    state = sage.CalState(pack, ms.N, device=device, dtype=cdtype)
+    if args.initsol:
+        hdr, tiles = solutions.read_solutions(args.initsol)
+        if tiles:
+            state.J = solutions.reorder_read_tile(
+                tiles[0], state.nchunks).to(device=device, dtype=cdtype)
+    opts = sage.SageSolveOptions(
+        max_emiter=args.max_emiter, max_iter=args.max_iter,
+        solver_mode=args.solver_mode, robust_nulow=args.nulow,
+        robust_nuhigh=args.nuhigh, lbfgs_iters=args.max_lbfgs if
+        args.max_lbfgs > 0 else 0)
+    writer = None
+    if args.solfile:
+        writer = solutions.SolutionWriter(
+            args.solfile, ms.freq0, ms.fdelta,
+            ms.tilesz * ms.tdelta / 60.0, ms.N, state.M, state.Mt)
+    pinit = state.J.clone()
+    for ti, tile in enumerate(ms.tiles()):
+        t0 = time.time()
+        flags = uv_flags(tile, args)
+        cohs = sage.precalc_coherencies(pack, tile)
+        if device != 'cpu':
+            cohs = cohs.to(torch.complex64)
+        bb = ms.bb_tensor(device=device)
+        res0, res1 = sage.sagefit(state, cohs, tile, bb, opts, flags=flags)
+        ccid = args.ccid if args.ccid != -99999 else None
+        xres = sage.calculate_residuals_multifreq(
+            state, pack, tile, bb, ccid=ccid, rho=args.rho_corr)
+        ms.write_column(args.outcol, ti, xres)
+        if writer:
+            writer.write_tile(state)
+        # divergence safeguard (fullbatch_mode.cpp:622-632)
+        if not np.isfinite(res1) or (res0 > 0 and res1 > 5 * res0):
+            state.J = pinit.clone()
+        mean_nu = float(state.nu.mean())
+        print(f"tile {ti}: residual {res0:.6f} -> {res1:.6f}, mean nu "
+              f"{mean_nu:.1f} ({time.time() - t0:.1f}s)")
+    ms.save()
+    if writer:
+        writer.close()
+
+
+def run_simulation(args):
+    """-a 1/2/3: predict (optionally corrupted by solutions), write/add/
+    subtract (fullbatch_mode.cpp:536-591)."""
+    from ..solvers import sage
+    from ..ops import dispatch as ops
+    from .. import solutions
+    ms, pack, clusters, device, dtype = load_context(args)
+    cdtype = torch.complex64 if dtype == torch.float32 else torch.complex128
+    state = sage.CalState(pack, ms.N, device=device, dtype=cdtype)
+    if args.solfile:
+        hdr, tiles = solutions.read_solutions(args.solfile)
+        if tiles:
+            state.J = solutions.reorder_read_tile(
+                tiles[0], state.nchunks).to(device=device, dtype=cdtype)
+    bb = ms.bb_tensor(device=device)
+    for ti, tile in enumerate(ms.tiles()):
+        V = torch.zeros_like(tile.xo)
+        fdelta_ch = tile.fdelta / len(tile.freqs)
+        for fi, f in enumerate(tile.freqs):
+            cohs = ops.predict_coh(pack, tile.u, tile.v, tile.w, float(f),
+                                   tile.freq0, fdelta_ch, tile.tdelta,
+                                   tile.dec0)
+            if cohs.dtype != cdtype:
+                cohs = cohs.to(cdtype)
+            V[fi] = sage.total_model(state, cohs, bb, tile.tilesz,
+                                     tile.Nbase)
+        if args.dosim == 2:
+            out = tile.xo + V
+        elif args.dosim == 3:
+            out = tile.xo - V
+        else:
+            out = V
+        ms.write_column(args.outcol, ti, out)
+        print(f"tile {ti}: simulated (mode {args.dosim})")
+    ms.save()
+
+
+def run_stochastic(args):
+    """-N epochs: minibatch (bandpass-consensus when -w > 1) calibration
+    (minibatch_mode.cpp / minibatch_consensus_mode.cpp)."""
+    from ..solvers.stochastic import MinibatchConsensusCalibration
+    ms, pack, clusters, device, dtype = load_context(args)
+    cdtype = torch.complex64 if dtype == torch.float32 else torch.complex128
+    cal = MinibatchConsensusCalibration(
+        pack, ms.N, ms.freqs, nsolbw=args.minibands, Npoly=args.npoly,
+        poly_type=args.polytype, rho=args.admm_rho, device=device,
+        dtype=cdtype)
+    bb = ms.bb_tensor(device=device)
+    for ti, tile in enumerate(ms.tiles()):
+        for ep in range(args.epochs):
+            cal.epoch(tile, bb, nmb=args.minibatches,
+                      lbfgs_iters=args.max_lbfgs,
+                      robust_nu=(args.nulow + args.nuhigh) / 2)
+        xres = cal.residuals(tile, bb)
+        ms.write_column(args.outcol, ti, xres)
+        print(f"tile {ti}: stochastic epochs {args.epochs}, "
+              f"res {float(xres.abs().pow(2).mean()):.6f}")
+    ms.save()
+
+
+def main(argv=None):
+    args = build_argparser().parse_args(argv)
+    if not args.ms and args.mslist:
+        with open(args.mslist) as f:
+            names = [l.strip() for l in f if l.strip()]
+        args.ms = names[0]
+    if not args.ms:
+        print("need -d MS or -f MSlist", file=sys.stderr)
+        return 1
+    if args.dosim > 0:
+        run_simulation(args)
+    elif args.epochs > 0:
+        run_stochastic(args)
+    else:
+        run_calibration(args)
+    return 0
+
+
+if __name__ == '__main__':
+    sys.exit(main())

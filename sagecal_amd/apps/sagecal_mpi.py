@@ -1,0 +1,133 @@
+"""sagecal-mpi CLI — distributed multi-band consensus calibration.
+
+The reference's MPI hub (src/MPI/sagecal_master.cpp + sagecal_slave.cpp)
+re-designed for one rank per GPU over RCCL/xGMI (torchrun launcher, gloo on
+CPU): every rank loads its own sub-band MS; the master's Z-update runs
+replicated on all ranks after one fused all-reduce (consensus/admm.py).
+
+Launch:
+  torchrun --nnodes 1 --nproc-per-node W --master-addr 127.0.0.1 \\
+      sagecal_mpi.py -f mslist.txt -s sky.txt -c cluster.txt -A 10 -P 2
+
+mslist.txt: one NpzMS path per line; rank r takes line r (more MSs than
+ranks: round-robin multiplexing like sagecal_master.cpp:1055).
+"""
+import argparse
+import os
+import sys
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+def build_argparser():
+    ap = argparse.ArgumentParser(prog='sagecal-mpi')
+    ap.add_argument('-f', dest='mslist', required=True,
+                    help='file listing per-band MS paths (one per rank)')
+    ap.add_argument('-s', dest='sky', required=True)
+    ap.add_argument('-c', dest='cluster', required=True)
+    ap.add_argument('-p', dest='solfile', help='per-rank solutions file')
+    ap.add_argument('-F', dest='format', type=int, default=0)
+    ap.add_argument('-e', dest='max_emiter', type=int, default=3)
+    ap.add_argument('-g', dest='max_iter', type=int, default=10)
+    ap.add_argument('-l', dest='max_lbfgs', type=int, default=0)
+    ap.add_argument('-t', dest='tilesz', type=int, default=10)
+    ap.add_argument('-j', dest='solver_mode', type=int, default=5)
+    ap.add_argument('-A', dest='nadmm', type=int, default=10)
+    ap.add_argument('-P', dest='npoly', type=int, default=2)
+    ap.add_argument('-Q', dest='polytype', type=int, default=0)
+    ap.add_argument('-r', dest='admm_rho', type=float, default=5.0)
+    ap.add_argument('-G', dest='rhofile',
+                    help='per-cluster regularization file')
+    ap.add_argument('-a', dest='use_bb', type=int, default=0,
+                    help='1: Barzilai-Borwein adaptive rho')
+    ap.add_argument('-U', dest='use_global', type=int, default=0,
+                    help='1: residuals from the global solution B Z')
+    ap.add_argument('-L', dest='nulow', type=float, default=2.0)
+    ap.add_argument('-H', dest='nuhigh', type=float, default=30.0)
+    ap.add_argument('-O', dest='outcol', default='residual')
+    ap.add_argument('-V', dest='verbose', action='store_true')
+    return ap
+
+
+def main(argv=None):
+    args = build_argparser().parse_args(argv)
+    rank = int(os.environ.get('RANK', '0'))
+    world = int(os.environ.get('WORLD_SIZE', '1'))
+    local_rank = int(os.environ.get('LOCAL_RANK', str(rank)))
+    use_gpu = torch.cuda.is_available()
+    if world > 1 and not dist.is_initialized():
+        dist.init_process_group('nccl' if use_gpu else 'gloo')
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    device = f'cuda:{local_rank}' if use_gpu else 'cpu'
+    dtype = torch.float32 if use_gpu else torch.float64
+    cdtype = torch.complex64 if use_gpu else torch.complex128
+
+    from .. import sky as skymod, msdata, solutions
+    from ..ops.reference import SourcePack
+    from ..solvers import sage
+    from ..consensus.admm import ConsensusADMM
+
+    with open(args.mslist) as f:
+        names = [l.strip() for l in f if l.strip()]
+    my_ms = names[rank % len(names)] if world <= len(names) else \
+        names[rank]
+    ms = msdata.NpzMS(my_ms, tilesz=args.tilesz, device=device,
+                      dtype=dtype)
+    clusters = skymod.read_sky_cluster(args.sky, args.cluster, ms.ra0,
+                                       ms.dec0, ms.freq0, fmt=args.format)
+    pack = SourcePack(clusters)
+    state = sage.CalState(pack, ms.N, device=device, dtype=cdtype)
+
+    rho = torch.full((pack.M,), args.admm_rho)
+    if args.rhofile:
+        arho, _ = skymod.read_arho_file(args.rhofile, clusters)
+        rho = torch.tensor(arho, dtype=torch.float64)
+        rho[rho == 0] = args.admm_rho
+
+    # gather every band's centre frequency (TAG_MSAUX metadata exchange)
+    f0s = torch.zeros(world)
+    f0s[rank] = ms.freq0
+    if world > 1:
+        dist.all_reduce(f0s)
+    freq0_global = float(f0s.mean())
+
+    adm = ConsensusADMM(state, f0s.tolist(), freq0_global, rank, world,
+                        Npoly=min(args.npoly, world), poly_type=args.polytype,
+                        rho=rho, use_bb=bool(args.use_bb))
+    opts = sage.SageSolveOptions(
+        max_emiter=args.max_emiter, max_iter=args.max_iter,
+        solver_mode=args.solver_mode, robust_nulow=args.nulow,
+        robust_nuhigh=args.nuhigh)
+    writer = None
+    if args.solfile:
+        writer = solutions.SolutionWriter(
+            f"{args.solfile}.rank{rank}", ms.freq0, ms.fdelta,
+            ms.tilesz * ms.tdelta / 60.0, ms.N, state.M, state.Mt)
+    bb = ms.bb_tensor(device=device)
+    for ti, tile in enumerate(ms.tiles()):
+        cohs = sage.precalc_coherencies(pack, tile)
+        if cohs.dtype != cdtype:
+            cohs = cohs.to(cdtype)
+        res0, res1 = adm.run(cohs, tile, bb, opts, n_admm=args.nadmm)
+        if args.use_global:
+            state.J = adm.global_solution()
+        xres = sage.calculate_residuals_multifreq(state, pack, tile, bb)
+        ms.write_column(args.outcol, ti, xres)
+        if writer:
+            writer.write_tile(state)
+        if rank == 0 or args.verbose:
+            print(f"rank {rank} tile {ti}: res {res0:.6f} -> {res1:.6f} "
+                  f"rho[0]={float(adm.rho[0]):.2f}")
+    ms.save()
+    if writer:
+        writer.close()
+    if world > 1:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == '__main__':
+    sys.exit(main())

@@ -198,3 +198,122 @@ class SyntheticMS:
         return TileData(u, v, w, x, xo, flags, self.freqs, self.freq0,
                         self.fdelta, self.tdelta, T, self.Nbase,
                         dec0=self.dec0)
+
+
+class NpzMS:
+    """Simple on-disk measurement-set container (.npz) behind the same tile
+    interface as SyntheticMS.
+
+    Plays the role of the casacore MS backend (Data::loadData/writeData,
+    src/MS/data.cpp:604/:1393) in environments without casacore: DATA
+    stored as [Ntime*Nbase, Nchan, 2, 2] complex64 rows (time-major,
+    pairs p<q), uvw in metres, plus aux metadata. `writeData` saves
+    residuals/corrected data to a chosen column name.
+
+    Fields: u,v,w [rows] f64 (metres); data [rows, F, 2, 2] c64;
+    flags [rows] bool; freqs [F]; N, Nbase, tilesz, tdelta, ra0, dec0.
+    """
+
+    def __init__(self, path, tilesz=None, device='cpu',
+                 dtype=torch.float64):
+        self.path = path
+        z = np.load(path)
+        self.N = int(z['N'])
+        self.Nbase = self.N * (self.N - 1) // 2
+        self.freqs = z['freqs']
+        self.Nchan = len(self.freqs)
+        self.freq0 = float(np.mean(self.freqs))
+        self.fdelta = float(z['fdelta']) if 'fdelta' in z else \
+            float(self.freqs.max() - self.freqs.min() + 1)
+        self.tdelta = float(z['tdelta'])
+        self.ra0 = float(z['ra0'])
+        self.dec0 = float(z['dec0'])
+        self.tilesz = int(tilesz or z.get('tilesz', 10))
+        rows = z['u'].shape[0]
+        self.Ntime = rows // self.Nbase
+        self._z = {k: z[k] for k in z.files}
+        self.pairs = baseline_pairs(self.N)
+        self.device = device
+        self.dtype = dtype
+        self.columns = {}
+
+    @staticmethod
+    def create(path, ms, data, flags=None):
+        """Write an NpzMS from a SyntheticMS-like object + data
+        [rows, F, 2, 2] complex."""
+        rows = data.shape[0]
+        T = rows // ms.Nbase
+        us, vs, ws = [], [], []
+        for ti in range(max(1, T // ms.tilesz)):
+            ub, vb, wb = ms.uvw_for(ti)
+            us.append(ub); vs.append(vb); ws.append(wb)
+        np.savez_compressed(
+            path, N=ms.N, freqs=ms.freqs, fdelta=ms.fdelta,
+            tdelta=ms.tdelta, ra0=ms.ra0, dec0=ms.dec0, tilesz=ms.tilesz,
+            u=np.concatenate(us) * C_LIGHT, v=np.concatenate(vs) * C_LIGHT,
+            w=np.concatenate(ws) * C_LIGHT,
+            data=data.astype(np.complex64),
+            flags=(flags if flags is not None
+                   else np.zeros(rows, dtype=bool)))
+
+    def bb_tensor(self, device=None):
+        dev = device or self.device
+        bb = np.tile(self.pairs, (self.tilesz, 1))
+        return torch.tensor(bb, dtype=torch.long, device=dev)
+
+    def n_tiles(self):
+        return max(1, self.Ntime // self.tilesz)
+
+    def tiles(self):
+        for ti in range(self.n_tiles()):
+            yield self.load_tile(ti)
+
+    def load_tile(self, ti):
+        T = self.tilesz
+        r0, r1 = ti * T * self.Nbase, (ti + 1) * T * self.Nbase
+        cdtype = torch.complex128 if self.dtype == torch.float64 \
+            else torch.complex64
+        u = torch.tensor(self._z['u'][r0:r1] / C_LIGHT, dtype=self.dtype,
+                         device=self.device)
+        v = torch.tensor(self._z['v'][r0:r1] / C_LIGHT, dtype=self.dtype,
+                         device=self.device)
+        w = torch.tensor(self._z['w'][r0:r1] / C_LIGHT, dtype=self.dtype,
+                         device=self.device)
+        xo = torch.tensor(self._z['data'][r0:r1], dtype=cdtype,
+                          device=self.device).permute(1, 0, 2, 3).contiguous()
+        flags = torch.tensor(self._z['flags'][r0:r1], dtype=torch.bool,
+                             device=self.device)
+        x = xo.mean(dim=0)
+        return TileData(u, v, w, x, xo, flags, self.freqs, self.freq0,
+                        self.fdelta, self.tdelta, T, self.Nbase,
+                        dec0=self.dec0)
+
+    def write_column(self, name, ti, xres):
+        """Stage output rows (residual/corrected data) for tile ti; save()
+        persists (Data::writeData analog)."""
+        col = self.columns.setdefault(
+            name, np.zeros_like(self._z['data']))
+        T = self.tilesz
+        r0 = ti * T * self.Nbase
+        arr = xres.permute(1, 0, 2, 3).cpu().numpy().astype(np.complex64)
+        col[r0:r0 + arr.shape[0]] = arr
+
+    def save(self, path=None):
+        out = dict(self._z)
+        for k, v in self.columns.items():
+            out[k] = v
+        np.savez_compressed(path or self.path, **out)
+
+
+def make_synthetic_npz(path, N=8, tilesz=4, Ntime=4, Nchan=2, pack=None,
+                       **kw):
+    """Generate a synthetic observation and persist it as NpzMS."""
+    ms = SyntheticMS(N=N, tilesz=tilesz, Ntime=Ntime, Nchan=Nchan,
+                     pack=pack, **kw)
+    datas = []
+    for ti in range(max(1, Ntime // tilesz)):
+        tile = ms.load_tile(ti)
+        datas.append(tile.xo.permute(1, 0, 2, 3).cpu().numpy())
+    data = np.concatenate(datas, axis=0)
+    NpzMS.create(path, ms, data)
+    return ms
