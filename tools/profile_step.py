@@ -25,7 +25,8 @@ def main():
     ap.add_argument('--chan', type=int, default=8)
     ap.add_argument('--emiter', type=int, default=3)
     ap.add_argument('--maxiter', type=int, default=12)
-    ap.add_argument('--em-group', type=int, default=2)
+    ap.add_argument('--em-group', type=int, default=3)
+    ap.add_argument('--reps', type=int, default=2)
     args = ap.parse_args()
     sys.argv = [sys.argv[0]]
 
@@ -52,7 +53,7 @@ def main():
     sage.sagefit(state, cohs, tile, bb, opts)
     sage.calculate_residuals_multifreq(state, pack, tile, bb)
 
-    for rep in range(2):
+    for rep in range(args.reps):
         state.reset()
         t0 = t_sync()
         cohs = sage.precalc_coherencies(pack, tile).to(torch.complex64)
