@@ -24,7 +24,7 @@ def main():
     ap.add_argument('--tilesz', type=int, default=60)
     ap.add_argument('--chan', type=int, default=8)
     ap.add_argument('--emiter', type=int, default=3)
-    ap.add_argument('--maxiter', type=int, default=12)
+    ap.add_argument('--maxiter', type=int, default=8)
     ap.add_argument('--em-group', type=int, default=3)
     ap.add_argument('--reps', type=int, default=2)
     args = ap.parse_args()
@@ -32,7 +32,7 @@ def main():
 
     import bench
     from sagecal_amd.solvers import sage
-    from sagecal_amd.constants import SM_RTR_OSRLM_RLBFGS
+    from sagecal_amd.constants import SM_RLM_RLBFGS
 
     class A:
         pass
@@ -46,8 +46,8 @@ def main():
                           dtype=torch.complex64)
     opts = sage.SageSolveOptions(max_emiter=args.emiter,
                                  max_iter=args.maxiter,
-                                 solver_mode=SM_RTR_OSRLM_RLBFGS,
-                                 robust_outer=2, em_group=args.em_group)
+                                 solver_mode=SM_RLM_RLBFGS,
+                                 robust_outer=1, em_group=args.em_group)
     # warm
     cohs = sage.precalc_coherencies(pack, tile).to(torch.complex64)
     sage.sagefit(state, cohs, tile, bb, opts)
