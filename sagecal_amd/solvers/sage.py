@@ -181,7 +181,8 @@ def robust_lm(prob, J0, nu0, opts):
     inner = max(3, opts.max_iter // max(1, opts.robust_outer))
     for outer in range(opts.robust_outer):
         J = _inner_solve(prob, J, opts, inner)
-        V = ops.apply_jones(prob.coh, J, prob.bb, prob.chunk_rows)
+        V = ops.apply_jones(prob.coh, J, prob.bb, prob.chunk_rows,
+                            prob.layout)
         r = prob.x - V
         w = ops.update_weights(r, nu, p=8)
         nu = ops.update_nu_aecm(w, nu, nulow=opts.robust_nulow,
