@@ -10,7 +10,7 @@ def main():
     dev = 'cuda:0'
     import bench
     from sagecal_amd.solvers import sage
-    from sagecal_amd.constants import SM_RTR_OSRLM_RLBFGS
+    from sagecal_amd.constants import SM_RLM_RLBFGS
 
     class A: pass
     a = A(); a.__dict__.update(stations=64, dirs=10, srcs=5, tilesz=60,
@@ -19,19 +19,16 @@ def main():
 
     configs = [
         # (emiter, maxiter, robust_outer, em_group, label)
-        (3, 12, 2, 2, 'current'),
-        (3, 8, 1, 2, 'lean'),
-        (2, 10, 1, 2, 'leaner'),
-        (2, 8, 1, 2, 'min'),
-        (3, 12, 2, 3, 'g3'),
-        (4, 8, 1, 2, 'more-em-lean'),
         (3, 8, 1, 3, 'lean-g3'),
-        (2, 6, 1, 2, 'tiny'),
+        (3, 8, 1, 5, 'lean-g5'),
+        (3, 8, 1, 10, 'lean-g10'),
+        (4, 8, 1, 5, 'em4-g5'),
+        (3, 10, 1, 5, 'it10-g5'),
     ]
     for emiter, maxiter, ro, eg, label in configs:
         opts = sage.SageSolveOptions(
             max_emiter=emiter, max_iter=maxiter,
-            solver_mode=SM_RTR_OSRLM_RLBFGS, robust_outer=ro, em_group=eg)
+            solver_mode=SM_RLM_RLBFGS, robust_outer=ro, em_group=eg)
         state = sage.CalState(pack, 64, device=dev, dtype=torch.complex64)
         cohs = sage.precalc_coherencies(pack, tile).to(torch.complex64)
         sage.sagefit(state, cohs, tile, bb, opts)   # warm
