@@ -43,6 +43,16 @@ class GPUPack:
         self.sphi = torch.where(up, f32(pack.sphi), zero).contiguous()
         self.stype = pack.stype.to(device=device,
                                    dtype=torch.int32).contiguous()
+        # shapelet sources are handled by the torch path (shapelet.py):
+        # zero their fluxes for the kernel so they contribute nothing there
+        self.shapelets = dict(getattr(pack, 'shapelets', {}))
+        if self.shapelets:
+            sh = self.stype == 4
+            for k in ('sI', 'sQ', 'sU', 'sV', 'sI0', 'sQ0', 'sU0', 'sV0'):
+                t = getattr(self, k).clone()
+                t[sh] = 0.0
+                setattr(self, k, t.contiguous())
+        self.pack_ref = pack
         # zero shape angles for point sources keep the envelope branch out
         self.cluster_off = pack.cluster_off.to(device=device,
                                                dtype=torch.int32).contiguous()
@@ -103,7 +113,56 @@ def predict_coh(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0, **kw):
         gp.eX, gp.eY, gp.eP, gp.cxi, gp.sxi, gp.cphi, gp.sphi, r1,
         gp.stype, gp.cluster_off, float(freq), float(fdelta) * 0.5,
         float(tdelta))
-    return out.view(gp.M, -1, 2, 2)
+    out = out.view(gp.M, -1, 2, 2)
+    if gp.shapelets:
+        out = out + _shapelet_coh(gp, u64, v64, w64, freq, freq0, fdelta,
+                                  tdelta, dec0).to(out.dtype)
+    return out
+
+
+def _shapelet_coh(gp, u, v, w, freq, freq0, fdelta, tdelta, dec0):
+    """Additive coherency of shapelet sources, computed with torch on
+    device (mirrors reference.predict_coh for stype==4 sources only)."""
+    from .. import shapelet as shmod
+    import numpy as np
+    M = gp.M
+    B = u.shape[0]
+    out = torch.zeros(M, B, 2, 2, dtype=torch.complex64, device=u.device)
+    co = gp.cluster_off.cpu().numpy()
+    pk = gp.pack_ref
+    sIs, sQs, sUs, sVs = gp.fluxes_at(freq, freq0)
+    for gi, (n0, beta, modes) in gp.shapelets.items():
+        ci = int(np.searchsorted(co, gi, side='right') - 1)
+        ll = float(gp.ll[gi]); mm = float(gp.mm[gi]); nn1 = float(gp.nn1[gi])
+        G = 2.0 * math.pi * (u * ll + v * mm + w * nn1)
+        ph = torch.remainder(G * freq, 2.0 * math.pi)
+        phc = torch.complex(torch.cos(ph), torch.sin(ph))
+        smf = G * (fdelta * 0.5)
+        sm = torch.where(smf.abs() > 1e-12,
+                         (torch.sin(smf) / smf).abs(),
+                         torch.ones_like(smf))
+        if tdelta > 0:
+            bl = torch.sqrt(u * u + v * v + w * w) * freq
+            r1 = math.sqrt(ll * ll + (math.sin(dec0) * mm) ** 2)
+            prod = 7.2921150e-5 * tdelta * bl * r1
+            sm = sm * torch.where(prod > 1e-12,
+                                  1.0645 * torch.erf(0.8326 * prod)
+                                  / prod.clamp_min(1e-12),
+                                  torch.ones_like(prod))
+        envc = shmod.shapelet_contrib(
+            u * freq, v * freq, w * freq, float(gp.eX[gi]),
+            float(gp.eY[gi]), float(gp.eP[gi]), float(pk.cxi[gi]),
+            float(pk.sxi[gi]), float(pk.cphi[gi]), float(pk.sphi[gi]),
+            bool(pk.use_proj[gi]), beta, n0, torch.as_tensor(modes))
+        term = (phc * sm.to(phc.dtype) * envc).to(torch.complex64)
+        # note: use_proj baked as identity when off; contrib uses proj form
+        I = float(sIs[gi]); Q = float(sQs[gi])
+        U_ = float(sUs[gi]); V_ = float(sVs[gi])
+        out[ci, :, 0, 0] += term * (I + Q)
+        out[ci, :, 0, 1] += term * complex(U_, V_)
+        out[ci, :, 1, 0] += term * complex(U_, -V_)
+        out[ci, :, 1, 1] += term * (I - Q)
+    return out
 
 
 class BaselineLayout:

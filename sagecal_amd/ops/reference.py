@@ -66,6 +66,13 @@ class SourcePack:
             np.concatenate([c.use_proj for c in clusters]).astype(np.int32),
             dtype=torch.int32, device=device)
         self.K = int(self.cluster_off[-1])
+        # shapelet metadata: global source idx -> (n0, beta, modes)
+        self.shapelets = {}
+        for ci, c in enumerate(clusters):
+            base = offs[ci]
+            for (li, n0, beta, coeff) in c.shapelets:
+                self.shapelets[base + li] = (int(n0), float(beta),
+                                             np.asarray(coeff))
 
     def to(self, device, dtype=None):
         for f_ in self.FIELDS:
@@ -194,10 +201,21 @@ def predict_coh(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0,
         env = _extended_envelope(pack, sel, uc * freq, vc * freq, wc * freq)
         if env is not None:
             phr, phi = phr * env, phi * env
-        if shapelet_env is not None and ci in shapelet_env:
-            # complex envelope per (B, S) for shapelet sources of cluster ci
-            for si, envc in shapelet_env[ci]:
-                pc = (phr[:, si] + 1j * phi[:, si]) * envc
+        # shapelet sources: complex uv envelope (shapelet.py)
+        if getattr(pack, 'shapelets', None):
+            from .. import shapelet as shmod
+            for gi, (n0, beta, modes) in pack.shapelets.items():
+                if not (s0 <= gi < s1):
+                    continue
+                si = gi - s0
+                envc = shmod.shapelet_contrib(
+                    u * freq, v * freq, w * freq,
+                    float(pack.eX[gi]), float(pack.eY[gi]),
+                    float(pack.eP[gi]), float(pack.cxi[gi]),
+                    float(pack.sxi[gi]), float(pack.cphi[gi]),
+                    float(pack.sphi[gi]), bool(pack.use_proj[gi]),
+                    beta, n0, modes)
+                pc = torch.complex(phr[:, si], phi[:, si]) * envc
                 phr[:, si] = pc.real
                 phi[:, si] = pc.imag
         I, Q, U, V = _source_flux_at(pack, sel, float(freq), freq0)

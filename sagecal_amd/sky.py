@@ -100,7 +100,7 @@ def _parse_sky_lines(path):
             yield line.split()
 
 
-def read_sky_model(path, fmt=0):
+def read_sky_model(path, fmt=0, modes_dir=None):
     """Parse an LSM sky model file into a dict name -> Source.
 
     fmt=0: 1-term spectral index (16 numeric cols after name)
@@ -132,8 +132,26 @@ def read_sky_model(path, fmt=0):
                 stype = STYPE_RING
             elif c0 == 'S':
                 stype = STYPE_SHAPELET
-        sources[name] = Source(name, ra, dec, sI, sQ, sU, sV,
-                               si0, si1, si2, RM, eX, eY, eP, f0, stype)
+        src = Source(name, ra, dec, sI, sQ, sU, sV,
+                     si0, si1, si2, RM, eX, eY, eP, f0, stype)
+        if stype == STYPE_SHAPELET:
+            # reference looks for <name>.fits.modes in the working dir
+            # (readsky.c:149); we search next to the sky file as well
+            import os
+            from . import shapelet as shmod
+            for d in (modes_dir, os.path.dirname(os.path.abspath(path)),
+                      '.'):
+                if d is None:
+                    continue
+                mf = os.path.join(d, name + '.fits.modes')
+                if os.path.exists(mf):
+                    src.sh_n0, src.sh_beta, src.sh_coeff =                         shmod.read_modes_file(mf)
+                    break
+            if src.eX == 0:
+                src.eX = 1.0
+            if src.eY == 0:
+                src.eY = 1.0
+        sources[name] = src
     return sources
 
 
