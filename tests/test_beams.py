@@ -1,0 +1,105 @@
+"""Beam model tests: array factor, tile beamformer, element patterns,
+beam-aware predict."""
+import numpy as np
+import pytest
+import torch
+
+from sagecal_amd import beams, sky, coords
+from sagecal_amd.ops.reference import SourcePack
+
+
+def _cfg(K=16, seed=0, shared=True, nsta=4):
+    rng = np.random.default_rng(seed)
+    lon, lat = 0.1, 0.92
+    b_ra0, b_dec0 = 0.0, np.pi / 4
+    el = [rng.uniform(-10, 10, (K, 3)) * np.array([1, 1, 0.05])
+          for _ in range(nsta)]
+    if shared:
+        el = [el[0]] * nsta
+    return beams.ArrayConfig(el, lon, lat, b_ra0, b_dec0)
+
+
+def test_array_beam_unity_at_pointing():
+    """At the pointing direction the delay-steered factor is exactly 1."""
+    cfg = _cfg()
+    af = beams.array_beam(cfg, [cfg.b_ra0], [cfg.b_dec0], [150e6],
+                          [57000.0])
+    torch.testing.assert_close(af[0, 0, 0, 0],
+                               torch.complex(torch.tensor(1.0).double(),
+                                             torch.tensor(0.0).double()))
+
+
+def test_array_beam_attenuates_off_axis():
+    cfg = _cfg(K=64)
+    # a source a few degrees away
+    af0 = beams.array_beam(cfg, [cfg.b_ra0], [cfg.b_dec0], [150e6],
+                           [57000.0])
+    af1 = beams.array_beam(cfg, [cfg.b_ra0 + 0.3], [cfg.b_dec0 - 0.2],
+                           [150e6], [57000.0])
+    assert abs(af1[0, 0, 0, 0]) < abs(af0[0, 0, 0, 0])
+
+
+def test_array_beam_below_horizon_zero():
+    cfg = _cfg()
+    # anti-pointing: below horizon
+    af = beams.array_beam(cfg, [cfg.b_ra0 + np.pi], [-cfg.b_dec0],
+                          [150e6], [57000.0])
+    assert abs(af[0, 0, 0, 0]) == 0.0
+
+
+def test_tile_beam_product():
+    cfg = _cfg(K=8)
+    cfg.tile_enu = np.array([[0.0, 0.0, 0.0], [1.25, 0, 0], [0, 1.25, 0],
+                             [1.25, 1.25, 0]])
+    tb = beams.tile_beam(cfg, [cfg.b_ra0], [cfg.b_dec0], [150e6],
+                         [57000.0])
+    torch.testing.assert_close(
+        tb[0, 0, 0, 0].real, torch.tensor(1.0).double())
+
+
+def test_element_coeffs_roundtrip(tmp_path):
+    c = beams.make_synthetic_element_coeffs(n0=3, freqs=[120e6, 150e6])
+    p = str(tmp_path / 'elem.npz')
+    c.save(p)
+    c2 = beams.ElementCoeffs.load(p)
+    assert c2.n0 == 3
+    np.testing.assert_allclose(c2.ctheta_x, c.ctheta_x)
+
+
+def test_element_beam_jones():
+    c = beams.make_synthetic_element_coeffs(n0=3)
+    az = np.array([0.0, 1.0, 2.0])
+    el = np.array([1.3, 0.9, 0.5])
+    E = beams.element_beam(c, az, el, 150e6)
+    assert E.shape == (3, 2, 2)
+    assert torch.isfinite(torch.view_as_real(E)).all()
+    # X and Y dipole responses differ (orthogonal patterns)
+    assert float((E[:, 0, 0] - E[:, 1, 1]).abs().max()) > 1e-6
+
+
+def test_predict_withbeam_changes_vis():
+    srcs, clist = sky.make_synthetic_sky(M=2, nsrc_per_cluster=2, seed=3)
+    clusters = sky.build_clusters(srcs, clist, 0.0, np.pi / 4, 150e6)
+    pack = SourcePack(clusters)
+    cfg = _cfg(K=32)
+    N, T = 6, 2
+    Nbase = N * (N - 1) // 2
+    rng = np.random.default_rng(4)
+    B = Nbase * T
+    u = torch.tensor(rng.uniform(-1e-5, 1e-5, B))
+    v = torch.tensor(rng.uniform(-1e-5, 1e-5, B))
+    w = torch.zeros(B, dtype=torch.float64)
+    bb = torch.tensor([(p, q) for p in range(N) for q in range(p + 1, N)]
+                      * T)
+    tmjd = np.array([57000.0, 57000.0001])
+    # beam config needs one element layout per station
+    cfg = beams.ArrayConfig([cfg.elements(0)] * N, cfg.lon, cfg.lat,
+                            cfg.b_ra0, cfg.b_dec0)
+    coh_b = beams.predict_coh_withbeam(pack, u, v, w, 150e6, 150e6, 0.0,
+                                       0.0, np.pi / 4, cfg, tmjd, bb,
+                                       Nbase, T)
+    from sagecal_amd.ops import reference as R
+    coh0 = R.predict_coh(pack, u, v, w, 150e6, 150e6, 0.0, 0.0, np.pi / 4)
+    assert coh_b.shape == coh0.shape
+    d = float((coh_b - coh0).abs().mean() / coh0.abs().mean())
+    assert 1e-4 < d < 1.0, d   # beam modifies but does not destroy
