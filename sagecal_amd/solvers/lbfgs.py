@@ -177,3 +177,36 @@ def polish(state, cohs, tile, bb, opts):
     v0 = _pack_params(state.J)
     v1, _, _ = lbfgs_fit(fg, v0, maxiter=opts.lbfgs_iters, m=7)
     state.J = _unpack_params(v1, Mt, N, state.J.dtype)
+
+
+def lbfgs_fit_bounded(fg, p0, lbound, ubound, maxiter=50, m=7, gtol=1e-9,
+                      mem=None):
+    """Bound-constrained LBFGS (the role of lbfgsb.c, Dirac.h:1797-1846),
+    via gradient projection: iterates are projected onto [lbound, ubound],
+    the search direction is the two-loop direction with bound-active
+    coordinates zeroed, and the line search walks the projected path."""
+    lb = torch.as_tensor(lbound, dtype=p0.dtype, device=p0.device)
+    ub = torch.as_tensor(ubound, dtype=p0.dtype, device=p0.device)
+
+    def proj(p):
+        return torch.minimum(torch.maximum(p, lb), ub)
+
+    p = proj(p0.clone())
+    if mem is None:
+        mem = LBFGSMemory(m, p.numel(), dtype=p.dtype, device=p.device)
+    f, g = fg(p)
+    for it in range(maxiter):
+        active = ((p <= lb) & (g > 0)) | ((p >= ub) & (g < 0))
+        gf = torch.where(active, torch.zeros_like(g), g)
+        if float(gf.abs().max()) < gtol:
+            break
+        d = -mem.two_loop(gf)
+        d = torch.where(active, torch.zeros_like(d), d)
+        alpha = _backtracking(lambda q: fg(proj(q))[0], p, f, g, d)
+        if alpha == 0.0:
+            break
+        pn = proj(p + alpha * d)
+        fn, gn = fg(pn)
+        mem.push(pn - p, gn - g)
+        p, f, g = pn, fn, gn
+    return p, mem, {'f1': float(f)}
