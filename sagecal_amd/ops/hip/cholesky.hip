@@ -47,25 +47,8 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
 
   if (tid == 0) bad = 0;
   const float m = mu[bid];
-  // vectorized copy of lower-triangle rows (quads) + fused damping
-  {
-    const int nv = n >> 2;
-    for (int idx = tid; idx < n * nv; idx += NTH) {
-      const int r = idx / nv, c4 = (idx - r * nv) << 2;
-      if (c4 <= r) {
-        float4 v = *(const float4*)(A + (size_t)r * n + c4);
-        const int d = r - c4;
-        if (d < 4) {   // diagonal falls in this quad
-          if (d == 0) v.x += m;
-          else if (d == 1) v.y += m;
-          else if (d == 2) v.z += m;
-          else v.w += m;
-        }
-        *(float4*)(L + (size_t)r * n + c4) = v;
-      }
-    }
-  }
-  __syncthreads();
+  // no A->L copy: left-looking panels stage straight from the damped A;
+  // L/LT only ever hold FACTORED panels.
   if (stages == 0) return;
 
   for (int k = 0; k < n; k += NB) {
@@ -74,8 +57,15 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     // stage S = damped A panel rows k..n, cols k..k+NB into LDS (float4)
     for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
       const int r = idx >> 3, c4 = (idx & 7) << 2;
-      *(float4*)(pan + r * PST + c4) =
-          *(const float4*)(L + (size_t)(k + r) * n + k + c4);
+      float4 v = *(const float4*)(A + (size_t)(k + r) * n + k + c4);
+      const int d = r - c4;
+      if (d >= 0 && d < 4) {     // diagonal falls in this quad
+        if (d == 0) v.x += m;
+        else if (d == 1) v.y += m;
+        else if (d == 2) v.z += m;
+        else v.w += m;
+      }
+      *(float4*)(pan + r * PST + c4) = v;
     }
     __syncthreads();
     // ---- LEFT-LOOKING update: S -= Lp(rows x k) * Lc(32 x k)^T
