@@ -28,6 +28,20 @@ def build_problem(args, device, dtype, rank=0, world=1):
 
     srcs, clist = sky.make_synthetic_sky(
         M=args.dirs, nsrc_per_cluster=args.srcs, seed=17)
+    nshap = getattr(args, 'shapelet_dirs', 0)
+    if nshap:
+        # make the first sources of the first nshap clusters shapelets
+        from sagecal_amd import shapelet as shmod
+        rng_s = np.random.default_rng(23)
+        for ci in range(min(nshap, len(clist))):
+            name = clist[ci][2][0]
+            s = srcs[name]
+            s.stype = 4
+            s.eX = s.eY = 1.0
+            s.sh_n0 = 3
+            s.sh_beta = 1e-3
+            s.sh_coeff = rng_s.standard_normal(9) * [1, .3, .1, .3, .2,
+                                                     .05, .1, .05, .02]
     clusters = sky.build_clusters(srcs, clist, 0.0, np.pi / 4,
                                   args.freq0)
     pack = SourcePack(clusters)
@@ -97,6 +111,10 @@ def main():
     ap.add_argument('--joint', type=int, default=0)
     ap.add_argument('--em-group', type=int, default=5)
     ap.add_argument('--npoly', type=int, default=2)
+    ap.add_argument('--solver', choices=['lm', 'rtr'], default='lm',
+                    help='rtr: Riemannian trust-region (SKA config 5)')
+    ap.add_argument('--shapelet-dirs', type=int, default=0,
+                    help='make this many clusters shapelet (config 5)')
     ap.add_argument('--admm-rho', type=float, default=5.0)
     ap.add_argument('--no-consensus', action='store_true')
     ap.add_argument('--cpu', action='store_true')
@@ -120,7 +138,7 @@ def main():
         dtype = torch.float64
 
     from sagecal_amd.solvers import sage
-    from sagecal_amd.constants import SM_RLM_RLBFGS
+    from sagecal_amd.constants import SM_RLM_RLBFGS, SM_RTR_OSRLM_RLBFGS
 
     pack, ms, tile, bb = build_problem(args, device, dtype, rank, world)
     cdtype = torch.complex64 if dtype == torch.float32 else torch.complex128
@@ -135,7 +153,8 @@ def main():
     opts = sage.SageSolveOptions(
         max_emiter=1 if adm is not None else args.emiter,
         max_iter=args.maxiter,
-        solver_mode=SM_RLM_RLBFGS,  # robust LM (BASELINE.json names LM)
+        solver_mode=(SM_RTR_OSRLM_RLBFGS if args.solver == 'rtr'
+                     else SM_RLM_RLBFGS),
         robust_outer=args.robust_outer, em_group=args.em_group,
         joint_iters=args.joint)
 

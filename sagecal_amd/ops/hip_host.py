@@ -177,11 +177,11 @@ class BaselineLayout:
         key = (N, Nbase, str(device))
         if key not in _pidx_cache:
             pidx = torch.full((N, N), -1, dtype=torch.int32)
-            pcpu = self.pairs.cpu()
-            for i in range(Nbase):
-                p, q = int(pcpu[i, 0]), int(pcpu[i, 1])
-                if p >= 0 and q >= 0:
-                    pidx[min(p, q), max(p, q)] = i
+            pcpu = self.pairs.cpu().long()
+            ok = (pcpu[:, 0] >= 0) & (pcpu[:, 1] >= 0)
+            lo = torch.minimum(pcpu[:, 0], pcpu[:, 1])[ok]
+            hi = torch.maximum(pcpu[:, 0], pcpu[:, 1])[ok]
+            pidx[lo, hi] = torch.arange(Nbase, dtype=torch.int32)[ok]
             _pidx_cache[key] = pidx.to(device).contiguous()
         self.pidx = _pidx_cache[key]
 
