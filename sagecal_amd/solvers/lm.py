@@ -48,7 +48,23 @@ def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
              verbose=False):
     if (prob.layout is not None and prob.x.is_cuda
             and os.environ.get('SAGECAL_NO_GRAPH') != '1'):
-        return _lm_solve_graphed(prob, J0, maxiter, tau, eps1, eps2)
+        # run the fixed-iteration graph in segments of 4 with a cheap
+        # convergence poll between: late EM sweeps often converge in a few
+        # iterations and the remaining replays are wasted work
+        seg = 4 if maxiter > 4 else maxiter
+        J = J0
+        info = None
+        done = 0
+        cost_prev = None
+        while done < maxiter:
+            n = min(seg, maxiter - done)
+            J, info = _lm_solve_graphed(prob, J, n, tau, eps1, eps2)
+            done += n
+            c = float(info['final_cost'].sum())
+            if cost_prev is not None and                     abs(cost_prev - c) < 1e-4 * abs(cost_prev):
+                break
+            cost_prev = c
+        return J, info
     return _lm_solve_eager(prob, J0, maxiter, tau, eps1, eps2)
 
 
