@@ -47,9 +47,14 @@ class GPUPack:
         # zero their fluxes for the kernel so they contribute nothing there
         self.shapelets = dict(getattr(pack, 'shapelets', {}))
         if self.shapelets:
+            # keep the true fluxes for the host shapelet pass, zero them
+            # in the kernel arrays
             sh = self.stype == 4
+            self._shfull = {}
             for k in ('sI', 'sQ', 'sU', 'sV', 'sI0', 'sQ0', 'sU0', 'sV0'):
-                t = getattr(self, k).clone()
+                t = getattr(self, k)
+                self._shfull[k] = t.clone()
+                t = t.clone()
                 t[sh] = 0.0
                 setattr(self, k, t.contiguous())
         self.pack_ref = pack
@@ -130,7 +135,24 @@ def _shapelet_coh(gp, u, v, w, freq, freq0, fdelta, tdelta, dec0):
     out = torch.zeros(M, B, 2, 2, dtype=torch.complex64, device=u.device)
     co = gp.cluster_off.cpu().numpy()
     pk = gp.pack_ref
-    sIs, sQs, sUs, sVs = gp.fluxes_at(freq, freq0)
+    # true (un-zeroed) fluxes at this channel
+    full = gp._shfull
+    if abs(freq - freq0) < 1.0:
+        sIs, sQs, sUs, sVs = (full['sI'], full['sQ'], full['sU'],
+                              full['sV'])
+    else:
+        lf = torch.log(torch.tensor(float(freq), dtype=torch.float64,
+                                    device=gp.ll.device) / gp.f0)
+        lf = lf.to(torch.float32)
+        flog = (gp.spec_idx * lf + gp.spec_idx1 * lf ** 2
+                + gp.spec_idx2 * lf ** 3)
+
+        def _scale(s0):
+            mag = torch.exp(torch.log(s0.abs().clamp_min(1e-30)) + flog)
+            return torch.where(s0 == 0, torch.zeros_like(s0),
+                               torch.sign(s0) * mag)
+        sIs, sQs, sUs, sVs = (_scale(full['sI0']), _scale(full['sQ0']),
+                              _scale(full['sU0']), _scale(full['sV0']))
     for gi, (n0, beta, modes) in gp.shapelets.items():
         ci = int(np.searchsorted(co, gi, side='right') - 1)
         ll = float(gp.ll[gi]); mm = float(gp.mm[gi]); nn1 = float(gp.nn1[gi])
