@@ -52,6 +52,10 @@ def build_argparser():
     ap.add_argument('-k', dest='ccid', type=int, default=-99999,
                     help='correct residuals with this cluster id')
     ap.add_argument('-o', dest='rho_corr', type=float, default=1e-9)
+    ap.add_argument('-J', dest='phase_only', type=int, default=0,
+                    help='1: phase-only correction')
+    ap.add_argument('-i', dest='dodiag', type=int, default=0,
+                    help='1: replace output with influence diagnostics')
     ap.add_argument('-j', dest='solver_mode', type=int, default=5,
                     help='0 OSLM,1 LM,2 OSRLM,3 RLM,4 RTR,5 RRTR,6 NSD')
     ap.add_argument('-L', dest='nulow', type=float, default=2.0)
@@ -123,8 +127,56 @@ def run_calibration(args):
         bb = ms.bb_tensor(device=device)
         res0, res1 = sage.sagefit(state, cohs, tile, bb, opts, flags=flags)
         ccid = args.ccid if args.ccid != -99999 else None
+        if args.phase_only and ccid is not None:
+            # phase-only: normalize the correcting cluster's J to unit
+            # amplitude before inversion (residual.c phase-only path)
+            ids = getattr(pack, 'cluster_ids', list(range(state.M)))
+            match = [i for i, c in enumerate(ids) if c == ccid]
+            if match:
+                o = state.chunk_off[match[0]]
+                nc = state.nchunks[match[0]]
+                Jc = state.J[o:o + nc]
+                state.J[o:o + nc] = Jc / Jc.abs().clamp_min(1e-12)
         xres = sage.calculate_residuals_multifreq(
             state, pack, tile, bb, ccid=ccid, rho=args.rho_corr)
+        if args.dochan:
+            # per-channel refinement (-b 1, fullbatch_mode.cpp:453-499):
+            # polish each channel's solutions with a short joint LBFGS and
+            # recompute that channel's residual
+            from ..solvers import lbfgs as lbfgs_mod
+            from ..ops import dispatch as dops
+            fdelta_ch = tile.fdelta / len(tile.freqs)
+            for fi, f in enumerate(tile.freqs):
+                cohs_f = dops.predict_coh(pack, tile.u, tile.v, tile.w,
+                                          float(f), tile.freq0, fdelta_ch,
+                                          tile.tdelta, tile.dec0)
+                if cohs_f.dtype != state.J.dtype:
+                    cohs_f = cohs_f.to(state.J.dtype)
+
+                class _T:  # channel view of the tile
+                    pass
+                tf = _T()
+                tf.x = tile.xo[fi]
+                tf.tilesz, tf.Nbase = tile.tilesz, tile.Nbase
+                st2 = sage.CalState(pack, ms.N, device=tile.x.device,
+                                    dtype=state.J.dtype)
+                st2.J = state.J.clone()
+                st2.nu = state.nu.clone()
+                o2 = sage.SageSolveOptions(
+                    max_emiter=1, max_iter=4,
+                    solver_mode=args.solver_mode,
+                    lbfgs_iters=max(args.max_lbfgs, 4))
+                lbfgs_mod.polish(st2, cohs_f, tf, bb, o2)
+                V = sage.total_model(st2, cohs_f, bb, tile.tilesz,
+                                     tile.Nbase)
+                xres[fi] = tile.xo[fi] - V
+        if args.dodiag:
+            from ..solvers import diagnostics as diagmod
+            cohs0 = sage.precalc_coherencies(pack, tile)
+            if cohs0.dtype != state.J.dtype:
+                cohs0 = cohs0.to(state.J.dtype)
+            lev = diagmod.influence_map(state, cohs0, tile, bb)
+            xres = lev[None, :, None, None].expand_as(xres).to(xres.dtype)
         ms.write_column(args.outcol, ti, xres)
         if writer:
             writer.write_tile(state)

@@ -280,6 +280,13 @@ def sagefit(state, cohs, tile, bb, opts, flags=None, admm_terms=None):
     B = x.shape[0]
     valid = (~tile.flags) if flags is None else ~flags
     nvalid = max(int(valid.sum()), 1)
+    if nvalid < B:
+        # zero flagged rows in data and coherencies so they contribute
+        # nothing to any solve (preset_flags_and_data,
+        # baseline_utils.c semantics)
+        vm = valid[:, None, None].to(x.dtype)
+        x = x * vm
+        cohs = cohs * valid[None, :, None, None].to(cohs.dtype)
 
     def resnorm(res):
         return float((res[valid].abs() ** 2).sum().sqrt() / (8.0 * nvalid) ** 0.5)

@@ -145,3 +145,55 @@ def test_sagecal_mpi_two_bands(tmp_path):
         assert 'residual' in z.files
         assert np.abs(z['residual']).mean() < \
             0.4 * np.abs(z['data']).mean(), f"band {r}"
+
+
+def test_cli_dochan_and_diag(obs):
+    from sagecal_amd.apps import sagecal as app
+    tmp, skyf, clf, msf = obs
+    rc = app.main(['-d', msf, '-s', skyf, '-c', clf, '-t', '4', '-e', '3',
+                   '-j', '3', '-l', '0', '-b', '1', '-O', 'res_chan'])
+    assert rc == 0
+    z = np.load(msf)
+    assert np.abs(z['res_chan']).mean() < 0.4 * np.abs(z['data']).mean()
+    rc = app.main(['-d', msf, '-s', skyf, '-c', clf, '-t', '4', '-e', '1',
+                   '-j', '3', '-l', '0', '-i', '1', '-O', 'lev'])
+    assert rc == 0
+    z = np.load(msf)
+    assert np.isfinite(z['lev']).all()
+    assert z['lev'].real.max() > 0
+
+
+def test_cli_correct_phase_only(obs):
+    from sagecal_amd.apps import sagecal as app
+    tmp, skyf, clf, msf = obs
+    rc = app.main(['-d', msf, '-s', skyf, '-c', clf, '-t', '4', '-e', '3',
+                   '-j', '3', '-l', '0', '-k', '1', '-J', '1',
+                   '-O', 'res_corr'])
+    assert rc == 0
+    z = np.load(msf)
+    assert np.isfinite(z['res_corr']).all()
+
+
+def test_sagefit_with_flags():
+    """Flagged rows contribute nothing; solution still converges on the
+    valid part."""
+    import torch
+    from sagecal_amd.solvers import sage
+    from sagecal_amd.constants import SM_LM_LBFGS
+    from tests.test_sage import setup_ms
+    ms, pack = setup_ms(M=2, noise=1e-4)
+    tile = ms.load_tile(0)
+    # flag 20% of rows with garbage data
+    rng = np.random.default_rng(0)
+    bad = torch.zeros(tile.x.shape[0], dtype=torch.bool)
+    bad[rng.choice(tile.x.shape[0], tile.x.shape[0] // 5,
+                   replace=False)] = True
+    tile.x[bad] = 1000.0
+    tile.flags = bad
+    bb = ms.bb_tensor()
+    state = sage.CalState(pack, ms.N)
+    cohs = sage.precalc_coherencies(pack, tile)
+    opts = sage.SageSolveOptions(max_emiter=3, max_iter=12,
+                                 solver_mode=SM_LM_LBFGS, joint_iters=4)
+    res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
+    assert res1 < 0.05 * res0
