@@ -44,6 +44,9 @@ def build_argparser():
                     help='1: Barzilai-Borwein adaptive rho')
     ap.add_argument('-U', dest='use_global', type=int, default=0,
                     help='1: residuals from the global solution B Z')
+    ap.add_argument('-X', dest='spatial',
+                    help='L2,L1,order,fista_iters,cadence (spatial reg)')
+    ap.add_argument('-u', dest='spatial_alpha', type=float, default=0.0)
     ap.add_argument('-L', dest='nulow', type=float, default=2.0)
     ap.add_argument('-H', dest='nuhigh', type=float, default=30.0)
     ap.add_argument('-O', dest='outcol', default='residual')
@@ -94,9 +97,19 @@ def main(argv=None):
         dist.all_reduce(f0s)
     freq0_global = float(f0s.mean())
 
+    spatial = None
+    centroids = None
+    if args.spatial:
+        v = args.spatial.split(',')
+        spatial = (float(v[0]), float(v[1]), int(v[2]), int(v[3]),
+                   int(v[4]))
+        centroids = (np.array([c.ll.mean() for c in clusters]),
+                     np.array([c.mm.mean() for c in clusters]))
     adm = ConsensusADMM(state, f0s.tolist(), freq0_global, rank, world,
                         Npoly=min(args.npoly, world), poly_type=args.polytype,
-                        rho=rho, use_bb=bool(args.use_bb))
+                        rho=rho, use_bb=bool(args.use_bb), spatial=spatial,
+                        spatial_alpha=args.spatial_alpha,
+                        centroids=centroids)
     opts = sage.SageSolveOptions(
         max_emiter=args.max_emiter, max_iter=args.max_iter,
         solver_mode=args.solver_mode, robust_nulow=args.nulow,

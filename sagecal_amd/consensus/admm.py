@@ -10,6 +10,7 @@ the per-cluster Npoly x Npoly solve replicated on every rank. Control tags
 disappear (SPMD loop); Zold/Yhat are rank-local; BB rho updates piggyback
 on a tiny all-reduce.
 """
+import numpy as np
 import torch
 import torch.distributed as dist
 
@@ -34,7 +35,8 @@ class ConsensusADMM:
 
     def __init__(self, state, freqs_all, freq0, rank, world, Npoly=2,
                  poly_type=0, rho=None, use_bb=False, rho_upper=1e3,
-                 group=None, federated_alpha=0.0):
+                 group=None, federated_alpha=0.0, spatial=None,
+                 spatial_alpha=0.0, centroids=None):
         self.state = state
         self.rank, self.world = rank, world
         self.group = group
@@ -59,6 +61,21 @@ class ConsensusADMM:
         self.Zold = None
         self.Yhat_prev = None
         self.J_prev = None
+        # spatial regularization (-X / -u, sagecal_master.cpp:293-423 +
+        # :887-985): spatial = (lam, mu, order, fista_iters, cadence)
+        self.spatial = spatial
+        self.spatial_alpha = spatial_alpha
+        self.centroids = centroids          # (ll[M], mm[M])
+        self.Zspat = None
+        self.Xs = torch.zeros_like(self.Z)  # Lagrange multiplier
+        self._Phi = None
+        if spatial is not None and centroids is not None:
+            from . import fista as fista_mod
+            lam, mu_l1, order, fiters, cadence = spatial
+            beta = float(max(np.max(np.abs(centroids[0])),
+                             np.max(np.abs(centroids[1])), 1e-3))
+            self._Phi = fista_mod.spatial_basis(centroids[0], centroids[1],
+                                                order, beta)
 
     def _update_bii(self):
         # rho may differ per band after fratio scaling; here identical
@@ -116,6 +133,10 @@ class ConsensusADMM:
         acc = (Bfd[None, :, None, None, None].to(self.cdtype)
                * contrib[:, None]).contiguous()
         self._allreduce(torch.view_as_real(acc))
+        if self.spatial is not None and self.Zspat is not None:
+            # augmented spatial constraint: z += alpha (Zspat - Xs); the
+            # Bii inverses already carry +alpha I (federated machinery)
+            acc = acc + self.spatial_alpha * (self.Zspat - self.Xs)
         self.Z = poly.update_global_z(acc, self.Bii)
         return self.Z
 
@@ -179,7 +200,34 @@ class ConsensusADMM:
             self.y_update(BZ)
             if self.use_bb and BZ_old is not None:
                 self.bb_update(BZ_old)
+            if self.spatial is not None and self._Phi is not None:
+                lam, mu_l1, order, fiters, cadence = self.spatial
+                if (it + 1) % max(cadence, 1) == 0:
+                    self.spatial_update(lam, mu_l1, fiters)
         return res0, res1
+
+    def spatial_update(self, lam, mu_l1, fiters):
+        """Fit the spatial (elastic-net shapelet) model to the per-cluster
+        Z and form the smoothed constraint Zspat + multiplier update
+        (sagecal_master.cpp:887-985)."""
+        from . import fista as fista_mod
+        M = self.state.M
+        Zb = torch.view_as_real(self.Z).reshape(M, -1)
+        Zb = torch.complex(Zb[:, 0::2], Zb[:, 1::2])   # [M, K]
+        Zsp = fista_mod.update_spatialreg_fista(
+            Zb.cpu(), self._Phi, lam=lam, mu=mu_l1, maxiter=fiters)
+        Zhat = (self._Phi.to(Zsp.dtype) @ Zsp.T)       # [M, K]
+        flat = torch.empty(M, Zhat.shape[1] * 2, dtype=torch.float64)
+        flat[:, 0::2] = Zhat.real
+        flat[:, 1::2] = Zhat.imag
+        self.Zspat = torch.view_as_complex(
+            flat.reshape(*self.Z.shape, 2).contiguous()).to(
+                device=self.dev, dtype=self.cdtype)
+        if self.spatial_alpha > 0:
+            self.Xs = self.Xs + self.spatial_alpha * (self.Z - self.Zspat)
+            # rebuild Bii with the +alpha I term
+            self.alpha = self.spatial_alpha
+            self._update_bii()
 
     def global_solution(self):
         """J = B_f Z (use_global_solution path, sagecal_master:1064)."""

@@ -164,3 +164,24 @@ def test_admm_two_bands_gloo(tmp_path):
         assert res1 < 0.1 * res0, f"rank {rank}: {res0} -> {res1}"
         assert err < 0.05, f"rank {rank} consensus model err {err}"
         assert zdiff < 1e-6, f"Z not replicated: {zdiff}"
+
+
+def test_admm_with_spatial_regularization():
+    """ADMM with the FISTA spatial constraint still converges and produces
+    a spatial model (Zspat)."""
+    from sagecal_amd.consensus.admm import ConsensusADMM
+    from sagecal_amd.solvers import sage
+    from sagecal_amd.constants import SM_LM_LBFGS
+    pack, ms, tile, bb, cohs, Jtrue, freqs_all, f0 = _band_problem(0, 1)
+    state = sage.CalState(pack, ms.N)
+    opts = sage.SageSolveOptions(max_emiter=2, max_iter=10,
+                                 solver_mode=SM_LM_LBFGS, mode='batched')
+    cent = (np.array([0.01, -0.02]), np.array([0.02, 0.01]))
+    adm = ConsensusADMM(state, freqs_all, f0, 0, 1, Npoly=1,
+                        rho=torch.full((pack.M,), 1.0),
+                        spatial=(0.01, 1e-4, 2, 20, 2),
+                        spatial_alpha=0.1, centroids=cent)
+    res0, res1 = adm.run(cohs, tile, bb, opts, n_admm=4)
+    assert res1 < 0.1 * res0
+    assert adm.Zspat is not None
+    assert torch.isfinite(torch.view_as_real(adm.Zspat)).all()
