@@ -125,13 +125,16 @@ def main():
     local_rank = int(os.environ.get('LOCAL_RANK', str(rank)))
     dist = world > 1
     if dist:
-        torch.distributed.init_process_group(
-            backend='nccl' if torch.cuda.is_available() else 'gloo')
+        backend = os.environ.get('SAGECAL_BENCH_BACKEND')
+        if backend is None:
+            backend = 'nccl' if torch.cuda.is_available() else 'gloo'
+        torch.distributed.init_process_group(backend=backend)
 
     use_gpu = torch.cuda.is_available() and not args.cpu
     if use_gpu:
-        torch.cuda.set_device(local_rank)
-        device = f'cuda:{local_rank}'
+        ngpu = torch.cuda.device_count()
+        torch.cuda.set_device(local_rank % ngpu)
+        device = f'cuda:{local_rank % ngpu}'
         dtype = torch.float32
     else:
         device = 'cpu'
