@@ -23,7 +23,7 @@
 
 #define NB 32
 #define PST 36   // padded LDS panel row stride (floats)
-#define NTH 256
+#define NTH 512
 #define MAXN_CHOL 1024
 
 extern "C" __global__ void __launch_bounds__(NTH)
@@ -75,11 +75,11 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     // ~6x fewer load-issue slots than scalar float4 tiles (the previous
     // version was load-issue-bound on a single CU).
     if (k > 0 && stages != 4) {
-      const int w = tid >> 6;              // wave id (4 waves)
+      const int w = tid >> 6;              // wave id (NTH/64 waves)
       const int l15 = lane & 15, l4 = lane >> 4;
       const int ntI = (rows + 15) >> 4;
       const int ntiles = ntI * 2;          // 2 col-tiles of 16
-      for (int tile = w; tile < ntiles; tile += 4) {
+      for (int tile = w; tile < ntiles; tile += NTH / 64) {
         const int I = tile >> 1, Jt = tile & 1;
         const int arow = k + I * 16 + l15;       // S row this lane loads
         const int bcol = k + Jt * 16 + l15;

@@ -69,7 +69,7 @@ hipError_t launch_chol_solve(
     const float* JtJ, const float* Jtr, const float* mu, int n, int batch,
     float* Lbuf, float* dp, int* info, int stages, hipStream_t stream) {
   const size_t shmem = (size_t)(n + 40) * PST * sizeof(float);
-  hipLaunchKernelGGL(k_chol_solve, dim3(batch), dim3(256), shmem, stream,
+  hipLaunchKernelGGL(k_chol_solve, dim3(batch), dim3(512), shmem, stream,
       JtJ, Jtr, mu, n, Lbuf, dp, info, stages);
   return hipGetLastError();
 }
