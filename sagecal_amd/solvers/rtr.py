@@ -31,29 +31,50 @@ def _to_X(J):
     return J.permute(0, 1, 2, 3).reshape(n, 2 * N, 2)
 
 
+def _herm2_eig(S):
+    """Closed-form eigendecomposition of Hermitian 2x2 batch:
+    S = U diag(l1,l2) U^H. Returns (l [n,2], U [n,2,2])."""
+    a = S[:, 0, 0].real
+    d = S[:, 1, 1].real
+    b = S[:, 0, 1]
+    tr = a + d
+    disc = torch.sqrt(((a - d) * 0.5) ** 2 + (b.abs() ** 2)).clamp_min(0)
+    l1 = 0.5 * tr + disc
+    l2 = 0.5 * tr - disc
+    # eigenvector for l1: (b, l1 - a) unless b ~ 0
+    bz = b.abs() < 1e-30
+    v1a = torch.where(bz, torch.ones_like(b), b)
+    v1b = torch.where(bz, torch.zeros_like(b), (l1 - a).to(b.dtype))
+    nrm = torch.sqrt(v1a.abs() ** 2 + v1b.abs() ** 2).clamp_min(1e-30)
+    v1a, v1b = v1a / nrm.to(b.dtype), v1b / nrm.to(b.dtype)
+    # orthogonal second eigenvector
+    v2a = -v1b.conj()
+    v2b = v1a.conj()
+    U = torch.stack([torch.stack([v1a, v2a], -1),
+                     torch.stack([v1b, v2b], -1)], -2)
+    lam = torch.stack([l1, l2], -1)
+    return lam, U
+
+
 def _proj(J, Z):
     """Project ambient direction Z onto the horizontal space at J
-    (fns_proj): Z - J om with (I2 (x) X^H X + (X^H X)^T (x) I2) om =
-    vec(X^H Z - Z^H X). J, Z: [nchunk, N, 2, 2]."""
-    n = J.shape[0]
+    (fns_proj): Z - J Om where Om solves the Sylvester equation
+    (X^H X) Om + Om (X^H X) = X^H Z - Z^H X.
+
+    Solved in CLOSED FORM via the 2x2 Hermitian eigendecomposition of
+    S = X^H X (no batched LAPACK: S = U L U^H => Om' = R'/(l_i + l_j) in
+    the eigenbasis) — the reference's 4x4 zgels (rtr_solve.c:340-410)
+    is equivalent but launch/LAPACK-bound on GPU. J, Z: [n, N, 2, 2]."""
     X = _to_X(J)
     Zt = _to_X(Z)
-    XX = X.conj().transpose(-1, -2) @ X          # [n,2,2]
+    S = X.conj().transpose(-1, -2) @ X           # [n,2,2] Hermitian
     XZ = X.conj().transpose(-1, -2) @ Zt
-    Rr = XZ - XZ.conj().transpose(-1, -2)        # skew part
-    # A om_vec = b with column-major vec (matches reference layout)
-    A = torch.zeros(n, 4, 4, dtype=J.dtype, device=J.device)
-    xx00, xx01 = XX[:, 0, 0], XX[:, 0, 1]
-    xx10, xx11 = XX[:, 1, 0], XX[:, 1, 1]
-    A[:, 0, 0] = 2 * xx00
-    A[:, 1, 1] = A[:, 2, 2] = xx11 + xx00
-    A[:, 3, 3] = 2 * xx11
-    A[:, 0, 1] = A[:, 2, 0] = A[:, 2, 3] = A[:, 3, 1] = xx01
-    A[:, 0, 2] = A[:, 1, 0] = A[:, 1, 3] = A[:, 3, 2] = xx10
-    b = torch.stack([Rr[:, 0, 0], Rr[:, 1, 0], Rr[:, 0, 1], Rr[:, 1, 1]],
-                    dim=1)
-    om = torch.linalg.lstsq(A, b.unsqueeze(-1)).solution.squeeze(-1)
-    Om = om.reshape(n, 2, 2).transpose(-1, -2)   # col-major -> matrix
+    Rr = XZ - XZ.conj().transpose(-1, -2)        # skew-Hermitian RHS
+    lam, U = _herm2_eig(S)
+    Rp = U.conj().transpose(-1, -2) @ Rr @ U
+    denom = (lam[:, :, None] + lam[:, None, :]).clamp_min(1e-30)
+    Omp = Rp / denom.to(Rp.dtype)
+    Om = U @ Omp @ U.conj().transpose(-1, -2)
     return Z - torch.einsum('nsij,njk->nsik', J, Om)
 
 
@@ -111,6 +132,9 @@ def _tcg(obj, J, grad, delta, maxit=30, kappa=0.01, theta=1.0):
     d = -r
     r0 = _inner(r, r)
     rnorm0 = r0.sqrt()
+    done = torch.zeros(n, dtype=torch.bool, device=J.device)
+    tol = rnorm0 * torch.minimum(rnorm0 ** theta,
+                                 torch.full_like(rnorm0, kappa))
     for it in range(maxit):
         Hd = obj.hess_vec(J, d)
         dHd = _inner(d, Hd)
@@ -119,27 +143,26 @@ def _tcg(obj, J, grad, delta, maxit=30, kappa=0.01, theta=1.0):
         enorm = _inner(eta_new, eta_new).sqrt()
         neg = dHd <= 0
         big = enorm > delta
-        if bool((neg | big).any()):
-            # to the boundary along d for those chunks
-            ee = _inner(eta, eta)
-            ed = _inner(eta, d)
-            dd = _inner(d, d)
-            tau = (-ed + (ed ** 2 + dd * (delta ** 2 - ee)).clamp_min(0)
-                   .sqrt()) / dd.clamp_min(1e-30)
-            bnd = eta + tau[:, None, None, None] * d
-            eta_new = torch.where((neg | big)[:, None, None, None], bnd,
-                                  eta_new)
-            eta = eta_new
-            break
-        eta = eta_new
+        hit = (neg | big) & ~done
+        # boundary step along d for chunks that hit
+        ee = _inner(eta, eta)
+        ed = _inner(eta, d)
+        dd = _inner(d, d)
+        tau = (-ed + (ed ** 2 + dd * (delta ** 2 - ee)).clamp_min(0)
+               .sqrt()) / dd.clamp_min(1e-30)
+        bnd = eta + tau[:, None, None, None] * d
+        eta_new = torch.where(hit[:, None, None, None], bnd, eta_new)
+        eta = torch.where(done[:, None, None, None], eta, eta_new)
+        done = done | hit
         r = r + alpha[:, None, None, None] * Hd
         rn = _inner(r, r)
-        if bool((rn.sqrt() <= rnorm0 * torch.minimum(
-                rnorm0 ** theta, torch.full_like(rnorm0, kappa))).all()):
-            break
+        done = done | (rn.sqrt() <= tol)
         beta = rn / r0.clamp_min(1e-30)
         d = -r + beta[:, None, None, None] * d
         r0 = rn
+        # poll every 8 iters only (host-sync hygiene)
+        if (it & 7) == 7 and bool(done.all()):
+            break
     return eta
 
 
@@ -195,12 +218,12 @@ def rtr_solve(prob, J0, maxiter=20, rsd_iters=2, delta0=None,
         J = torch.where(accept[:, None, None, None], Jn, J)
         cost = torch.where(accept, cn, cost)
         delta = torch.where(rho < 0.25, delta * 0.25,
-                            torch.where((rho > 0.75), (delta * 2.0)
-                                        .clamp(max=float(delta_bar.max())),
-                                        delta))
+                            torch.where(rho > 0.75,
+                                        torch.minimum(delta * 2.0,
+                                                      delta_bar), delta))
         obj.refresh(J)
         g = obj.grad(J)
-        if float(_inner(g, g).sqrt().max()) < 1e-9:
+        if (it & 3) == 3 and float(_inner(g, g).sqrt().max()) < 1e-9:
             break
     return J, {'init_cost': init_cost, 'final_cost': cost,
                'niter': maxiter}
