@@ -106,6 +106,8 @@ def run_bandpass(args, device, dtype, rank, world):
                                          seed=17)
     clusters = sky.build_clusters(srcs, clist, 0.0, np.pi / 4, args.freq0)
     pack = SourcePack(clusters)
+    if device != 'cpu':
+        pack.to(device)
     cdt = torch.complex64 if dtype == torch.float32 else torch.complex128
     ms = msdata.SyntheticMS(
         N=args.stations, tilesz=args.tilesz, Ntime=args.tilesz,
