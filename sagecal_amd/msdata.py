@@ -152,7 +152,9 @@ class SyntheticMS:
             g = (rng.standard_normal((self.N, 2, 2))
                  + 1j * rng.standard_normal((self.N, 2, 2)))
             J[ci] = np.eye(2)[None] + self.gain_amp * g
-        return torch.tensor(J, device=self.device)
+        cdt = (torch.complex128 if self.dtype == torch.float64
+               else torch.complex64)
+        return torch.tensor(J, device=self.device, dtype=cdt)
 
     def tiles(self):
         for ti in range(max(1, self.Ntime // self.tilesz)):
