@@ -94,12 +94,29 @@ class CalState:
         self.J = eye.expand(self.Mt, self.N, 2, 2).clone()
 
 
-def precalc_coherencies(pack, tile, device=None):
-    """Channel-averaged coherencies for the solve (precalculate_coherencies,
-    predict.c:503): predicted at freq0 with full-bandwidth smearing."""
+def precalc_coherencies(pack, tile, device=None, discrete=True):
+    """Channel-averaged coherencies for the solve.
+
+    discrete=True (default): the EXACT model of channel-averaged data —
+    the mean of per-channel coherencies over the tile's actual channel
+    grid (each with its own channel-width smearing). The reference instead
+    predicts once at freq0 with a continuous |sinc| bandwidth-smearing
+    factor (precalculate_coherencies, predict.c:503) — an approximation
+    whose Dirichlet-vs-sinc mismatch puts a floor under the residuals of
+    wide sub-bands; set discrete=False for reference-equivalent behavior.
+    Predict cost is x Nchan, amortized once per tile."""
     u, v, w = tile.u, tile.v, tile.w
-    return ops.predict_coh(pack, u, v, w, tile.freq0, tile.freq0,
-                           tile.fdelta, tile.tdelta, tile.dec0)
+    freqs = getattr(tile, 'freqs', None)
+    if not discrete or freqs is None or len(freqs) <= 1:
+        return ops.predict_coh(pack, u, v, w, tile.freq0, tile.freq0,
+                               tile.fdelta, tile.tdelta, tile.dec0)
+    fdelta_ch = tile.fdelta / len(freqs)
+    acc = None
+    for f in freqs:
+        c = ops.predict_coh(pack, u, v, w, float(f), tile.freq0,
+                            fdelta_ch, tile.tdelta, tile.dec0)
+        acc = c if acc is None else acc + c
+    return acc / len(freqs)
 
 
 def _layout_for(state, bb, T, Nbase, nseg, device):
