@@ -409,3 +409,87 @@ def _lm_solve_graphed(prob, J0, maxiter, tau, eps1, eps2):
     J, cost, init_cost = ent['outs']
     return J.clone(), {'init_cost': init_cost.clone(),
                        'final_cost': cost.clone(), 'niter': maxiter}
+
+
+_robust_graph_cache = {}
+
+
+def _robust_body(x, coh, bb, N, nchunk, chunk_rows, layout, J0, nu_t,
+                 inner, tau, prob_admm):
+    """Capture-safe robust IRLS body (robust_outer=1 schedule):
+    solve -> Student's-t weight update (stale nu, refreshed on the host
+    after replay) -> weighted solve. Returns (J, w, e2sum)."""
+    class _P:
+        pass
+    p1 = _P(); p1.admm = prob_admm
+    J1, _, init_cost = _lm_body(x, coh, bb, N, nchunk, chunk_rows, None,
+                                layout, J0, inner, tau, p1)
+    from ..ops import dispatch as ops_d
+    V = ops_d.apply_jones(coh, J1, bb, chunk_rows, layout)
+    r = x - V
+    e2 = (r.abs() ** 2).sum(dim=(-1, -2)).to(torch.float32)
+    w = (nu_t + 8.0) / (nu_t + e2)
+    J2, cost, _ = _lm_body(x, coh, bb, N, nchunk, chunk_rows, w, layout,
+                           J1, inner, tau, p1)
+    logsumw = (torch.log(w) - w).mean()
+    return J2, w, cost, init_cost, logsumw
+
+
+def robust_lm_graphed(prob, J0, nu0, inner, tau=1e-3):
+    """hipGraph-captured robust LM (one replay for the whole IRLS
+    schedule). The nu AECM update runs on the host AFTER the replay from
+    the captured mean(log w - w) (one sync per group-solve)."""
+    x, coh, bb, N = prob.x, prob.coh, prob.bb, prob.N
+    nchunk = prob.nchunk
+    dev = x.device
+    B = x.shape[0]
+    key = ('robust', B, nchunk, N, inner, id(prob.layout),
+           prob.admm is not None)
+    ent = _robust_graph_cache.get(key)
+    if ent is None:
+        ent = {'x': torch.empty_like(x), 'coh': torch.empty_like(coh),
+               'J0': torch.empty_like(J0),
+               'rows': torch.zeros(B, dtype=torch.long, device=dev),
+               'nu': torch.zeros((), dtype=torch.float32, device=dev),
+               'bb': bb}
+        ent['admm'] = None
+        if prob.admm is not None:
+            ent['admm'] = tuple(torch.empty_like(t) for t in prob.admm)
+        ent['x'].copy_(x); ent['coh'].copy_(coh); ent['J0'].copy_(J0)
+        if prob.chunk_rows is not None:
+            ent['rows'].copy_(prob.chunk_rows)
+        ent['nu'].fill_(nu0)
+        if prob.admm is not None:
+            for d, sr in zip(ent['admm'], prob.admm):
+                d.copy_(sr)
+        st = torch.cuda.Stream()
+        st.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(st):
+            for _ in range(2):
+                _robust_body(ent['x'], ent['coh'], ent['bb'], N, nchunk,
+                             ent['rows'], prob.layout, ent['J0'],
+                             ent['nu'], inner, tau, ent['admm'])
+        torch.cuda.current_stream().wait_stream(st)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            outs = _robust_body(ent['x'], ent['coh'], ent['bb'], N, nchunk,
+                                ent['rows'], prob.layout, ent['J0'],
+                                ent['nu'], inner, tau, ent['admm'])
+        ent['graph'] = g
+        ent['outs'] = outs
+        _robust_graph_cache[key] = ent
+    ent['x'].copy_(x)
+    ent['coh'].copy_(coh)
+    ent['J0'].copy_(J0)
+    if prob.chunk_rows is not None:
+        ent['rows'].copy_(prob.chunk_rows)
+    else:
+        ent['rows'].zero_()
+    ent['nu'].fill_(nu0)
+    if prob.admm is not None:
+        for d, sr in zip(ent['admm'], prob.admm):
+            d.copy_(sr)
+    ent['graph'].replay()
+    J, w, cost, init_cost, logsumw = ent['outs']
+    return (J.clone(), float(logsumw),
+            {'init_cost': init_cost.clone(), 'final_cost': cost.clone()})

@@ -193,9 +193,26 @@ def robust_lm(prob, J0, nu0, opts):
     """IRLS Student's-t wrapper (robustlm.c rlevmar_der_single_* /
     rtr_solve_robust.c): alternate the weighted inner solver with weight +
     nu AECM updates."""
+    import os as _os
     nu = nu0 if nu0 > 0 else 2.0
-    J = J0
     inner = max(3, opts.max_iter // max(1, opts.robust_outer))
+    if (opts.robust_outer == 1 and prob.layout is not None
+            and prob.x.is_cuda and opts.solver_mode != 5 or
+            (opts.robust_outer == 1 and prob.layout is not None
+             and prob.x.is_cuda and prob.N <= 40))             and _os.environ.get('SAGECAL_NO_GRAPH') != '1':
+        # single-IRLS-pass schedule: whole solve as one graph replay; the
+        # nu grid update runs on the host from the captured mean(log w - w)
+        J, logsumw, _ = lm_mod.robust_lm_graphed(prob, J0, nu, inner)
+        import math as _math
+        dgm = float(torch.special.digamma(torch.tensor((nu + 8) * 0.5)))
+        dgm -= _math.log((nu + 8) * 0.5)
+        grid = torch.linspace(opts.robust_nulow, opts.robust_nuhigh,
+                              NU_GRID)
+        q = (-torch.special.digamma(grid * 0.5) + torch.log(grid * 0.5)
+             + logsumw + dgm + 1.0)
+        nu = float(grid[torch.argmin(q.abs())])
+        return J, nu
+    J = J0
     for outer in range(opts.robust_outer):
         J = _inner_solve(prob, J, opts, inner)
         V = ops.apply_jones(prob.coh, J, prob.bb, prob.chunk_rows,
