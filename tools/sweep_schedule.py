@@ -19,11 +19,12 @@ def main():
 
     configs = [
         # (emiter, maxiter, robust_outer, em_group, label)
-        (3, 8, 1, 3, 'lean-g3'),
-        (3, 8, 1, 5, 'lean-g5'),
-        (3, 8, 1, 10, 'lean-g10'),
         (4, 8, 1, 5, 'em4-g5'),
-        (3, 10, 1, 5, 'it10-g5'),
+        (3, 8, 1, 5, 'em3-g5'),
+        (5, 8, 1, 5, 'em5-g5'),
+        (4, 6, 1, 5, 'em4-it6'),
+        (6, 6, 1, 5, 'em6-it6'),
+        (4, 8, 1, 10, 'em4-g10'),
     ]
     for emiter, maxiter, ro, eg, label in configs:
         opts = sage.SageSolveOptions(
