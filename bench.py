@@ -139,7 +139,7 @@ def main():
     ap.add_argument('--freq0', type=float, default=150e6)
     ap.add_argument('--bandwidth', type=float, default=180e3)
     ap.add_argument('--emiter', type=int, default=4)
-    ap.add_argument('--maxiter', type=int, default=8)
+    ap.add_argument('--maxiter', type=int, default=6)
     ap.add_argument('--robust-outer', type=int, default=1)
     ap.add_argument('--joint', type=int, default=0)
     ap.add_argument('--em-group', type=int, default=5)
