@@ -156,39 +156,6 @@ def total_model(state, cohs, bb, T, Nbase):
     return V
 
 
-def _solve_cluster(state, ci, xsub, coh_ci, bb, rows, opts):
-    """One cluster's (robust) LM solve over its chunks."""
-    nchunk = state.nchunks[ci]
-    N = state.N
-    prob = lm_mod.LMProblem(xsub, coh_ci, bb, N, nchunk, rows)
-    J0 = state.cluster_J(ci)
-    if opts.robust:
-        J, nu_new = robust_lm(prob, J0, float(state.nu[ci]), opts)
-        state.nu[ci] = nu_new
-    else:
-        if opts.nsubsets > 1:
-            J, _ = lm_mod.os_lm_solve(prob, J0, maxiter=opts.max_iter,
-                                      nsubsets=opts.nsubsets)
-        else:
-            J, _ = lm_mod.lm_solve(prob, J0, maxiter=opts.max_iter)
-    return J
-
-
-def _inner_solve(prob, J, opts, maxiter):
-    """Dispatch one deterministic inner solve by solver mode, honoring the
-    reference's LMCUT heuristic (sagecalmain.h:24: N <= 40 -> RTR/NSD
-    replaced by LM)."""
-    mode = opts.solver_mode
-    if prob.N > LMCUT and mode in (SM_RTR_OSLM_LBFGS,
-                                   SM_RTR_OSRLM_RLBFGS):
-        from . import rtr as rtr_mod
-        return rtr_mod.rtr_solve(prob, J, maxiter=maxiter)[0]
-    if prob.N > LMCUT and mode == SM_NSD_RLBFGS:
-        from . import rtr as rtr_mod
-        return rtr_mod.nsd_solve(prob, J, maxiter=2 * maxiter)[0]
-    return lm_mod.lm_solve(prob, J, maxiter=maxiter)[0]
-
-
 def robust_lm(prob, J0, nu0, opts):
     """IRLS Student's-t wrapper (robustlm.c rlevmar_der_single_* /
     rtr_solve_robust.c): alternate the weighted inner solver with weight +

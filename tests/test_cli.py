@@ -197,3 +197,27 @@ def test_sagefit_with_flags():
                                  solver_mode=SM_LM_LBFGS, joint_iters=4)
     res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
     assert res1 < 0.05 * res0
+
+
+def test_npz_multi_tile_and_uvwriter(tmp_path):
+    """Multi-tile observation: per-tile solutions blocks; uvwriter runs."""
+    from sagecal_amd.apps import sagecal as app, uvwriter
+    (tmp_path / 'sky.txt').write_text(SKY)
+    (tmp_path / 'cl.txt').write_text(CLUSTER)
+    clusters = sky.read_sky_cluster(str(tmp_path / 'sky.txt'),
+                                    str(tmp_path / 'cl.txt'),
+                                    0.0, np.pi / 4, 150e6)
+    pack = SourcePack(clusters)
+    msf = str(tmp_path / 'obs2.npz')
+    msdata.make_synthetic_npz(msf, N=8, tilesz=3, Ntime=6, Nchan=2,
+                              pack=pack, bandwidth=50e3, noise_sigma=1e-3,
+                              seed=8, ra0=0.0, dec0=np.pi / 4)
+    sol = str(tmp_path / 'sol2.txt')
+    rc = app.main(['-d', msf, '-s', str(tmp_path / 'sky.txt'),
+                   '-c', str(tmp_path / 'cl.txt'), '-p', sol, '-t', '3',
+                   '-e', '5', '-j', '3', '-l', '0'])
+    assert rc == 0
+    hdr, tiles = solutions.read_solutions(sol)
+    assert len(tiles) == 2          # two solution intervals
+    rc = uvwriter.main(['-d', msf])
+    assert rc == 0
