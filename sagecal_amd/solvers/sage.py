@@ -156,6 +156,21 @@ def total_model(state, cohs, bb, T, Nbase):
     return V
 
 
+def _inner_solve(prob, J, opts, maxiter):
+    """Dispatch one deterministic inner solve by solver mode, honoring the
+    reference's LMCUT heuristic (sagecalmain.h:24: N <= 40 -> RTR/NSD
+    replaced by LM)."""
+    mode = opts.solver_mode
+    if prob.N > LMCUT and mode in (SM_RTR_OSLM_LBFGS,
+                                   SM_RTR_OSRLM_RLBFGS):
+        from . import rtr as rtr_mod
+        return rtr_mod.rtr_solve(prob, J, maxiter=maxiter)[0]
+    if prob.N > LMCUT and mode == SM_NSD_RLBFGS:
+        from . import rtr as rtr_mod
+        return rtr_mod.nsd_solve(prob, J, maxiter=2 * maxiter)[0]
+    return lm_mod.lm_solve(prob, J, maxiter=maxiter)[0]
+
+
 def robust_lm(prob, J0, nu0, opts):
     """IRLS Student's-t wrapper (robustlm.c rlevmar_der_single_* /
     rtr_solve_robust.c): alternate the weighted inner solver with weight +
