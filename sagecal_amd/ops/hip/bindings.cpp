@@ -13,7 +13,7 @@ hipError_t launch_predict_coh(const double*, const double*, const double*,
     hipStream_t);
 hipError_t launch_jtj_accum(const float2*, const float2*, const float2*,
     const int*, const int*, const float*, int, int, int, int, float2*,
-    float2*, float2*, float*, int, hipStream_t);
+    float2*, float2*, float*, int, int, hipStream_t);
 hipError_t launch_jtj_expand(const float2*, const float2*, const float2*,
     const int*, int, int, int, float*, float*, hipStream_t);
 hipError_t launch_model_cost(const float2*, const float2*, const float2*,
@@ -80,7 +80,7 @@ std::vector<torch::Tensor> jtj_jtr(
   CHECK_HIP(launch_jtj_accum(ccptr(x), ccptr(coh), ccptr(J),
       pairs.data_ptr<int>(), chunk_tab.data_ptr<int>(), wp,
       (int)Nbase, (int)T, (int)N, (int)nseg, cptr(D), cptr(g), cptr(Cx),
-      cost.data_ptr<float>(), (int)npair, cur_stream()));
+      cost.data_ptr<float>(), (int)npair, 0, cur_stream()));
   auto JtJ = torch::empty({Mt, 8 * N, 8 * N}, fopts);
   auto Jtr = torch::empty({Mt, 8 * N}, fopts);
   CHECK_HIP(launch_jtj_expand(ccptr(D), ccptr(g), ccptr(Cx),
@@ -134,7 +134,27 @@ std::vector<torch::Tensor> chol_solve(
   return {dp, info};
 }
 
+std::vector<torch::Tensor> jtr_grad(
+    torch::Tensor x, torch::Tensor coh, torch::Tensor J,
+    torch::Tensor pairs, torch::Tensor chunk_tab,
+    c10::optional<torch::Tensor> wts, int64_t Nbase, int64_t T, int64_t N,
+    int64_t nseg, int64_t Mt) {
+  // gradient-only accumulation (LBFGS path): returns (g [Mt*N,4] c64,
+  // cost [Mt]); g's interleaved memory IS vecR order.
+  auto fopts = torch::dtype(torch::kFloat).device(x.device());
+  auto copts = torch::dtype(torch::kComplexFloat).device(x.device());
+  auto g = torch::zeros({Mt * N, 4}, copts);
+  auto cost = torch::zeros({Mt}, fopts);
+  const float* wp = wts.has_value() ? wts->data_ptr<float>() : nullptr;
+  CHECK_HIP(launch_jtj_accum(ccptr(x), ccptr(coh), ccptr(J),
+      pairs.data_ptr<int>(), chunk_tab.data_ptr<int>(), wp,
+      (int)Nbase, (int)T, (int)N, (int)nseg, nullptr, cptr(g), nullptr,
+      cost.data_ptr<float>(), (int)Nbase, 1, cur_stream()));
+  return {g, cost};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("jtr_grad", &jtr_grad, "gradient-only accumulation (gfx950)");
   m.def("chol_solve", &chol_solve, "batched damped Cholesky solve (gfx950)");
   m.def("predict_coh", &predict_coh, "coherency predict (gfx950)");
   m.def("jtj_jtr", &jtj_jtr, "fused JtJ/Jtr assembly (gfx950)");

@@ -34,7 +34,7 @@ k_jtj_accum(const float2* __restrict__ x, const float2* __restrict__ coh,
             int Nbase, int T, int N,
             float2* __restrict__ D, float2* __restrict__ g,
             float2* __restrict__ Cx, float* __restrict__ cost,
-            int npair_slots) {
+            int npair_slots, int grad_only) {
   const int b = blockIdx.x;        // pair index
   const int seg = blockIdx.y;
   const int lane = threadIdx.x;
@@ -71,14 +71,16 @@ k_jtj_accum(const float2* __restrict__ x, const float2* __restrict__ coh,
 #pragma unroll
       for (int i = 0; i < 4; ++i) res[i] = csub(X[i], V[i]);
       cst += wt * (cabs2(res[0]) + cabs2(res[1]) + cabs2(res[2]) + cabs2(res[3]));
-      // diag: Ap += w*conj(G1 G1^H); Aq += w*conj(K^H K)
       cf t4[4];
-      m2mulh(G1, G1, t4);
+      if (!grad_only) {
+        // diag: Ap += w*conj(G1 G1^H); Aq += w*conj(K^H K)
+        m2mulh(G1, G1, t4);
 #pragma unroll
-      for (int i = 0; i < 4; ++i) Ap[i] = cadd(Ap[i], cscale(conjf2(t4[i]), wt));
-      m2hmul(K, K, t4);
+        for (int i = 0; i < 4; ++i) Ap[i] = cadd(Ap[i], cscale(conjf2(t4[i]), wt));
+        m2hmul(K, K, t4);
 #pragma unroll
-      for (int i = 0; i < 4; ++i) Aq[i] = cadd(Aq[i], cscale(conjf2(t4[i]), wt));
+        for (int i = 0; i < 4; ++i) Aq[i] = cadd(Aq[i], cscale(conjf2(t4[i]), wt));
+      }
       // grads: gp += w*(res G1^H); gq += w*(res^H K)
       m2mulh(res, G1, t4);
 #pragma unroll
@@ -87,6 +89,7 @@ k_jtj_accum(const float2* __restrict__ x, const float2* __restrict__ coh,
 #pragma unroll
       for (int i = 0; i < 4; ++i) gq[i] = cadd(gq[i], cscale(t4[i], wt));
       // cross: Xc[(i,j),(k,b2)] += w * conj(G1[j,k]) * K[i,b2]
+      if (grad_only) continue;
 #pragma unroll
       for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -103,31 +106,41 @@ k_jtj_accum(const float2* __restrict__ x, const float2* __restrict__ coh,
     // wave reduction of all partials
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      Ap[i].x = wave_sum(Ap[i].x); Ap[i].y = wave_sum(Ap[i].y);
-      Aq[i].x = wave_sum(Aq[i].x); Aq[i].y = wave_sum(Aq[i].y);
       gp[i].x = wave_sum(gp[i].x); gp[i].y = wave_sum(gp[i].y);
       gq[i].x = wave_sum(gq[i].x); gq[i].y = wave_sum(gq[i].y);
     }
+    if (!grad_only) {
 #pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      Xc[i].x = wave_sum(Xc[i].x); Xc[i].y = wave_sum(Xc[i].y);
+      for (int i = 0; i < 4; ++i) {
+        Ap[i].x = wave_sum(Ap[i].x); Ap[i].y = wave_sum(Ap[i].y);
+        Aq[i].x = wave_sum(Aq[i].x); Aq[i].y = wave_sum(Aq[i].y);
+      }
+#pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        Xc[i].x = wave_sum(Xc[i].x); Xc[i].y = wave_sum(Xc[i].y);
+      }
     }
     cst = wave_sum(cst);
     if (lane == 0) {
-      float2* Dp = D + ((size_t)c * N + p) * 4;
-      float2* Dq = D + ((size_t)c * N + q) * 4;
       float2* Gp = g + ((size_t)c * N + p) * 4;
       float2* Gq = g + ((size_t)c * N + q) * 4;
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        atomicAddCf(Dp + i, Ap[i]);
-        atomicAddCf(Dq + i, Aq[i]);
         atomicAddCf(Gp + i, gp[i]);
         atomicAddCf(Gq + i, gq[i]);
       }
-      float2* Xo = Cx + ((size_t)c * npair_slots + b) * 16;
+      if (!grad_only) {
+        float2* Dp = D + ((size_t)c * N + p) * 4;
+        float2* Dq = D + ((size_t)c * N + q) * 4;
 #pragma unroll
-      for (int i = 0; i < 16; ++i) Xo[i] = Xc[i];   // unique writer
+        for (int i = 0; i < 4; ++i) {
+          atomicAddCf(Dp + i, Ap[i]);
+          atomicAddCf(Dq + i, Aq[i]);
+        }
+        float2* Xo = Cx + ((size_t)c * npair_slots + b) * 16;
+#pragma unroll
+        for (int i = 0; i < 16; ++i) Xo[i] = Xc[i];   // unique writer
+      }
       atomicAdd(&cost[c], cst);
     }
     t0 = t1;

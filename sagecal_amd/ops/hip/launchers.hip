@@ -27,9 +27,10 @@ hipError_t launch_jtj_accum(
     const float2* x, const float2* coh, const float2* J, const int* pairs,
     const int* chunk_tab, const float* wts, int Nbase, int T, int N,
     int nseg, float2* D, float2* g, float2* Cx, float* cost, int npair,
-    hipStream_t stream) {
+    int grad_only, hipStream_t stream) {
   hipLaunchKernelGGL(k_jtj_accum, dim3(Nbase, nseg), dim3(64), 0, stream,
-      x, coh, J, pairs, chunk_tab, wts, Nbase, T, N, D, g, Cx, cost, npair);
+      x, coh, J, pairs, chunk_tab, wts, Nbase, T, N, D, g, Cx, cost, npair,
+      grad_only);
   return hipGetLastError();
 }
 

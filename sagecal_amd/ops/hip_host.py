@@ -271,10 +271,56 @@ def residual_total(x, cohs, J, chunk_tabs, layout):
 
 def lbfgs_cost_grad(x, cohs, J_packed, chunk_off, nchunks, bb, T, Nbase,
                     robust_nu=None, weights=None):
-    # torch-on-GPU path (cold op; HIP kernel in a later wave)
+    """Full-parameter cost/gradient on GPU via the grad-only accumulation
+    kernel: r = x - sum_ci V_ci (fused residual kernel), then per cluster
+    Jtr with x := r + V_ci (so the kernel's internal residual IS the total
+    residual); robust scale folded as per-baseline weights. Falls back to
+    the torch path when the row structure is absent."""
     from . import reference as R
-    return R.lbfgs_cost_grad(x, cohs, J_packed, chunk_off, nchunks, bb, T,
-                             Nbase, robust_nu, weights)
+    B = x.shape[0]
+    M = cohs.shape[0]
+    N = J_packed.shape[1]
+    Mt = J_packed.shape[0]
+    dev = x.device
+    if B % Nbase != 0 or (B // Nbase) != T:
+        return R.lbfgs_cost_grad(x, cohs, J_packed, chunk_off, nchunks,
+                                 bb, T, Nbase, robust_nu, weights)
+    lay = BaselineLayout(bb, Nbase, T, 1, N, dev)
+    ext = _ext()
+    Jc = J_packed.to(torch.complex64).reshape(-1, 4).contiguous()
+    # per-cluster chunk tables (global chunk ids)
+    cts = []
+    Vs = []
+    Vtot = torch.zeros(B, 4, dtype=torch.complex64, device=dev)
+    cohs4 = cohs.to(torch.complex64).reshape(M, B, 4)
+    for ci in range(M):
+        rows = R.chunk_rows_for(ci, nchunks, T, Nbase, B, dev)
+        if rows is None:
+            rows = torch.zeros(B, dtype=torch.long, device=dev)
+        rows = rows + chunk_off[ci]
+        ct = rows[::Nbase].to(torch.int32).contiguous()
+        cts.append(ct)
+        V = ext.apply_jones(None, cohs4[ci].contiguous(), Jc, lay.pairs,
+                            ct, Nbase, T, N, 1, 1, 0)
+        Vs.append(V)
+        Vtot += V
+    x4 = x.to(torch.complex64).reshape(B, 4)
+    rtot = x4 - Vtot
+    e2 = (rtot.abs() ** 2).sum(dim=1)
+    if robust_nu is not None:
+        cost = torch.log1p(e2 / robust_nu).sum()
+        scale = (1.0 / (robust_nu + e2)).to(torch.float32).contiguous()
+    else:
+        cost = e2.sum()
+        scale = None
+    grad = torch.zeros(Mt * N, 4, dtype=torch.complex64, device=dev)
+    for ci in range(M):
+        xsub = (rtot + Vs[ci]).contiguous()
+        gci, _ = ext.jtr_grad(xsub, cohs4[ci].contiguous(), Jc, lay.pairs,
+                              cts[ci], scale, Nbase, T, N, 1, Mt)
+        grad += gci
+    g = -2.0 * torch.view_as_real(grad).reshape(-1)
+    return cost.to(torch.float32), g
 
 
 _chol_scratch = {}
