@@ -214,24 +214,41 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts,
     """Solve a group of clusters as ONE batched LM problem (block-diagonal
     across clusters via the chunk axis), then update the running residual
     incrementally: res += sum(V_old - V_new) over the group."""
-    xs, cs, bbs, rows_all, Vold = [], [], [], [], []
-    chunk_counts = []
     dev = res.device
-    for ci in group:
-        Vc, rows = _model_cluster(state, ci, cohs[ci], bb, T, Nbase, B)
-        Vold.append(Vc)
-        xs.append(res + Vc)
-        cs.append(cohs[ci])
-        bbs.append(bb)
-        r = rows if rows is not None else torch.zeros(B, dtype=torch.long,
-                                                      device=dev)
-        rows_all.append(r + sum(chunk_counts))
-        chunk_counts.append(state.nchunks[ci])
+    # static per-(group, cohs) concatenations cached across EM sweeps:
+    # the cluster coherencies, pair table and chunk rows never change
+    # within a tile — only xsub does
+    cache = getattr(state, '_grp_cache', None)
+    if cache is None:
+        cache = state._grp_cache = {}
+    key = (tuple(group), id(cohs), B)
+    ent = cache.get(key)
+    if ent is None:
+        cs, bbs, rows_all, chunk_counts = [], [], [], []
+        for ci in group:
+            rows = R.chunk_rows_for(ci, state.nchunks, T, Nbase, B, dev)
+            r = rows if rows is not None else torch.zeros(
+                B, dtype=torch.long, device=dev)
+            cs.append(cohs[ci])
+            bbs.append(bb)
+            rows_all.append(r + sum(chunk_counts))
+            chunk_counts.append(state.nchunks[ci])
+        ent = {
+            'ccat': torch.cat(cs), 'bbcat': torch.cat(bbs),
+            'rcat': torch.cat(rows_all), 'counts': chunk_counts,
+            'xbuf': torch.empty(len(group) * B, 2, 2, dtype=res.dtype,
+                                device=dev),
+        }
+        cache[key] = ent
+    chunk_counts = ent['counts']
     nch_tot = sum(chunk_counts)
-    xcat = torch.cat(xs)
-    ccat = torch.cat(cs)
-    bbcat = torch.cat(bbs)
-    rcat = torch.cat(rows_all)
+    ccat, bbcat, rcat = ent['ccat'], ent['bbcat'], ent['rcat']
+    Vold = []
+    xcat = ent['xbuf']
+    for gi, ci in enumerate(group):
+        Vc, _ = _model_cluster(state, ci, cohs[ci], bb, T, Nbase, B)
+        Vold.append(Vc)
+        torch.add(res, Vc, out=xcat[gi * B:(gi + 1) * B])
     J0 = torch.cat([state.cluster_J(ci) for ci in group])
     lay = _layout_for(state, bb, T, Nbase, len(group), xcat.device)
     admm = None
