@@ -221,7 +221,14 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts,
     cache = getattr(state, '_grp_cache', None)
     if cache is None:
         cache = state._grp_cache = {}
-    key = (tuple(group), id(cohs), B)
+    # token-based key: id() is recycled by the allocator across tiles
+    tok = getattr(cohs, '_sagecal_token', None)
+    if tok is None:
+        tok = getattr(sagefit, '_next_token', 0)
+        sagefit._next_token = tok + 1
+        cohs._sagecal_token = tok
+        cache.clear()          # new tile: drop stale group buffers
+    key = (tuple(group), tok, B)
     ent = cache.get(key)
     if ent is None:
         cs, bbs, rows_all, chunk_counts = [], [], [], []
