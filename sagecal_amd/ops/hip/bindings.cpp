@@ -22,6 +22,8 @@ hipError_t launch_model_cost(const float2*, const float2*, const float2*,
 hipError_t launch_apply_jones(const float2*, const float2*, const float2*,
     const int*, const int*, int, int, int, int, int, int, float2*,
     hipStream_t);
+hipError_t launch_chol_mw(const float*, const float*, const float*, int,
+                          int, float*, float*, int*, hipStream_t);
 hipError_t launch_chol_solve(const float*, const float*, const float*, int,
     int, float*, float*, int*, int, hipStream_t);
 }
@@ -134,6 +136,22 @@ std::vector<torch::Tensor> chol_solve(
   return {dp, info};
 }
 
+std::vector<torch::Tensor> chol_solve_mw(
+    torch::Tensor JtJ, torch::Tensor Jtr, torch::Tensor mu,
+    torch::Tensor scratch) {
+  // multi-workgroup right-looking path: same contract as chol_solve
+  const int64_t batch = JtJ.size(0);
+  const int64_t n = JtJ.size(1);
+  auto dp = torch::empty({batch, n},
+      torch::dtype(torch::kFloat).device(JtJ.device()));
+  auto info = torch::zeros({batch},
+      torch::dtype(torch::kInt).device(JtJ.device()));
+  CHECK_HIP(launch_chol_mw(JtJ.data_ptr<float>(), Jtr.data_ptr<float>(),
+      mu.data_ptr<float>(), (int)n, (int)batch, scratch.data_ptr<float>(),
+      dp.data_ptr<float>(), info.data_ptr<int>(), cur_stream()));
+  return {dp, info};
+}
+
 std::vector<torch::Tensor> jtr_grad(
     torch::Tensor x, torch::Tensor coh, torch::Tensor J,
     torch::Tensor pairs, torch::Tensor chunk_tab,
@@ -156,6 +174,8 @@ std::vector<torch::Tensor> jtr_grad(
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("jtr_grad", &jtr_grad, "gradient-only accumulation (gfx950)");
   m.def("chol_solve", &chol_solve, "batched damped Cholesky solve (gfx950)");
+  m.def("chol_solve_mw", &chol_solve_mw,
+        "multi-workgroup damped Cholesky solve (gfx950)");
   m.def("predict_coh", &predict_coh, "coherency predict (gfx950)");
   m.def("jtj_jtr", &jtj_jtr, "fused JtJ/Jtr assembly (gfx950)");
   m.def("model_cost", &model_cost, "per-chunk model cost (gfx950)");

@@ -270,3 +270,259 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
     if (tid == 0) atomicOr(&info[bid], 1);
   }
 }
+
+// ======================================================================
+// Multi-workgroup right-looking variant.
+//
+// The single-kernel solver above runs ONE workgroup per problem: at the
+// LM batch sizes of the headline config (B≈5, n=512) that is <2% of the
+// 256 CUs and k_chol_solve measures ~65% of all GPU time (753 us/call,
+// profiles/lm_kernel_stats_final.csv). Here the factorization is split
+// into a kernel SEQUENCE on one stream (graph-captured, so launch cost
+// is node dispatch only):
+//   k_cholmw_init   — L := A + mu*I (working copy; right-looking mutates)
+//   per panel k:
+//     k_cholmw_panel — factor 32x32 diag + row-solve the panel
+//                      (1 WG/problem; serial critical path)
+//     k_cholmw_syrk  — trailing update: ONE 32x32 MFMA tile per 64-lane
+//                      wave, grid = batch * n_lower_tiles — the O(n^3)
+//                      work spreads over the whole chip instead of 1 CU
+//   k_cholmw_subst  — both triangular substitutions + NaN poison
+// Same scratch contract as k_chol_solve (2*n*n: L | L^T mirror).
+// ======================================================================
+
+extern "C" __global__ void __launch_bounds__(NTH)
+k_cholmw_init(const float* __restrict__ JtJ, const float* __restrict__ mu,
+              int n, float* __restrict__ Lbuf) {
+  const int bid = blockIdx.x, tid = threadIdx.x;
+  const float* A = JtJ + (size_t)bid * n * n;
+  float* L = Lbuf + (size_t)bid * 2 * n * n;
+  const float m = mu[bid];
+  const int nq = n * n / 4;
+  for (int idx = tid; idx < nq; idx += NTH) {
+    const size_t p = 4ull * idx;
+    float4 v = *(const float4*)(A + p);
+    const int r = (int)(p / n), c = (int)(p % n);
+    const int d = r - c;
+    if (d >= 0 && d < 4) {
+      if (d == 0) v.x += m;
+      else if (d == 1) v.y += m;
+      else if (d == 2) v.z += m;
+      else v.w += m;
+    }
+    *(float4*)(L + p) = v;
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(NTH)
+k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
+               int* __restrict__ info) {
+  const int bid = blockIdx.x, tid = threadIdx.x, lane = tid & 63;
+  float* L = Lbuf + (size_t)bid * 2 * n * n;
+  float* LT = L + (size_t)n * n;
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* pan = smem;
+  __shared__ float yv[NB];
+  __shared__ int bad;
+  if (tid == 0) bad = 0;
+  const int rows = n - k;
+  // stage panel rows k..n, cols k..k+NB from the (already-updated) L copy
+  for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
+    const int r = idx >> 3, c4 = (idx & 7) << 2;
+    *(float4*)(pan + r * PST + c4) =
+        *(const float4*)(L + (size_t)(k + r) * n + k + c4);
+  }
+  __syncthreads();
+  // wave-synchronous 32x32 factor (same scheme as k_chol_solve)
+  if (tid < 64 && lane < 32) {
+    const int r = lane;
+    float row[NB];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) row[c] = pan[r * PST + c];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      float pivraw = __shfl(row[c], c, 64);
+      if (pivraw <= 1e-30f) {
+        if (lane == 0) bad = 1;
+        pivraw = 1e-30f;
+      }
+      const float pv = sqrtf(pivraw);
+      if (r >= c) row[c] /= pv;
+      yv[r] = row[c];
+      const float lrc = row[c];
+#pragma unroll
+      for (int cc = c + 1; cc < NB; ++cc) {
+        if (r >= cc) row[cc] -= lrc * yv[cc];
+      }
+    }
+#pragma unroll
+    for (int c = 0; c < NB; ++c) pan[r * PST + c] = row[c];
+  }
+  __syncthreads();
+  // row-solve sub-panel rows NB..rows
+  for (int r = NB + tid; r < rows; r += NTH) {
+    float rw[NB];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) rw[c] = pan[r * PST + c];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      float s = rw[c];
+#pragma unroll
+      for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * pan[c * PST + c2];
+      rw[c] = s / pan[c * PST + c];
+    }
+#pragma unroll
+    for (int c = 0; c < NB; ++c) pan[r * PST + c] = rw[c];
+  }
+  __syncthreads();
+  // write back: L (the syrk's consumers read LT; subst reads both)
+  for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
+    const int r = idx >> 3, c4 = (idx & 7) << 2;
+    *(float4*)(L + (size_t)(k + r) * n + k + c4) =
+        *(const float4*)(pan + r * PST + c4);
+  }
+#pragma unroll
+  for (int c = 0; c < NB; ++c) {
+    for (int r = tid; r < rows; r += NTH) {
+      LT[(size_t)(k + c) * n + k + r] = pan[r * PST + c];
+    }
+  }
+  __syncthreads();
+  if (bad && tid == 0) atomicOr(&info[bid], 1);
+}
+
+// trailing SYRK: tile t of the lower-triangular 32x32 tile grid of the
+// trailing matrix (rows/cols k+NB..n). One 64-lane wave per tile: 2x2
+// fragments of mfma_f32_16x16x4, K = NB, operands read COALESCED from
+// the LT mirror (same fragment scheme as the left-looking update above).
+extern "C" __global__ void __launch_bounds__(64)
+k_cholmw_syrk(int n, int k, int ntiles, float* __restrict__ Lbuf) {
+  const int g = blockIdx.x;
+  const int bid = g / ntiles, t = g % ntiles;
+  int I = (int)((sqrtf(8.0f * t + 1.0f) - 1.0f) * 0.5f + 1e-4f);
+  while ((I + 1) * (I + 2) / 2 <= t) ++I;
+  while (I * (I + 1) / 2 > t) --I;
+  const int J = t - I * (I + 1) / 2;
+  float* L = Lbuf + (size_t)bid * 2 * n * n;
+  const float* LT = L + (size_t)n * n;
+  const int r0 = k + NB + I * NB;
+  const int c0 = k + NB + J * NB;
+  const int lane = threadIdx.x & 63;
+  const int l15 = lane & 15, l4 = lane >> 4;
+  const float* LTb = LT + (size_t)(k + l4) * n;
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
+  f32x4 a00 = {0.f, 0.f, 0.f, 0.f}, a01 = a00, a10 = a00, a11 = a00;
+#pragma unroll
+  for (int kk = 0; kk < NB; kk += 4) {
+    const float x0 = LTb[(size_t)kk * n + r0 + l15];
+    const float x1 = LTb[(size_t)kk * n + r0 + 16 + l15];
+    const float y0 = LTb[(size_t)kk * n + c0 + l15];
+    const float y1 = LTb[(size_t)kk * n + c0 + 16 + l15];
+    a00 = __builtin_amdgcn_mfma_f32_16x16x4f32(x0, y0, a00, 0, 0, 0);
+    a01 = __builtin_amdgcn_mfma_f32_16x16x4f32(x0, y1, a01, 0, 0, 0);
+    a10 = __builtin_amdgcn_mfma_f32_16x16x4f32(x1, y0, a10, 0, 0, 0);
+    a11 = __builtin_amdgcn_mfma_f32_16x16x4f32(x1, y1, a11, 0, 0, 0);
+  }
+  // D layout: col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int ra = r0 + l4 * 4 + reg, rb = ra + 16;
+    const int ca = c0 + l15, cb = ca + 16;
+    L[(size_t)ra * n + ca] -= a00[reg];
+    L[(size_t)ra * n + cb] -= a01[reg];
+    L[(size_t)rb * n + ca] -= a10[reg];
+    L[(size_t)rb * n + cb] -= a11[reg];
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(NTH)
+k_cholmw_subst(int n, const float* __restrict__ Jtr,
+               const float* __restrict__ Lbuf, float* __restrict__ dp,
+               const int* __restrict__ info) {
+  const int bid = blockIdx.x, tid = threadIdx.x, lane = tid & 63;
+  const float* L = Lbuf + (size_t)bid * 2 * n * n;
+  const float* LT = L + (size_t)n * n;
+  const float* b = Jtr + (size_t)bid * n;
+  float* xo = dp + (size_t)bid * n;
+  __shared__ float pan[NB * PST];
+  __shared__ float yv[NB];
+  for (int idx = tid; idx < n; idx += NTH) xo[idx] = b[idx];
+  __syncthreads();
+  // forward: L y = b
+  for (int k = 0; k < n; k += NB) {
+    for (int idx = tid; idx < NB * NB; idx += NTH) {
+      const int r = idx >> 5, c = idx & 31;
+      if (c <= r) pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];
+    }
+    __syncthreads();
+    if (tid < 64) {
+      const int r = lane & 31;
+      float rv[NB];
+      float bv = (lane < 32) ? xo[k + r] : 0.f;
+      if (lane < 32) {
+#pragma unroll
+        for (int c = 0; c < NB; ++c)
+          rv[c] = (c <= r) ? pan[r * PST + c] : 0.f;
+      }
+#pragma unroll
+      for (int c = 0; c < NB; ++c) {
+        float yc;
+        if (lane == c) bv /= rv[c];
+        yc = __shfl(bv, c, 64);
+        if (lane < 32 && r > c) bv -= rv[c] * yc;
+        if (lane == c) yv[c] = bv;
+      }
+      if (lane < 32) xo[k + r] = bv;
+    }
+    __syncthreads();
+    for (int i = k + NB + tid; i < n; i += NTH) {
+      float s = 0.f;
+      const float* Lr = L + (size_t)i * n + k;
+#pragma unroll
+      for (int c = 0; c < NB; ++c) s += Lr[c] * yv[c];
+      xo[i] -= s;
+    }
+    __syncthreads();
+  }
+  // backward: L^T x = y (LT rows coalesced)
+  for (int k = ((n - 1) / NB) * NB; k >= 0; k -= NB) {
+    for (int idx = tid; idx < NB * NB; idx += NTH) {
+      const int r = idx >> 5, c = idx & 31;
+      if (c >= r) pan[r * PST + c] = LT[(size_t)(k + r) * n + k + c];
+    }
+    __syncthreads();
+    if (tid < 64) {
+      const int r = lane & 31;
+      float rv[NB];
+      float bv = (lane < 32) ? xo[k + r] : 0.f;
+      if (lane < 32) {
+#pragma unroll
+        for (int c = 0; c < NB; ++c)
+          rv[c] = (c >= r) ? pan[r * PST + c] : 0.f;
+      }
+#pragma unroll
+      for (int ci = 0; ci < NB; ++ci) {
+        const int c = NB - 1 - ci;
+        float xc;
+        if (lane == c) bv /= rv[c];
+        xc = __shfl(bv, c, 64);
+        if (lane < 32 && r < c) bv -= rv[c] * xc;
+        if (lane == c) yv[c] = bv;
+      }
+      if (lane < 32) xo[k + r] = bv;
+    }
+    __syncthreads();
+    for (int i = tid; i < k; i += NTH) {
+      float s = 0.f;
+      const float* Lr = LT + (size_t)i * n + k;
+#pragma unroll
+      for (int c = 0; c < NB; ++c) s += Lr[c] * yv[c];
+      xo[i] -= s;
+    }
+    __syncthreads();
+  }
+  if (info[bid]) {
+    const float qn = __int_as_float(0x7fc00000);
+    for (int idx = tid; idx < n; idx += NTH) xo[idx] = qn;
+  }
+}

@@ -74,4 +74,27 @@ hipError_t launch_chol_solve(
       JtJ, Jtr, mu, n, Lbuf, dp, info, stages);
   return hipGetLastError();
 }
+
+// multi-workgroup right-looking variant: kernel sequence on one stream
+// (graph-capturable). See cholesky.hip for the rationale.
+hipError_t launch_chol_mw(
+    const float* JtJ, const float* Jtr, const float* mu, int n, int batch,
+    float* Lbuf, float* dp, int* info, hipStream_t stream) {
+  hipLaunchKernelGGL(k_cholmw_init, dim3(batch), dim3(512), 0, stream,
+      JtJ, mu, n, Lbuf);
+  const size_t shmem = (size_t)(n + 40) * PST * sizeof(float);
+  for (int k = 0; k < n; k += NB) {
+    hipLaunchKernelGGL(k_cholmw_panel, dim3(batch), dim3(512), shmem,
+        stream, n, k, Lbuf, info);
+    const int tcnt = (n - k - NB) / NB;
+    const int ntiles = tcnt * (tcnt + 1) / 2;
+    if (ntiles > 0) {
+      hipLaunchKernelGGL(k_cholmw_syrk, dim3(batch * ntiles), dim3(64), 0,
+          stream, n, k, ntiles, Lbuf);
+    }
+  }
+  hipLaunchKernelGGL(k_cholmw_subst, dim3(batch), dim3(512), 0, stream,
+      n, Jtr, Lbuf, dp, info);
+  return hipGetLastError();
+}
 }  // extern "C"

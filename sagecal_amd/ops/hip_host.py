@@ -6,6 +6,8 @@ projection identity into unused rotation terms, chunk tables, station-pair
 index tables, and complex64 views.
 """
 import math
+import os
+
 import torch
 
 from . import dispatch
@@ -348,7 +350,19 @@ def chol_solve_damped(JtJ, Jtr, mu):
     if sc is None or tuple(sc.shape) != want:
         sc = torch.empty(want, dtype=JtJ.dtype, device=JtJ.device)
         _chol_scratch[key] = sc
-    dp, info = _ext().chol_solve(JtJ.contiguous(), Jtr.contiguous(),
-                                 mu.to(torch.float32).contiguous(), sc, 3)
+    # n >= 256: multi-workgroup right-looking path — the trailing-update
+    # SYRK tiles spread over the whole chip instead of 1 WG/problem
+    # (65% of step time at [5,512,512]; see profiles/). Small n stays on
+    # the single fused kernel (lower launch count wins).
+    mw = os.environ.get('SAGECAL_CHOL_MW')
+    use_mw = (JtJ.shape[1] >= 256) if mw is None else mw == '1'
+    if use_mw:
+        dp, info = _ext().chol_solve_mw(JtJ.contiguous(), Jtr.contiguous(),
+                                        mu.to(torch.float32).contiguous(),
+                                        sc)
+    else:
+        dp, info = _ext().chol_solve(JtJ.contiguous(), Jtr.contiguous(),
+                                     mu.to(torch.float32).contiguous(),
+                                     sc, 3)
     # failed factorizations already return NaN rows (kernel poisons dp)
     return dp[:, :n]
