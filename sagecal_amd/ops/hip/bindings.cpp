@@ -23,7 +23,7 @@ hipError_t launch_apply_jones(const float2*, const float2*, const float2*,
     const int*, const int*, int, int, int, int, int, int, float2*,
     hipStream_t);
 hipError_t launch_chol_mw(const float*, const float*, const float*, int,
-                          int, float*, float*, int*, hipStream_t);
+                          int, float*, float*, int*, int, hipStream_t);
 hipError_t launch_chol_solve(const float*, const float*, const float*, int,
     int, float*, float*, int*, int, hipStream_t);
 }
@@ -138,7 +138,7 @@ std::vector<torch::Tensor> chol_solve(
 
 std::vector<torch::Tensor> chol_solve_mw(
     torch::Tensor JtJ, torch::Tensor Jtr, torch::Tensor mu,
-    torch::Tensor scratch) {
+    torch::Tensor scratch, int64_t stages) {
   // multi-workgroup right-looking path: same contract as chol_solve
   const int64_t batch = JtJ.size(0);
   const int64_t n = JtJ.size(1);
@@ -148,7 +148,8 @@ std::vector<torch::Tensor> chol_solve_mw(
       torch::dtype(torch::kInt).device(JtJ.device()));
   CHECK_HIP(launch_chol_mw(JtJ.data_ptr<float>(), Jtr.data_ptr<float>(),
       mu.data_ptr<float>(), (int)n, (int)batch, scratch.data_ptr<float>(),
-      dp.data_ptr<float>(), info.data_ptr<int>(), cur_stream()));
+      dp.data_ptr<float>(), info.data_ptr<int>(), (int)stages,
+      cur_stream()));
   return {dp, info};
 }
 

@@ -291,15 +291,21 @@ k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
 // Same scratch contract as k_chol_solve (2*n*n: L | L^T mirror).
 // ======================================================================
 
+// grid (batch, SLICES): B workgroups alone leave the copy bandwidth-bound
+// on B CUs (~75 us for [5,512,512]); sliced over 32 WGs/problem it is ~8 us.
+// Also seeds dp := b (the panel kernels run the forward substitution
+// in-place as each panel factors).
+#define INIT_SLICES 32
 extern "C" __global__ void __launch_bounds__(NTH)
-k_cholmw_init(const float* __restrict__ JtJ, const float* __restrict__ mu,
-              int n, float* __restrict__ Lbuf) {
-  const int bid = blockIdx.x, tid = threadIdx.x;
+k_cholmw_init(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
+              const float* __restrict__ mu, int n, float* __restrict__ Lbuf,
+              float* __restrict__ dp) {
+  const int bid = blockIdx.x, slice = blockIdx.y, tid = threadIdx.x;
   const float* A = JtJ + (size_t)bid * n * n;
   float* L = Lbuf + (size_t)bid * 2 * n * n;
   const float m = mu[bid];
   const int nq = n * n / 4;
-  for (int idx = tid; idx < nq; idx += NTH) {
+  for (int idx = slice * NTH + tid; idx < nq; idx += NTH * INIT_SLICES) {
     const size_t p = 4ull * idx;
     float4 v = *(const float4*)(A + p);
     const int r = (int)(p / n), c = (int)(p % n);
@@ -312,11 +318,14 @@ k_cholmw_init(const float* __restrict__ JtJ, const float* __restrict__ mu,
     }
     *(float4*)(L + p) = v;
   }
+  if (slice == 0) {
+    for (int i = tid; i < n; i += NTH) dp[(size_t)bid * n + i] = Jtr[(size_t)bid * n + i];
+  }
 }
 
 extern "C" __global__ void __launch_bounds__(NTH)
 k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
-               int* __restrict__ info) {
+               float* __restrict__ dp, int* __restrict__ info) {
   const int bid = blockIdx.x, tid = threadIdx.x, lane = tid & 63;
   float* L = Lbuf + (size_t)bid * 2 * n * n;
   float* LT = L + (size_t)n * n;
@@ -375,12 +384,10 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
     for (int c = 0; c < NB; ++c) pan[r * PST + c] = rw[c];
   }
   __syncthreads();
-  // write back: L (the syrk's consumers read LT; subst reads both)
-  for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
-    const int r = idx >> 3, c4 = (idx & 7) << 2;
-    *(float4*)(L + (size_t)(k + r) * n + k + c4) =
-        *(const float4*)(pan + r * PST + c4);
-  }
+  // write back the LT mirror ONLY: every consumer of the factored panel
+  // (syrk fragments, fused forward subst via LDS, backward subst) reads
+  // L^T; the L working copy keeps unfactored trailing values, which is
+  // all later panel stages ever read.
 #pragma unroll
   for (int c = 0; c < NB; ++c) {
     for (int r = tid; r < rows; r += NTH) {
@@ -388,6 +395,36 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
     }
   }
   __syncthreads();
+  // fused forward-substitution step: y_k = L_kk^-1 xo[k..k+NB] on wave 0,
+  // then xo[k+NB..n] -= panel . y_k straight from LDS (the separate subst
+  // kernel then only runs the backward pass)
+  float* xo = dp + (size_t)bid * n;
+  if (tid < 64) {
+    const int r = lane & 31;
+    float rv[NB];
+    float bv = (lane < 32) ? xo[k + r] : 0.f;
+    if (lane < 32) {
+#pragma unroll
+      for (int c = 0; c < NB; ++c)
+        rv[c] = (c <= r) ? pan[r * PST + c] : 0.f;
+    }
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      float yc;
+      if (lane == c) bv /= rv[c];
+      yc = __shfl(bv, c, 64);
+      if (lane < 32 && r > c) bv -= rv[c] * yc;
+      if (lane == c) yv[c] = bv;
+    }
+    if (lane < 32) xo[k + r] = bv;
+  }
+  __syncthreads();
+  for (int r = NB + tid; r < rows; r += NTH) {
+    float s = 0.f;
+#pragma unroll
+    for (int c = 0; c < NB; ++c) s += pan[r * PST + c] * yv[c];
+    xo[k + r] -= s;
+  }
   if (bad && tid == 0) atomicOr(&info[bid], 1);
 }
 
@@ -435,55 +472,17 @@ k_cholmw_syrk(int n, int k, int ntiles, float* __restrict__ Lbuf) {
   }
 }
 
+// backward substitution only: dp already holds y = L^-1 b (forward pass
+// fused into the panel kernels)
 extern "C" __global__ void __launch_bounds__(NTH)
-k_cholmw_subst(int n, const float* __restrict__ Jtr,
-               const float* __restrict__ Lbuf, float* __restrict__ dp,
-               const int* __restrict__ info) {
+k_cholmw_subst(int n, const float* __restrict__ Lbuf,
+               float* __restrict__ dp, const int* __restrict__ info) {
   const int bid = blockIdx.x, tid = threadIdx.x, lane = tid & 63;
   const float* L = Lbuf + (size_t)bid * 2 * n * n;
   const float* LT = L + (size_t)n * n;
-  const float* b = Jtr + (size_t)bid * n;
   float* xo = dp + (size_t)bid * n;
   __shared__ float pan[NB * PST];
   __shared__ float yv[NB];
-  for (int idx = tid; idx < n; idx += NTH) xo[idx] = b[idx];
-  __syncthreads();
-  // forward: L y = b
-  for (int k = 0; k < n; k += NB) {
-    for (int idx = tid; idx < NB * NB; idx += NTH) {
-      const int r = idx >> 5, c = idx & 31;
-      if (c <= r) pan[r * PST + c] = L[(size_t)(k + r) * n + k + c];
-    }
-    __syncthreads();
-    if (tid < 64) {
-      const int r = lane & 31;
-      float rv[NB];
-      float bv = (lane < 32) ? xo[k + r] : 0.f;
-      if (lane < 32) {
-#pragma unroll
-        for (int c = 0; c < NB; ++c)
-          rv[c] = (c <= r) ? pan[r * PST + c] : 0.f;
-      }
-#pragma unroll
-      for (int c = 0; c < NB; ++c) {
-        float yc;
-        if (lane == c) bv /= rv[c];
-        yc = __shfl(bv, c, 64);
-        if (lane < 32 && r > c) bv -= rv[c] * yc;
-        if (lane == c) yv[c] = bv;
-      }
-      if (lane < 32) xo[k + r] = bv;
-    }
-    __syncthreads();
-    for (int i = k + NB + tid; i < n; i += NTH) {
-      float s = 0.f;
-      const float* Lr = L + (size_t)i * n + k;
-#pragma unroll
-      for (int c = 0; c < NB; ++c) s += Lr[c] * yv[c];
-      xo[i] -= s;
-    }
-    __syncthreads();
-  }
   // backward: L^T x = y (LT rows coalesced)
   for (int k = ((n - 1) / NB) * NB; k >= 0; k -= NB) {
     for (int idx = tid; idx < NB * NB; idx += NTH) {

@@ -79,22 +79,24 @@ hipError_t launch_chol_solve(
 // (graph-capturable). See cholesky.hip for the rationale.
 hipError_t launch_chol_mw(
     const float* JtJ, const float* Jtr, const float* mu, int n, int batch,
-    float* Lbuf, float* dp, int* info, hipStream_t stream) {
-  hipLaunchKernelGGL(k_cholmw_init, dim3(batch), dim3(512), 0, stream,
-      JtJ, mu, n, Lbuf);
+    float* Lbuf, float* dp, int* info, int stages, hipStream_t stream) {
+  hipLaunchKernelGGL(k_cholmw_init, dim3(batch, INIT_SLICES), dim3(512), 0,
+      stream, JtJ, Jtr, mu, n, Lbuf, dp);
+  if (stages < 2) return hipGetLastError();
   const size_t shmem = (size_t)(n + 40) * PST * sizeof(float);
   for (int k = 0; k < n; k += NB) {
     hipLaunchKernelGGL(k_cholmw_panel, dim3(batch), dim3(512), shmem,
-        stream, n, k, Lbuf, info);
+        stream, n, k, Lbuf, dp, info);
     const int tcnt = (n - k - NB) / NB;
     const int ntiles = tcnt * (tcnt + 1) / 2;
-    if (ntiles > 0) {
+    if (ntiles > 0 && stages >= 3) {
       hipLaunchKernelGGL(k_cholmw_syrk, dim3(batch * ntiles), dim3(64), 0,
           stream, n, k, ntiles, Lbuf);
     }
   }
+  if (stages < 4) return hipGetLastError();
   hipLaunchKernelGGL(k_cholmw_subst, dim3(batch), dim3(512), 0, stream,
-      n, Jtr, Lbuf, dp, info);
+      n, Lbuf, dp, info);
   return hipGetLastError();
 }
 }  // extern "C"

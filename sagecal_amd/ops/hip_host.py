@@ -355,11 +355,11 @@ def chol_solve_damped(JtJ, Jtr, mu):
     # (65% of step time at [5,512,512]; see profiles/). Small n stays on
     # the single fused kernel (lower launch count wins).
     mw = os.environ.get('SAGECAL_CHOL_MW')
-    use_mw = (JtJ.shape[1] >= 256) if mw is None else mw == '1'
+    use_mw = (JtJ.shape[1] >= 384) if mw is None else mw == '1'
     if use_mw:
         dp, info = _ext().chol_solve_mw(JtJ.contiguous(), Jtr.contiguous(),
                                         mu.to(torch.float32).contiguous(),
-                                        sc)
+                                        sc, 4)
     else:
         dp, info = _ext().chol_solve(JtJ.contiguous(), Jtr.contiguous(),
                                      mu.to(torch.float32).contiguous(),
