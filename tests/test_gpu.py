@@ -256,3 +256,25 @@ def test_lbfgs_cost_grad_gpu():
         err = (g_g.cpu().double() - g_ref).abs().max() / \
             float(g_ref.abs().max())
         assert float(err) < 1e-4, float(err)
+
+
+def test_chol_mw_matches_single_kernel():
+    """Multi-workgroup right-looking path vs the fused single-WG kernel
+    (same scratch contract; dispatch picks mw for 384<=n<=576)."""
+    import sagecal_amd.ops.hip.dirac_hip as ext
+    dev = 'cuda:0'
+    rng = np.random.default_rng(5)
+    n, batch = 512, 4
+    Araw = torch.tensor(rng.standard_normal((batch, n, n)),
+                        dtype=torch.float32, device=dev)
+    A = ((Araw @ Araw.transpose(-1, -2)) / n
+         + 0.5 * torch.eye(n, device=dev).unsqueeze(0)).contiguous()
+    b = torch.tensor(rng.standard_normal((batch, n)), dtype=torch.float32,
+                     device=dev).contiguous()
+    mu = torch.full((batch,), 0.1, device=dev)
+    sc = torch.empty(batch, 2 * n * n, dtype=torch.float32, device=dev)
+    x1, i1 = ext.chol_solve(A, b, mu, sc, 3)
+    x2, i2 = ext.chol_solve_mw(A, b, mu, sc, 4)
+    assert int(i1.sum()) == 0 and int(i2.sum()) == 0
+    err = (x1 - x2).abs().max() / x1.abs().max()
+    assert float(err) < 1e-4, f"mw vs single rel err {float(err)}"
