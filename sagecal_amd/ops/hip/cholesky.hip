@@ -332,6 +332,7 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* pan = smem;
   __shared__ float yv[NB];
+  __shared__ float rdg[NB];   // reciprocal diagonal (div -> mul downstream)
   __shared__ int bad;
   if (tid == 0) bad = 0;
   const int rows = n - k;
@@ -356,8 +357,10 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
         pivraw = 1e-30f;
       }
       const float pv = sqrtf(pivraw);
-      if (r >= c) row[c] /= pv;
+      const float rpv = 1.0f / pv;
+      if (r >= c) row[c] *= rpv;
       yv[r] = row[c];
+      if (lane == c) rdg[c] = rpv;
       const float lrc = row[c];
 #pragma unroll
       for (int cc = c + 1; cc < NB; ++cc) {
@@ -368,7 +371,8 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
     for (int c = 0; c < NB; ++c) pan[r * PST + c] = row[c];
   }
   __syncthreads();
-  // row-solve sub-panel rows NB..rows
+  // row-solve sub-panel rows NB..rows (multiply by reciprocal diag: the
+  // 32 serial divides per row-thread were the panel's longest chain)
   for (int r = NB + tid; r < rows; r += NTH) {
     float rw[NB];
 #pragma unroll
@@ -378,7 +382,7 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
       float s = rw[c];
 #pragma unroll
       for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * pan[c * PST + c2];
-      rw[c] = s / pan[c * PST + c];
+      rw[c] = s * rdg[c];
     }
 #pragma unroll
     for (int c = 0; c < NB; ++c) pan[r * PST + c] = rw[c];
@@ -411,7 +415,7 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
 #pragma unroll
     for (int c = 0; c < NB; ++c) {
       float yc;
-      if (lane == c) bv /= rv[c];
+      if (lane == c) bv *= rdg[c];
       yc = __shfl(bv, c, 64);
       if (lane < 32 && r > c) bv -= rv[c] * yc;
       if (lane == c) yv[c] = bv;
