@@ -369,6 +369,10 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
     }
 #pragma unroll
     for (int c = 0; c < NB; ++c) pan[r * PST + c] = row[c];
+    // publish 1/pv on the panel diagonal: the LT mirror's diagonal is read
+    // ONLY by the backward substitution's divide (syrk/fwd never touch it),
+    // so storing the reciprocal there turns that divide into a multiply
+    pan[r * PST + r] = rdg[r];
   }
   __syncthreads();
   // row-solve sub-panel rows NB..rows (multiply by reciprocal diag: the
@@ -507,7 +511,7 @@ k_cholmw_subst(int n, const float* __restrict__ Lbuf,
       for (int ci = 0; ci < NB; ++ci) {
         const int c = NB - 1 - ci;
         float xc;
-        if (lane == c) bv /= rv[c];
+        if (lane == c) bv *= rv[c];   // diag holds 1/pv (panel kernel)
         xc = __shfl(bv, c, 64);
         if (lane < 32 && r < c) bv -= rv[c] * xc;
         if (lane == c) yv[c] = bv;
