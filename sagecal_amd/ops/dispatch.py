@@ -29,6 +29,25 @@ def _load_ext():
     return _ext
 
 
+_token_counter = [0]
+
+
+def obj_token(obj):
+    """Stable per-object cache token. id() is NOT a safe cache key — CPython
+    recycles addresses after GC, so an id()-keyed cache can serve STALE
+    entries (seen in practice: sage._solve_group served a freed tile's
+    coherency buffers). The token is stored on the object and dies with it."""
+    tok = getattr(obj, '_sagecal_token', None)
+    if tok is None:
+        _token_counter[0] += 1
+        tok = _token_counter[0]
+        try:
+            obj._sagecal_token = tok
+        except AttributeError:   # __slots__/tensor subclass without attr
+            return id(obj)
+    return tok
+
+
 def have_ext():
     return _load_ext() is not None
 

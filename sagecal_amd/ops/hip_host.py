@@ -102,8 +102,10 @@ _gpu_pack_cache = {}
 
 
 def gpu_pack(pack, device):
-    key = id(pack)
+    key = (dispatch.obj_token(pack), str(device))
     if key not in _gpu_pack_cache:
+        if len(_gpu_pack_cache) > 16:   # bound long multi-MS runs
+            _gpu_pack_cache.pop(next(iter(_gpu_pack_cache)))
         _gpu_pack_cache[key] = GPUPack(pack, device)
     return _gpu_pack_cache[key]
 
@@ -348,6 +350,8 @@ def chol_solve_damped(JtJ, Jtr, mu):
     sc = _chol_scratch.get(key)
     want = (JtJ.shape[0], 2 * JtJ.shape[1] * JtJ.shape[2])
     if sc is None or tuple(sc.shape) != want:
+        if len(_chol_scratch) > 8:   # bound long multi-config runs
+            _chol_scratch.pop(next(iter(_chol_scratch)))
         sc = torch.empty(want, dtype=JtJ.dtype, device=JtJ.device)
         _chol_scratch[key] = sc
     # n >= 256: multi-workgroup right-looking path — the trailing-update

@@ -344,7 +344,7 @@ def _lm_solve_graphed(prob, J0, maxiter, tau, eps1, eps2):
     nchunk = prob.nchunk
     dev = x.device
     B = x.shape[0]
-    key = (B, nchunk, N, maxiter, id(prob.layout), float(tau),
+    key = (B, nchunk, N, maxiter, ops.obj_token(prob.layout), float(tau),
            prob.admm is not None)
     ent = _lm_graph_cache.get(key)
     if ent is None:
@@ -443,7 +443,7 @@ def robust_lm_graphed(prob, J0, nu0, inner, tau=1e-3):
     nchunk = prob.nchunk
     dev = x.device
     B = x.shape[0]
-    key = ('robust', B, nchunk, N, inner, id(prob.layout),
+    key = ('robust', B, nchunk, N, inner, ops.obj_token(prob.layout),
            prob.admm is not None)
     ent = _robust_graph_cache.get(key)
     if ent is None:

@@ -222,12 +222,11 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts,
     if cache is None:
         cache = state._grp_cache = {}
     # token-based key: id() is recycled by the allocator across tiles
-    tok = getattr(cohs, '_sagecal_token', None)
-    if tok is None:
-        tok = getattr(sagefit, '_next_token', 0)
-        sagefit._next_token = tok + 1
-        cohs._sagecal_token = tok
-        cache.clear()          # new tile: drop stale group buffers
+    # (ops.obj_token); a fresh token means a new tile -> drop old buffers
+    new_tile = getattr(cohs, '_sagecal_token', None) is None
+    tok = ops.obj_token(cohs)
+    if new_tile:
+        cache.clear()
     key = (tuple(group), tok, B)
     ent = cache.get(key)
     if ent is None:

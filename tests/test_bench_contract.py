@@ -1,0 +1,61 @@
+"""The driver depends on bench.py's JSON contract (BASELINE.json metric,
+one JSON line on stdout from rank 0). Guard it for every bench mode so a
+schedule/solver change can't silently break the round-end measurement.
+
+Runs tiny CPU configs (--cpu) of each mode and validates the JSON keys
+and value sanity.
+"""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+REQUIRED_KEYS = {
+    'metric', 'value', 'unit', 'n_gpus', 'steps', 'warmup', 'ms_per_step',
+    'higher_is_better', 'scaling', 'vs_baseline', 'dtype', 'data', 'config',
+}
+
+
+def _run_bench(extra, timeout=600):
+    cmd = [sys.executable, os.path.join(ROOT, 'bench.py'), '--cpu',
+           '--steps', '1', '--warmup', '0'] + extra
+    out = subprocess.run(cmd, capture_output=True, text=True, cwd=ROOT,
+                         timeout=timeout)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    return json.loads(line)
+
+
+def _check(d):
+    assert REQUIRED_KEYS <= set(d), REQUIRED_KEYS - set(d)
+    assert d['metric'] == 'visibilities/sec calibrated'
+    assert d['value'] > 0 and d['ms_per_step'] > 0
+    assert d['higher_is_better'] is True
+    assert d['scaling'] == 'weak'
+    assert d['data'] == 'synthetic'
+    assert isinstance(d['config'], dict) and 'model' in d['config']
+
+
+def test_bench_default_contract():
+    d = _run_bench(['--stations', '8', '--dirs', '2', '--srcs', '2',
+                    '--tilesz', '4', '--chan', '2'])
+    _check(d)
+    assert d['config']['stations'] == 8
+
+
+def test_bench_rtr_contract():
+    d = _run_bench(['--stations', '48', '--dirs', '2', '--srcs', '2',
+                    '--tilesz', '2', '--chan', '2', '--solver', 'rtr',
+                    '--no-consensus'])
+    _check(d)
+
+
+def test_bench_bandpass_contract():
+    d = _run_bench(['--mode', 'bandpass', '--stations', '8', '--dirs', '2',
+                    '--srcs', '2', '--tilesz', '2', '--chan', '16',
+                    '--nsolbw', '4'])
+    _check(d)
