@@ -80,7 +80,7 @@ def load_context(args):
         else bool(args.gpupredict)
     device = 'cuda:0' if use_gpu else 'cpu'
     dtype = torch.float32 if use_gpu else torch.float64
-    ms = msdata.NpzMS(args.ms, tilesz=args.tilesz, device=device,
+    ms = msdata.open_ms(args.ms, tilesz=args.tilesz, device=device,
                       dtype=dtype)
     ignore = skymod.read_ignore_file(args.ignfile) if args.ignfile else ()
     clusters = skymod.read_sky_cluster(args.sky, args.cluster, ms.ra0,

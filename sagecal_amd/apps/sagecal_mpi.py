@@ -77,8 +77,8 @@ def main(argv=None):
         names = [l.strip() for l in f if l.strip()]
     my_ms = names[rank % len(names)] if world <= len(names) else \
         names[rank]
-    ms = msdata.NpzMS(my_ms, tilesz=args.tilesz, device=device,
-                      dtype=dtype)
+    ms = msdata.open_ms(my_ms, tilesz=args.tilesz, device=device,
+                        dtype=dtype)
     clusters = skymod.read_sky_cluster(args.sky, args.cluster, ms.ra0,
                                        ms.dec0, ms.freq0, fmt=args.format)
     pack = SourcePack(clusters)

@@ -307,6 +307,15 @@ class NpzMS:
         np.savez_compressed(path or self.path, **out)
 
 
+def open_ms(path, tilesz=10, device='cpu', dtype=torch.float64):
+    """Open a measurement set by suffix: .npz -> NpzMS (the container used
+    in casacore-free environments), anything else -> CasaMS
+    (python-casacore table on disk, e.g. a LOFAR .MS directory)."""
+    if str(path).endswith('.npz'):
+        return NpzMS(path, tilesz=tilesz, device=device, dtype=dtype)
+    return CasaMS(path, tilesz=tilesz, device=device, dtype=dtype)
+
+
 def make_synthetic_npz(path, N=8, tilesz=4, Ntime=4, Nchan=2, pack=None,
                        **kw):
     """Generate a synthetic observation and persist it as NpzMS."""
