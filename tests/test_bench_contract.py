@@ -59,3 +59,26 @@ def test_bench_bandpass_contract():
                     '--srcs', '2', '--tilesz', '2', '--chan', '16',
                     '--nsolbw', '4'])
     _check(d)
+
+
+def test_bench_two_rank_launch():
+    """Launch bench.py exactly the way the driver's scaling harness does
+    (torch.distributed.run, one process per 'GPU', 127.0.0.1 rendezvous)
+    with 2 CPU/gloo ranks: rank 0 must print the JSON contract line with
+    the whole-job aggregate value."""
+    env = dict(os.environ)
+    env['SAGECAL_BENCH_BACKEND'] = 'gloo'
+    cmd = [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+           '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
+           '--master-port', '29547', os.path.join(ROOT, 'bench.py'),
+           '--gpus', '2', '--cpu', '--steps', '1', '--warmup', '0',
+           '--stations', '8', '--dirs', '2', '--srcs', '2',
+           '--tilesz', '4', '--chan', '2']
+    out = subprocess.run(cmd, capture_output=True, text=True, cwd=ROOT,
+                         env=env, timeout=900)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.strip().splitlines()
+            if l.startswith('{')][-1]
+    d = json.loads(line)
+    _check(d)
+    assert d['config']['parallelism'].endswith('2')
