@@ -1,11 +1,11 @@
 """uvwriter — recompute and rewrite the UVW column of an MS.
 
 The reference tool (/root/reference/src/uvwriter/uvwriter.cpp:46-55)
-rewrites MS UVW for lunar reference frames via CSPICE. CSPICE is not
-available in this environment; this analog recomputes geocentric UVW from
-the array geometry + phase centre (the standard earth-frame path), with a
-pluggable frame hook where a lunar ephemeris transform would slot in.
-"""
+rewrites MS UVW for lunar reference frames via CSPICE. This analog
+implements both frames natively: the standard earth path from array
+geometry + phase centre, and `--frame lunar` via the WGCCRE analytic
+MOON_ME orientation in coords.py (no SPICE kernels needed; ~150 m-class
+frame accuracy, ample for UVW)."""
 import argparse
 import sys
 
@@ -20,9 +20,11 @@ def main(argv=None):
     ap.add_argument('-d', dest='ms', required=True, help='NpzMS path')
     ap.add_argument('--lat', type=float, default=0.92)
     ap.add_argument('--tdelta', type=float, default=None)
-    ap.add_argument('--frame', choices=['earth'], default='earth',
-                    help='reference frame (lunar requires an ephemeris '
-                         'backend; hook in msdata.enu_uvw)')
+    ap.add_argument('--frame', choices=['earth', 'lunar'], default='earth',
+                    help='earth: ENU + hour angle; lunar: WGCCRE MOON_ME '
+                         'body-fixed stations -> J2000 projection')
+    ap.add_argument('--jd0', type=float, default=2460000.5,
+                    help='lunar frame: JD (TDB) of the first time slot')
     args = ap.parse_args(argv)
     ms = msdata.NpzMS(args.ms)
     td = args.tdelta or ms.tdelta
@@ -32,8 +34,23 @@ def main(argv=None):
     p, q = ms.pairs[:, 0], ms.pairs[:, 1]
     us, vs, ws = [], [], []
     for t in range(ms.Ntime):
-        ha = (t + 0.5) * td * 7.2921150e-5 - 0.2
-        uu, vv, ww = msdata.enu_uvw(pos, args.lat, ha, ms.dec0)
+        if args.frame == 'lunar':
+            from .. import coords
+            # place ENU metres on the lunar surface at (lon=0.1, lat)
+            lon0, lat0 = 0.1, args.lat
+            sl, cl = np.sin(lon0), np.cos(lon0)
+            sb, cb = np.sin(lat0), np.cos(lat0)
+            east = np.array([-sl, cl, 0.0])
+            north = np.array([-sb * cl, -sb * sl, cb])
+            up = np.array([cb * cl, cb * sl, sb])
+            pos_me = (coords.MOON_RADIUS * up
+                      + pos[:, 0, None] * east + pos[:, 1, None] * north)
+            jd = args.jd0 + (t + 0.5) * td / 86400.0
+            uu, vv, ww = coords.lunar_uvw(pos_me, ms.ra0, ms.dec0, jd)
+            uu, vv, ww = uu[None], vv[None], ww[None]
+        else:
+            ha = (t + 0.5) * td * 7.2921150e-5 - 0.2
+            uu, vv, ww = msdata.enu_uvw(pos, args.lat, ha, ms.dec0)
         us.append((uu[:, p] - uu[:, q]).reshape(-1))
         vs.append((vv[:, p] - vv[:, q]).reshape(-1))
         ws.append((ww[:, p] - ww[:, q]).reshape(-1))

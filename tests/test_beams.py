@@ -103,3 +103,77 @@ def test_predict_withbeam_changes_vis():
     assert coh_b.shape == coh0.shape
     d = float((coh_b - coh0).abs().mean() / coh0.abs().mean())
     assert 1e-4 < d < 1.0, d   # beam modifies but does not destroy
+
+
+class TestLunarFrames:
+    """Native MOON_ME frame (coords.py) replacing the reference's CSPICE
+    path (cspice_utils.c): orientation series, conversions, az/el, UVW."""
+
+    def test_frame_orthonormal_and_periodic(self):
+        from sagecal_amd import coords
+        for jd in (2451545.0, 2460000.5, 2466123.25):
+            M = coords.j2000_to_moon_me(jd)
+            assert np.allclose(M @ M.T, np.eye(3), atol=1e-12)
+            assert abs(np.linalg.det(M) - 1.0) < 1e-12
+        # prime meridian advances ~360 deg per sidereal month
+        _, _, W0 = coords.moon_orientation(2460000.0)
+        _, _, W1 = coords.moon_orientation(2460000.0 + 27.321661)
+        dW = (W1 - W0) % (2 * np.pi)
+        assert min(dW, 2 * np.pi - dW) < np.deg2rad(4.0)
+
+    def test_sub_earth_point_within_libration(self):
+        from sagecal_amd import coords
+        # the ME +x axis points at the mean Earth: instantaneous sub-Earth
+        # lon/lat must stay inside the ~8 deg optical libration envelope
+        for jd in np.linspace(2459000, 2459730, 60):
+            e = coords.j2000_to_moon_me(jd) @ (
+                -coords.moon_position_j2000(jd))
+            lon = np.arctan2(e[1], e[0])
+            lat = np.arcsin(e[2])
+            assert abs(np.degrees(lon)) < 10.0
+            assert abs(np.degrees(lat)) < 10.0
+
+    def test_latlon_roundtrip_and_azel(self):
+        from sagecal_amd import coords
+        lon, lat, alt = coords.xyz_to_lunar_latlon(
+            *(coords.MOON_RADIUS + 120.0) * np.array(
+                [np.cos(0.7) * np.cos(0.3), np.cos(0.7) * np.sin(0.3),
+                 np.sin(0.7)]))
+        assert abs(lon - 0.3) < 1e-12 and abs(lat - 0.7) < 1e-12
+        assert abs(alt - 120.0) < 1e-6
+        # a source whose sub-source point IS the station sits at zenith
+        jd = 2460123.5
+        ra, dec = 1.1, -0.2
+        slon, slat = coords.lunar_radec_to_latlon(ra, dec, jd)
+        az, el = coords.lunar_azel(ra, dec, slon, slat, jd)
+        assert abs(el - np.pi / 2) < 1e-9
+
+    def test_lunar_uvw_projection(self):
+        from sagecal_amd import coords
+        jd = 2460321.25
+        ra0, dec0 = 0.8, 0.4
+        pos = np.array([[coords.MOON_RADIUS, 0.0, 0.0],
+                        [coords.MOON_RADIUS, 500.0, 0.0],
+                        [coords.MOON_RADIUS, 0.0, 800.0]])
+        u, v, w = coords.lunar_uvw(pos, ra0, dec0, jd)
+        # projection preserves baseline length
+        bl = np.array([u[1] - u[0], v[1] - v[0], w[1] - w[0]])
+        assert abs(np.linalg.norm(bl) - 500.0) < 1e-6
+        # w equals the baseline's component along the source direction
+        M = coords.j2000_to_moon_me(jd)
+        s = np.array([np.cos(dec0) * np.cos(ra0),
+                      np.cos(dec0) * np.sin(ra0), np.sin(dec0)])
+        b2000 = M.T @ (pos[2] - pos[0])
+        assert abs((w[2] - w[0]) - b2000 @ s) < 1e-6
+
+    def test_uvwriter_lunar_cli(self, tmp_path):
+        from sagecal_amd import msdata
+        from sagecal_amd.apps import uvwriter
+        path = str(tmp_path / 'obs.npz')
+        msdata.make_synthetic_npz(path, N=5, tilesz=2, Ntime=2, Nchan=2)
+        u_before = np.array(np.load(path)['u'])
+        assert uvwriter.main(['-d', path, '--frame', 'lunar']) == 0
+        z = np.load(path)
+        assert z['u'].shape == u_before.shape
+        assert not np.allclose(z['u'], u_before)
+        assert np.isfinite(z['u']).all()
