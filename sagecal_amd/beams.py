@@ -283,3 +283,38 @@ def _single_source_coh(pack, gi, u, v, w, freq, freq0, fdelta, tdelta,
                      and gi in pack.shapelets else {})
     return R.predict_coh(one, u, v, w, freq, freq0, fdelta, tdelta,
                          dec0)[0]
+
+
+def lunar_station_azel(ra, dec, lon, lat, jd):
+    """Vectorized topocentric (az, el) of J2000 sources over lunar
+    stations: sources [S] x stations [N] -> ([N,S], [N,S]).
+    Native replacement for the per-station loop in
+    cspice_element_beam_lunar (cspice_utils.c:160-195); the reference
+    approximates el = pi/2 - |dlat| from a haversine between sub-source
+    and station lon/lat — here the exact local-ENU elevation is used."""
+    from . import coords
+    ra = np.atleast_1d(np.asarray(ra, dtype=float))
+    dec = np.atleast_1d(np.asarray(dec, dtype=float))
+    lon = np.atleast_1d(np.asarray(lon, dtype=float))
+    lat = np.atleast_1d(np.asarray(lat, dtype=float))
+    v = np.stack([np.cos(dec) * np.cos(ra), np.cos(dec) * np.sin(ra),
+                  np.sin(dec)])                       # [3, S]
+    s = coords.j2000_to_moon_me(jd) @ v               # [3, S]
+    sl, cl = np.sin(lon)[:, None], np.cos(lon)[:, None]
+    sb, cb = np.sin(lat)[:, None], np.cos(lat)[:, None]
+    e = -sl * s[0] + cl * s[1]
+    n = -sb * cl * s[0] - sb * sl * s[1] + cb * s[2]
+    u = cb * cl * s[0] + cb * sl * s[1] + sb * s[2]
+    return np.arctan2(e, n), np.arcsin(np.clip(u, -1.0, 1.0))
+
+
+def element_beam_lunar(coeffs, ra, dec, lon, lat, jd, freq):
+    """Per-station lunar element E-Jones [N, S, 2, 2]
+    (cspice_element_beam_lunar): sources below the lunar horizon get a
+    zero response (the reference's el<0 'invalid' branch)."""
+    az, el = lunar_station_azel(ra, dec, lon, lat, jd)
+    N, S = az.shape
+    E = element_beam(coeffs, az.reshape(-1), el.reshape(-1), freq)
+    E = E.reshape(N, S, 2, 2)
+    E[torch.from_numpy(el < 0)] = 0
+    return E

@@ -177,3 +177,27 @@ class TestLunarFrames:
         assert z['u'].shape == u_before.shape
         assert not np.allclose(z['u'], u_before)
         assert np.isfinite(z['u']).all()
+
+    def test_element_beam_lunar(self):
+        from sagecal_amd import beams, coords
+        jd = 2460200.75
+        coeffs = beams.make_synthetic_element_coeffs(n0=3)
+        ra = np.array([0.5, 2.0])
+        dec = np.array([0.1, -0.9])
+        # stations: one at the sub-source point of src 0 (zenith), one
+        # antipodal to it (below horizon)
+        slon, slat = coords.lunar_radec_to_latlon(ra[0], dec[0], jd)
+        lon = np.array([slon, slon + np.pi])
+        lat = np.array([slat, -slat])
+        az, el = beams.lunar_station_azel(ra, dec, lon, lat, jd)
+        assert az.shape == (2, 2)
+        assert abs(el[0, 0] - np.pi / 2) < 1e-9
+        assert el[1, 0] < 0
+        E = beams.element_beam_lunar(coeffs, ra, dec, lon, lat, jd, 150e6)
+        assert E.shape == (2, 2, 2, 2)
+        # zenith response matches a direct zenith evaluation
+        Ez = beams.element_beam(coeffs, np.array([az[0, 0]]),
+                                np.array([np.pi / 2]), 150e6)[0]
+        assert torch.allclose(E[0, 0], Ez)
+        # below-horizon response is zeroed
+        assert float(E[1, 0].abs().max()) == 0.0
