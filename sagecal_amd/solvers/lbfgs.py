@@ -210,3 +210,183 @@ def lbfgs_fit_bounded(fg, p0, lbound, ubound, maxiter=50, m=7, gtol=1e-9,
         mem.push(pn - p, gn - g)
         p, f, g = pn, fn, gn
     return p, mem, {'f1': float(f)}
+
+
+def lbfgsb_fit(fg, p0, lbound, ubound, maxiter=100, m=7, gtol=1e-9,
+               ftol=1e-12):
+    """Full Byrd–Lu–Nocedal–Zhu L-BFGS-B (the reference's lbfgsb.c):
+    generalized Cauchy point along the projected-gradient path + direct
+    primal subspace minimization over the free variables using the
+    compact representation B = theta*I - W M W^T, then a projected Armijo
+    line search. `lbfgs_fit_bounded` remains as the cheap
+    projected-gradient variant for well-conditioned problems.
+
+    fg(p) -> (f, grad); returns (p, info) with info['f1'], info['iters'].
+    Validated against scipy.optimize L-BFGS-B in tests/test_lbfgsb.py.
+    """
+    dt, dev = p0.dtype, p0.device
+    lb = torch.as_tensor(lbound, dtype=dt, device=dev).expand_as(p0).clone()
+    ub = torch.as_tensor(ubound, dtype=dt, device=dev).expand_as(p0).clone()
+    n = p0.numel()
+
+    def proj(x):
+        return torch.minimum(torch.maximum(x, lb), ub)
+
+    x = proj(p0.clone())
+    f, g = fg(x)
+    Slist, Ylist = [], []
+    theta = 1.0
+    eps = torch.finfo(dt).eps
+
+    def compact():
+        """W [n,2k], M [2k,2k] with B = theta I - W M W^T."""
+        if not Slist:
+            return None, None
+        S = torch.stack(Slist, dim=1)          # [n, k]
+        Y = torch.stack(Ylist, dim=1)
+        SY = S.T @ Y                           # [k, k]
+        D = torch.diag(torch.diagonal(SY))
+        L = torch.tril(SY, diagonal=-1)
+        StS = S.T @ S
+        k = S.shape[1]
+        Mi = torch.zeros(2 * k, 2 * k, dtype=dt, device=dev)
+        Mi[:k, :k] = -D
+        Mi[:k, k:] = L.T
+        Mi[k:, :k] = L
+        Mi[k:, k:] = theta * StS
+        W = torch.cat([Y, theta * S], dim=1)   # [n, 2k]
+        return W, torch.linalg.inv(Mi)
+
+    info = {'f1': float(f), 'iters': 0}
+    for it in range(maxiter):
+        pg = x - proj(x - g)                   # projected gradient
+        if float(pg.abs().max()) < gtol:
+            break
+        W, M = compact()
+
+        # ---- generalized Cauchy point (Alg. CP) ----
+        t_brk = torch.full_like(x, float('inf'))
+        neg, pos = g < 0, g > 0
+        t_brk[neg] = (x[neg] - ub[neg]) / g[neg]
+        t_brk[pos] = (x[pos] - lb[pos]) / g[pos]
+        d = torch.where(t_brk > 0, -g, torch.zeros_like(g))
+        xcp = x.clone()
+        if W is not None:
+            p = W.T @ d
+            c = torch.zeros_like(p)
+        fp = -float(d @ d)
+        if W is not None:
+            fpp = -theta * fp - float(p @ (M @ p))
+        else:
+            fpp = -theta * fp
+        fpp = max(fpp, eps)
+        dt_min = -fp / fpp
+        t_old = 0.0
+        order = torch.argsort(t_brk)
+        for bi in order.tolist():
+            t_b = float(t_brk[bi])
+            if t_b <= 0 or not np_isfinite(t_b):
+                if t_b <= 0:
+                    continue
+                break
+            delta = t_b - t_old
+            if dt_min < delta:
+                break
+            # hit breakpoint bi: fix at its bound
+            t_old = t_b
+            xcp[bi] = ub[bi] if g[bi] < 0 else lb[bi]
+            z_b = float(xcp[bi] - x[bi])
+            g_b = float(g[bi])
+            if W is not None:
+                c = c + delta * p
+                w_b = W[bi]
+                wMc = float(w_b @ (M @ c))
+                wMp = float(w_b @ (M @ p))
+                wMw = float(w_b @ (M @ w_b))
+            else:
+                wMc = wMp = wMw = 0.0
+            fp += delta * fpp + g_b * g_b + theta * g_b * z_b - g_b * wMc
+            fpp += -theta * g_b * g_b - 2.0 * g_b * wMp - g_b * g_b * wMw
+            fpp = max(fpp, eps)
+            if W is not None:
+                p = p + g_b * w_b
+            d[bi] = 0.0
+            dt_min = -fp / fpp
+            if fp >= 0:
+                dt_min = 0.0
+                break
+        dt_min = max(dt_min, 0.0)
+        t_final = t_old + dt_min
+        move = t_brk > t_old                   # still-moving coordinates
+        xcp[move] = proj(x + t_final * d)[move]
+        if W is not None:
+            c = c + dt_min * p
+
+        # ---- subspace minimization over free variables at xcp ----
+        free = (xcp > lb) & (xcp < ub)
+        xn_target = xcp
+        if bool(free.any()):
+            r = (g + theta * (xcp - x))
+            if W is not None:
+                r = r - W @ (M @ c)
+            rf = r[free]
+            if W is not None:
+                Wf = W[free]
+                k2 = W.shape[1]
+                K = torch.eye(k2, dtype=dt, device=dev) \
+                    - (M @ (Wf.T @ Wf)) / theta
+                v = torch.linalg.solve(K, M @ (Wf.T @ rf))
+                du = -(rf / theta + (Wf @ v) / theta ** 2)
+            else:
+                du = -rf / theta
+            # truncate to the box
+            xf = xcp[free]
+            lbf, ubf = lb[free], ub[free]
+            with torch.no_grad():
+                pos_d = du > 0
+                neg_d = du < 0
+                amax = torch.ones_like(du)
+                amax[pos_d] = (ubf[pos_d] - xf[pos_d]) / du[pos_d]
+                amax[neg_d] = (lbf[neg_d] - xf[neg_d]) / du[neg_d]
+                alpha_max = float(amax.clamp(min=0.0).min()) \
+                    if du.numel() else 1.0
+            xn_target = xcp.clone()
+            xn_target[free] = xf + min(1.0, alpha_max) * du
+
+        # ---- projected Armijo backtracking from x toward xn_target ----
+        dstep = xn_target - x
+        gTd = float(g @ dstep)
+        if gTd > 0:                            # not a descent dir: fall back
+            dstep = -pg
+            gTd = float(g @ dstep)
+        alpha, fn, xn, gn = 1.0, None, None, None
+        for _ in range(30):
+            xt = proj(x + alpha * dstep)
+            ft, gt = fg(xt)
+            if float(ft) <= float(f) + 1e-4 * alpha * gTd or \
+                    float(ft) < float(f) - abs(float(f)) * 1e-16:
+                fn, xn, gn = ft, xt, gt
+                break
+            alpha *= 0.5
+        if xn is None:
+            break
+        s, y = xn - x, gn - g
+        sy = float(s @ y)
+        if sy > 1e-12 * float(s.norm()) * float(y.norm()) + 1e-300:
+            Slist.append(s)
+            Ylist.append(y)
+            if len(Slist) > m:
+                Slist.pop(0)
+                Ylist.pop(0)
+            theta = float(y @ y) / sy
+        fprev = float(f)
+        x, f, g = xn, fn, gn
+        info['iters'] = it + 1
+        if abs(fprev - float(f)) <= ftol * max(1.0, abs(float(f))):
+            break
+    info['f1'] = float(f)
+    return x, info
+
+
+def np_isfinite(v):
+    return v == v and v not in (float('inf'), float('-inf'))
