@@ -48,3 +48,45 @@ def manifold_average_projectback(J_bands, niter=2):
     out = torch.stack([Javg @ U[f].conj().T
                        for f in range(J_bands.shape[0])])
     return out, Javg
+
+
+def extract_phases(J, niter=2):
+    """Phase-only reduction of solutions with a common unitary ambiguity
+    (extract_phases, manifold_average.c:400): resolve the ambiguity by
+    Jacobi joint diagonalization (Cardoso-Souloumiac sweeps: for each
+    off-diagonal position, the Givens rotation maximizing the summed
+    diagonal energy comes from the top eigenvector of a real 3x3 H =
+    sum Re(h h^H), h = conj([a00-a11, a01+a10, i(a10-a01)])), then keep
+    only exp(i*arg) of the diagonal.
+
+    J: [N, 2, 2] complex -> [N, 2, 2] with unit-modulus diagonal, zero
+    off-diagonals."""
+    Jw = J.clone().to(torch.complex128)
+    N = Jw.shape[0]
+    for _ in range(niter):
+        for which in (0, 1):
+            a00, a01 = Jw[:, 0, 0], Jw[:, 0, 1]
+            a10, a11 = Jw[:, 1, 0], Jw[:, 1, 1]
+            if which == 0:
+                h = torch.stack([a00 - a11, a01 + a10,
+                                 1j * (a10 - a01)], dim=1).conj()
+            else:
+                h = torch.stack([a11 - a00, a10 + a01,
+                                 1j * (a01 - a10)], dim=1).conj()
+            H = (h.unsqueeze(2) * h.conj().unsqueeze(1)).sum(dim=0).real
+            w, V = torch.linalg.eigh(H)
+            z = V[:, -1]
+            if float(z[0]) >= 0.0:
+                c = torch.sqrt(0.5 + 0.5 * z[0]).to(torch.complex128)
+                s = 0.5 * (z[1] - 1j * z[2]) / c
+            else:
+                c = torch.sqrt(0.5 - 0.5 * z[0]).to(torch.complex128)
+                s = 0.5 * (-z[1] + 1j * z[2]) / c
+            G = torch.stack([torch.stack([c, -s]),
+                             torch.stack([s.conj(), c.conj()])])
+            Jw = Jw @ G.conj().T
+    out = torch.zeros_like(Jw)
+    for d in (0, 1):
+        ph = torch.angle(Jw[:, d, d])
+        out[:, d, d] = torch.complex(torch.cos(ph), torch.sin(ph))
+    return out.to(J.dtype)

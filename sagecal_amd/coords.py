@@ -224,3 +224,34 @@ def moon_position_j2000(jd):
     z = (np.cos(bet) * np.sin(lam) * np.sin(eps)
          + np.sin(bet) * np.cos(eps))
     return np.array([x, y, z])
+
+
+def precession_matrix(jd):
+    """IAU-1976 precession rotation from J2000 mean equinox to the mean
+    equinox of date `jd` (the role of the reference's NOVAS-derived
+    get_precession_params/precession, transforms.c; used to precess
+    catalogue positions to apparent, data.cpp:1616):
+    R = Rz(-z) Ry(theta) Rz(-zeta), v_date = R @ v_J2000."""
+    T = (np.asarray(jd, dtype=float) - _J2000_JD) / 36525.0
+    arc = np.deg2rad(1.0 / 3600.0)
+    zeta = (2306.2181 * T + 0.30188 * T ** 2 + 0.017998 * T ** 3) * arc
+    z = (2306.2181 * T + 1.09468 * T ** 2 + 0.018203 * T ** 3) * arc
+    th = (2004.3109 * T - 0.42665 * T ** 2 - 0.041833 * T ** 3) * arc
+
+    def Rz(a):
+        ca, sa = np.cos(a), np.sin(a)
+        return np.array([[ca, sa, 0.0], [-sa, ca, 0.0], [0.0, 0.0, 1.0]])
+
+    def Ry(a):
+        ca, sa = np.cos(a), np.sin(a)
+        return np.array([[ca, 0.0, -sa], [0.0, 1.0, 0.0], [sa, 0.0, ca]])
+    return Rz(-z) @ Ry(th) @ Rz(-zeta)
+
+
+def precess_radec(ra, dec, jd):
+    """(ra, dec) at J2000 -> mean of date jd."""
+    v = np.array([np.cos(dec) * np.cos(ra), np.cos(dec) * np.sin(ra),
+                  np.sin(dec)])
+    p = precession_matrix(jd) @ v
+    return float(np.arctan2(p[1], p[0]) % (2 * np.pi)), \
+        float(np.arcsin(np.clip(p[2], -1, 1)))

@@ -77,3 +77,54 @@ def test_synthetic_sky():
     clusters = sky.build_clusters(srcs, clist, 0.0, np.pi / 4, 150e6)
     assert len(clusters) == 3
     assert clusters[0].nsrc == 2
+
+
+def test_precession_matches_known_rates():
+    """IAU-1976 precession: equinox drifts ~50.29 arcsec/yr along the
+    ecliptic; a point on the equator at ra=0 precesses by m ~ 46.12
+    arcsec/yr (+1.2815 deg/cy) in ra and n ~ 20.04 arcsec/yr
+    (+0.5567 deg/cy) in dec (standard textbook rates)."""
+    from sagecal_amd import coords
+    M = coords.precession_matrix(2451545.0 + 36525.0)
+    assert np.allclose(M @ M.T, np.eye(3), atol=1e-12)
+    ra, dec = coords.precess_radec(0.0, 0.0, 2451545.0 + 36525.0)
+    assert abs(np.degrees(ra) - 1.2815) < 0.01
+    assert abs(np.degrees(dec) - 0.5567) < 0.01
+    # identity at J2000
+    ra0, dec0 = coords.precess_radec(1.0, 0.5, 2451545.0)
+    assert abs(ra0 - 1.0) < 1e-12 and abs(dec0 - 0.5) < 1e-12
+
+
+def test_extract_phases_joint_diagonalization():
+    """Diagonal phase Jones obscured by a COMMON unitary: extract_phases
+    must undo the ambiguity and return the diagonal phases, preserving
+    station-to-station phase differences (manifold_average.c:400)."""
+    import torch
+    from sagecal_amd.consensus.manifold import extract_phases
+    rng = np.random.default_rng(7)
+    N = 6
+    ph = rng.uniform(-np.pi, np.pi, (N, 2))
+    D = torch.zeros(N, 2, 2, dtype=torch.complex128)
+    D[:, 0, 0] = torch.tensor(np.exp(1j * ph[:, 0]))
+    D[:, 1, 1] = torch.tensor(np.exp(1j * ph[:, 1]))
+    # common unitary ambiguity
+    A = torch.tensor(rng.standard_normal((2, 2))
+                     + 1j * rng.standard_normal((2, 2)))
+    U, _ = torch.linalg.qr(A)
+    J = D @ U.conj().T
+    P = extract_phases(J, niter=10)
+    # off-diagonals zero, unit modulus diagonals
+    assert float(P[:, 0, 1].abs().max()) == 0.0
+    assert torch.allclose(P[:, 0, 0].abs(),
+                          torch.ones(N, dtype=torch.float64))
+    # phase DIFFERENCES between stations are recovered (up to a common
+    # per-column offset and possible column permutation)
+    for col in (0, 1):
+        got = np.angle(P[:, col, col].numpy())
+        for truecol in (0, 1):
+            want = ph[:, truecol]
+            d = np.angle(np.exp(1j * (got - want)))
+            if np.std(d - d[0]) < 1e-6:
+                break
+        else:
+            raise AssertionError("phase differences not recovered")
