@@ -85,7 +85,8 @@ def load_context(args):
     ignore = skymod.read_ignore_file(args.ignfile) if args.ignfile else ()
     clusters = skymod.read_sky_cluster(args.sky, args.cluster, ms.ra0,
                                        ms.dec0, ms.freq0, fmt=args.format,
-                                       ignore_ids=ignore)
+                                       ignore_ids=ignore,
+                                       jd=getattr(ms, 'jd0', None))
     pack = SourcePack(clusters)
     return ms, pack, clusters, device, dtype
 

@@ -80,7 +80,8 @@ def main(argv=None):
     ms = msdata.open_ms(my_ms, tilesz=args.tilesz, device=device,
                         dtype=dtype)
     clusters = skymod.read_sky_cluster(args.sky, args.cluster, ms.ra0,
-                                       ms.dec0, ms.freq0, fmt=args.format)
+                                       ms.dec0, ms.freq0, fmt=args.format,
+                                       jd=getattr(ms, 'jd0', None))
     pack = SourcePack(clusters)
     state = sage.CalState(pack, ms.N, device=device, dtype=cdtype)
 

@@ -231,6 +231,7 @@ class NpzMS:
         self.ra0 = float(z['ra0'])
         self.dec0 = float(z['dec0'])
         self.tilesz = int(tilesz or z.get('tilesz', 10))
+        self.jd0 = float(z['jd0']) if 'jd0' in z else None
         rows = z['u'].shape[0]
         self.Ntime = rows // self.Nbase
         self._z = {k: z[k] for k in z.files}
@@ -380,6 +381,12 @@ class CasaMS:
         self.Ntime = rows // self.Nbase
         self.tdelta = float(np.asarray(
             self.tab.getcol('EXPOSURE'))[self._sel[0]])
+        # casacore TIME is MJD seconds -> JD of the first row
+        try:
+            t0 = float(np.asarray(self.tab.getcol('TIME'))[self._sel[0]])
+            self.jd0 = t0 / 86400.0 + 2400000.5
+        except (KeyError, TypeError):
+            self.jd0 = None
         self.tilesz = int(tilesz)
         self.pairs = np.stack([a1[self._sel[:self.Nbase]],
                                a2[self._sel[:self.Nbase]]], axis=1)

@@ -251,9 +251,18 @@ def build_clusters(sources, cluster_list, ra0, dec0, freq0,
 
 
 def read_sky_cluster(sky_path, cluster_path, ra0, dec0, freq0, fmt=0,
-                     ignore_ids=()):
-    """One-call equivalent of reference read_sky_cluster (readsky.c:195)."""
+                     ignore_ids=(), jd=None):
+    """One-call equivalent of reference read_sky_cluster (readsky.c:195).
+
+    jd: if given, precess the J2000 catalogue positions AND the phase
+    centre to the mean equinox of that epoch before computing direction
+    cosines (the reference does this from the MS time, data.cpp:1616)."""
     sources = read_sky_model(sky_path, fmt=fmt)
+    if jd is not None:
+        from . import coords
+        for s in sources:
+            s.ra, s.dec = coords.precess_radec(s.ra, s.dec, jd)
+        ra0, dec0 = coords.precess_radec(ra0, dec0, jd)
     clist = read_cluster_file(cluster_path)
     return build_clusters(sources, clist, ra0, dec0, freq0, ignore_ids)
 
