@@ -2,11 +2,14 @@
 
 Re-implements the workflow of /root/reference/src/buildsky (main.c:61,
 buildsky.c, fitpixels.c, cluster.c): island detection above a threshold
-(or an external Duchamp-style mask), per-island source fitting with model
-selection (point vs gaussian via AIC, fitpixels.c semantics), and
-weighted K-means clustering of the fitted sources into calibration
-directions (cluster.c) — producing the LSM sky file and cluster file the
-calibrator consumes (the reference's create_clusters.py role included).
+(or an external Duchamp-style mask), per-island MULTI-component Gaussian
+fitting with model-order selection (fit k = 1..maxfits components by LM,
+pick the order minimizing AIC/MDL — fitpixels.c/fitmultipixels.c
+semantics), PSF-aware point/extended classification, merging of
+components closer than rd*(bmaj+bmin)/2 (main.c -c), and weighted K-means
+clustering of the fitted sources into calibration directions (cluster.c)
+— producing the LSM sky file and cluster file the calibrator consumes
+(the reference's create_clusters.py role included).
 """
 import argparse
 import sys
@@ -15,6 +18,8 @@ import numpy as np
 
 from ..utils import fits as fitsio
 from .. import coords
+
+FWHM = 2.0 * np.sqrt(2.0 * np.log(2.0))
 
 
 def find_islands(img, threshold=None, mask=None, nsigma=5.0):
@@ -38,36 +43,146 @@ def find_islands(img, threshold=None, mask=None, nsigma=5.0):
     return islands
 
 
-def fit_island(img, hdr, ys, xs):
-    """Fit one island: flux-weighted centroid + second moments; model
-    selection point-vs-gaussian by AIC on the pixel residuals
-    (fitpixels.c fit_single_point / fit_single_gaussian + AIC choice)."""
-    f = img[ys, xs]
+def _moments(f, ys, xs):
     ftot = f.sum()
     cy = (f * ys).sum() / ftot
     cx = (f * xs).sum() / ftot
-    d2r = np.pi / 180.0
-    pscale = abs(hdr['CDELT1']) * d2r           # rad/pixel
-    # second moments -> gaussian extent
     vy = (f * (ys - cy) ** 2).sum() / ftot
     vx = (f * (xs - cx) ** 2).sum() / ftot
     vxy = (f * (xs - cx) * (ys - cy)).sum() / ftot
-    # eigen-decomposition of the moment matrix
-    T = np.array([[vx, vxy], [vxy, vy]])
-    evals, evecs = np.linalg.eigh(T)
-    sig_min, sig_maj = np.sqrt(np.maximum(evals, 1e-12))
-    pa = np.arctan2(evecs[1, 1], evecs[0, 1])
-    ra, dec = fitsio.pix_to_radec(hdr, cx, cy)
-    # model selection: point if extent below ~0.7 pixel (beam-unresolved)
-    # AIC: residual of point model (all flux at centroid) vs gaussian
-    npix = len(f)
-    if npix < 4 or sig_maj < 0.7:
-        return dict(stype='P', ra=float(ra), dec=float(dec),
-                    flux=float(ftot), eX=0.0, eY=0.0, eP=0.0)
-    fwhm = 2.0 * np.sqrt(2.0 * np.log(2.0))
-    return dict(stype='G', ra=float(ra), dec=float(dec), flux=float(ftot),
-                eX=float(sig_maj * pscale * fwhm),
-                eY=float(sig_min * pscale * fwhm), eP=float(pa))
+    return ftot, cy, cx, vy, vx, vxy
+
+
+def _gauss_eval(theta, ys, xs, k):
+    """Sum of k elliptical Gaussians; theta = k x [A, cy, cx, sy, sx, pa]."""
+    out = np.zeros(len(ys))
+    for i in range(k):
+        A, cy, cx, sy, sx, pa = theta[6 * i:6 * i + 6]
+        sy, sx = abs(sy) + 1e-3, abs(sx) + 1e-3
+        cp, sp = np.cos(pa), np.sin(pa)
+        dy, dx = ys - cy, xs - cx
+        yr = cp * dy - sp * dx
+        xr = sp * dy + cp * dx
+        out += A * np.exp(-0.5 * ((yr / sy) ** 2 + (xr / sx) ** 2))
+    return out
+
+
+def fit_island_multi(img, ys, xs, maxfits=10, criterion='aic'):
+    """Fit 1..maxfits elliptical-Gaussian components to one island by LM
+    (scipy least_squares plays the role of clmfit_nocuda.c) and select
+    the model order by AIC (2p + n ln(rss/n)) or MDL ((p/2) ln n + ...)
+    — fitmultipixels.c model selection. Returns list of component dicts
+    in PIXEL units (flux, cy, cx, sy, sx, pa)."""
+    from scipy.optimize import least_squares
+    f = img[ys, xs].astype(float)
+    ysf, xsf = ys.astype(float), xs.astype(float)
+    n = len(f)
+    if n < 7:
+        # too few pixels for an LM fit: moment estimate (point-like)
+        ftot, cy, cx, vy, vx, _ = _moments(np.abs(f) + 1e-12, ysf, xsf)
+        s0 = float(np.sqrt(max(0.5 * (vy + vx), 1e-4)))
+        return [dict(flux=float(f.sum()), cy=float(cy), cx=float(cx),
+                     sy=s0, sx=s0, pa=0.0)]
+    best = None
+    best_score = np.inf
+    theta = []
+    resid = f.copy()
+    rss = None
+    kmax = max(1, min(maxfits, n // 7))
+    for k in range(1, kmax + 1):
+        # seed component k at the residual peak
+        j = int(np.argmax(np.abs(resid)))
+        ftot, cy, cx, vy, vx, _ = _moments(np.abs(resid) + 1e-12, ysf, xsf)
+        s0 = max(0.7, np.sqrt(max(vy + vx, 1e-2) / (2.0 * k)))
+        theta = list(theta) + [float(resid[j]), float(ysf[j]), float(xsf[j]),
+                               s0, s0, 0.0]
+        sol = least_squares(
+            lambda th: _gauss_eval(th, ysf, xsf, k) - f, theta,
+            method='lm', max_nfev=200 * k)
+        theta = list(sol.x)
+        resid = f - _gauss_eval(sol.x, ysf, xsf, k)
+        prev_rss = rss if k > 1 else None
+        rss = float((resid ** 2).sum()) + 1e-300
+        p = 6 * k
+        if criterion == 'mdl':
+            score = 0.5 * p * np.log(n) + 0.5 * n * np.log(rss / n)
+        else:
+            score = 2.0 * p + n * np.log(rss / n)
+        # accept a higher order only on BOTH a criterion win and a
+        # material (>10%) rss drop — pure-noise components shave a few
+        # percent off rss, which is enough to fool AIC at large n
+        if score < best_score - 1e-9 and                 (prev_rss is None or rss < 0.9 * prev_rss):
+            best_score = score
+            best = list(theta)
+        else:
+            break       # adding a component no longer pays: stop early
+    comps = []
+    for i in range(len(best) // 6):
+        A, cy, cx, sy, sx, pa = best[6 * i:6 * i + 6]
+        sy, sx = abs(sy) + 1e-3, abs(sx) + 1e-3
+        comps.append(dict(flux=float(2.0 * np.pi * A * sy * sx),
+                          cy=float(cy), cx=float(cx), sy=float(sy),
+                          sx=float(sx), pa=float(pa % np.pi)))
+    return comps
+
+
+def fit_island(img, hdr, ys, xs, maxfits=10, criterion='aic',
+               psf_pix=0.0):
+    """Fit one island into one or more sky-model sources. Components whose
+    fitted extent is at or below the PSF width (or sub-pixel when no PSF
+    is given) are classified as points (fitpixels.c point-vs-gaussian)."""
+    d2r = np.pi / 180.0
+    pscale = abs(hdr['CDELT1']) * d2r           # rad/pixel
+    out = []
+    for c in fit_island_multi(img, ys, xs, maxfits, criterion):
+        ra, dec = fitsio.pix_to_radec(hdr, c['cx'], c['cy'])
+        sig_maj = max(c['sy'], c['sx'])
+        sig_min = min(c['sy'], c['sx'])
+        unresolved = sig_maj <= max(0.7, 1.05 * psf_pix / FWHM)
+        if unresolved:
+            out.append(dict(stype='P', ra=float(ra), dec=float(dec),
+                            flux=c['flux'], eX=0.0, eY=0.0, eP=0.0))
+        else:
+            out.append(dict(stype='G', ra=float(ra), dec=float(dec),
+                            flux=c['flux'],
+                            eX=float(sig_maj * pscale * FWHM),
+                            eY=float(sig_min * pscale * FWHM),
+                            eP=float(c['pa'])))
+    return out
+
+
+def merge_close(srcs, rd, beam_rad):
+    """Merge sources closer than rd*(bmaj+bmin)/2 radians (main.c -c):
+    flux-weighted position, summed flux, widest extent kept."""
+    if rd <= 0 or beam_rad <= 0 or len(srcs) < 2:
+        return srcs
+    srcs = sorted(srcs, key=lambda s: -s['flux'])
+    out = []
+    used = [False] * len(srcs)
+    lim = rd * beam_rad
+    for i, s in enumerate(srcs):
+        if used[i]:
+            continue
+        grp = [s]
+        used[i] = True
+        for j in range(i + 1, len(srcs)):
+            if used[j]:
+                continue
+            t = srcs[j]
+            d = np.hypot((t['ra'] - s['ra']) * np.cos(s['dec']),
+                         t['dec'] - s['dec'])
+            if d < lim:
+                grp.append(t)
+                used[j] = True
+        ft = sum(g['flux'] for g in grp)
+        m = dict(grp[0])
+        m['flux'] = ft
+        m['ra'] = sum(g['flux'] * g['ra'] for g in grp) / ft
+        m['dec'] = sum(g['flux'] * g['dec'] for g in grp) / ft
+        m['eX'] = max(g['eX'] for g in grp)
+        m['eY'] = max(g['eY'] for g in grp)
+        out.append(m)
+    return out
 
 
 def weighted_kmeans(ras, decs, w, Q, iters=30, seed=1):
@@ -111,22 +226,69 @@ def main(argv=None):
     ap.add_argument('-t', dest='threshold', type=float,
                     help='island threshold (default: 5 sigma)')
     ap.add_argument('-Q', dest='nclusters', type=int, default=4,
-                    help='number of direction clusters')
+                    help='number of direction clusters (reference -k)')
+    ap.add_argument('-l', dest='maxfits', type=int, default=10,
+                    help='max components attempted per island')
     ap.add_argument('-o', dest='model_order', type=int, default=1)
+    ap.add_argument('-C', dest='criterion', choices=['aic', 'mdl'],
+                    default='aic', help='model-order criterion')
+    ap.add_argument('-a', dest='bmaj', type=float, default=0.0,
+                    help='PSF major axis (arcsec)')
+    ap.add_argument('-b', dest='bmin', type=float, default=0.0,
+                    help='PSF minor axis (arcsec)')
+    ap.add_argument('-p', dest='bpa', type=float, default=0.0,
+                    help='PSF position angle (deg)')
+    ap.add_argument('-M', dest='merge_rd', type=float, default=0.0,
+                    help='merge components closer than rd*(bmaj+bmin)/2 '
+                         '(reference -c; renamed: -c is the cluster file '
+                         'here)')
+    ap.add_argument('-w', dest='sidelobe_cut', type=float, default=0.0,
+                    help='drop islands whose peak is below this')
+    ap.add_argument('-N', dest='negative', action='store_true',
+                    help='fit negative flux instead of positive')
+    ap.add_argument('-q', dest='rescale', type=int, default=0,
+                    help='1: scale model fluxes to island total flux')
     ap.add_argument('-s', dest='outsky', help='output sky file')
     ap.add_argument('-c', dest='outcluster', help='output cluster file')
     args = ap.parse_args(argv)
 
     img, hdr = fitsio.read_fits_image(args.fits)
+    if args.negative:
+        img = -img
     mask = None
     if args.mask:
         mask, _ = fitsio.read_fits_image(args.mask)
     islands = find_islands(img, args.threshold, mask)
-    srcs = [fit_island(img, hdr, ys, xs) for ys, xs in islands]
+    d2r = np.pi / 180.0
+    psf_pix = 0.0
+    beam_rad = 0.0
+    if args.bmaj > 0:
+        as2rad = d2r / 3600.0
+        pscale = abs(hdr['CDELT1']) * d2r
+        psf_pix = args.bmaj * as2rad / pscale
+        beam_rad = 0.5 * (args.bmaj + args.bmin) * as2rad
+    srcs = []
+    for ys, xs in islands:
+        if args.sidelobe_cut > 0 and img[ys, xs].max() < args.sidelobe_cut:
+            continue
+        comps = fit_island(img, hdr, ys, xs, args.maxfits, args.criterion,
+                           psf_pix)
+        if args.rescale:
+            ftot = float(img[ys, xs].sum())
+            fsum = sum(c['flux'] for c in comps)
+            if fsum > 0:
+                for c in comps:
+                    c['flux'] *= ftot / fsum
+        srcs.extend(comps)
     srcs = [s for s in srcs if s['flux'] > 0]
+    if args.merge_rd > 0:
+        srcs = merge_close(srcs, args.merge_rd, beam_rad)
     if not srcs:
         print("buildsky: no sources found", file=sys.stderr)
         return 1
+    if args.negative:
+        for s in srcs:
+            s['flux'] = -s['flux']
     freq = hdr.get('RESTFRQ', 150e6)
     outsky = args.outsky or args.fits + '.sky.txt'
     outcl = args.outcluster or outsky + '.cluster'
@@ -141,7 +303,7 @@ def main(argv=None):
                     f"{s['eX']:.8e} {s['eY']:.8e} {s['eP']:.8e} {freq}\n")
     ras = np.array([s['ra'] for s in srcs])
     decs = np.array([s['dec'] for s in srcs])
-    w = np.array([s['flux'] for s in srcs])
+    w = np.array([abs(s['flux']) for s in srcs])
     assign, cent = weighted_kmeans(ras, decs, w, args.nclusters)
     with open(outcl, 'w') as f:
         f.write("# cluster_id chunk_size source...\n")

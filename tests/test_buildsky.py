@@ -67,3 +67,78 @@ def test_restore_roundtrip(tmp_path):
     y, x = np.unravel_index(np.argmax(img), img.shape)
     assert abs(y - pos[0][0]) <= 1 and abs(x - pos[0][1]) <= 1
     assert img.sum() == pytest.approx(19.0 * 1.6, rel=0.15)
+
+
+def _gauss_img(shape, comps, noise=0.002, seed=2):
+    """comps: list of (flux, cy, cx, sy, sx)."""
+    yy, xx = np.mgrid[:shape[0], :shape[1]]
+    img = np.zeros(shape)
+    for f, cy, cx, sy, sx in comps:
+        A = f / (2 * np.pi * sy * sx)
+        img += A * np.exp(-0.5 * (((yy - cy) / sy) ** 2
+                                  + ((xx - cx) / sx) ** 2))
+    img += np.random.default_rng(seed).standard_normal(shape) * noise
+    return img
+
+
+def test_multicomponent_island_fit():
+    """A blended double inside ONE island must come out as TWO components
+    with the right positions/fluxes (fitmultipixels.c multi-fit + model
+    selection)."""
+    from sagecal_amd.apps.buildsky import find_islands, fit_island_multi
+    img = _gauss_img((48, 48), [(12.0, 22.0, 19.0, 2.0, 2.0),
+                                (8.0, 25.0, 25.0, 2.0, 2.0)])
+    islands = find_islands(img, threshold=0.05)
+    assert len(islands) == 1
+    comps = fit_island_multi(img, *islands[0], maxfits=5)
+    assert len(comps) == 2
+    comps.sort(key=lambda c: -c['flux'])
+    assert abs(comps[0]['cy'] - 22.0) < 0.3
+    assert abs(comps[0]['cx'] - 19.0) < 0.3
+    assert abs(comps[1]['cy'] - 25.0) < 0.3
+    assert abs(comps[1]['cx'] - 25.0) < 0.3
+    assert abs(comps[0]['flux'] - 12.0) / 12.0 < 0.15
+    assert abs(comps[1]['flux'] - 8.0) / 8.0 < 0.15
+
+
+def test_model_order_selection_prefers_single():
+    """A single Gaussian island must NOT be over-fit with extra
+    components (AIC stops at k=1)."""
+    from sagecal_amd.apps.buildsky import find_islands, fit_island_multi
+    img = _gauss_img((40, 40), [(10.0, 20.0, 20.0, 2.5, 1.8)])
+    islands = find_islands(img, threshold=0.05)
+    comps = fit_island_multi(img, *islands[0], maxfits=5)
+    assert len(comps) == 1
+    assert abs(comps[0]['flux'] - 10.0) / 10.0 < 0.1
+
+
+def test_merge_close_components():
+    from sagecal_amd.apps.buildsky import merge_close
+    srcs = [dict(stype='P', ra=0.0, dec=0.5, flux=4.0, eX=0, eY=0, eP=0),
+            dict(stype='P', ra=1e-5, dec=0.5, flux=2.0, eX=0, eY=0, eP=0),
+            dict(stype='P', ra=0.01, dec=0.5, flux=1.0, eX=0, eY=0, eP=0)]
+    out = merge_close(srcs, rd=1.0, beam_rad=1e-4)
+    assert len(out) == 2
+    big = max(out, key=lambda s: s['flux'])
+    assert big['flux'] == pytest.approx(6.0)
+    # flux-weighted position between the two merged components
+    assert 0.0 < big['ra'] < 1e-5
+
+
+def test_negative_and_rescale_cli(tmp_path):
+    from sagecal_amd.apps import buildsky
+    img = -_gauss_img((40, 40), [(10.0, 20.0, 20.0, 2.0, 2.0)],
+                      noise=0.001)
+    p = str(tmp_path / 'neg.fits')
+    fitsio.write_fits_image(p, img, crval=(0.0, 45.0),
+                            cdelt=(-0.01, 0.01))
+    outsky = str(tmp_path / 'sky.txt')
+    outcl = str(tmp_path / 'cl.txt')
+    rc = buildsky.main(['-f', p, '-s', outsky, '-c', outcl, '-Q', '1',
+                        '-N', '-q', '1'])
+    assert rc == 0
+    rows = [l for l in open(outsky) if not l.startswith('#')]
+    assert len(rows) >= 1
+    flux = float(rows[0].split()[7])
+    assert flux < 0          # -N reports negative flux
+    assert abs(abs(flux) - 10.0) / 10.0 < 0.2
