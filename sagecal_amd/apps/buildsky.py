@@ -108,10 +108,19 @@ def fit_island_multi(img, ys, xs, maxfits=10, criterion='aic'):
             score = 0.5 * p * np.log(n) + 0.5 * n * np.log(rss / n)
         else:
             score = 2.0 * p + n * np.log(rss / n)
-        # accept a higher order only on BOTH a criterion win and a
-        # material (>10%) rss drop — pure-noise components shave a few
-        # percent off rss, which is enough to fool AIC at large n
-        if score < best_score - 1e-9 and                 (prev_rss is None or rss < 0.9 * prev_rss):
+        # accept a higher order only on a criterion win AND a material
+        # (>10%) rss drop AND physically sensible components — pure-noise
+        # components are sub-pixel spikes of negligible flux and can fool
+        # AIC/rss gates on small islands
+        fsum = float(np.abs(f).sum())
+        sane = all(
+            max(abs(theta[6 * i + 3]), abs(theta[6 * i + 4])) >= 0.5
+            and abs(2.0 * np.pi * theta[6 * i]
+                    * (abs(theta[6 * i + 3]) + 1e-3)
+                    * (abs(theta[6 * i + 4]) + 1e-3)) >= 0.02 * fsum
+            for i in range(k))
+        if (score < best_score - 1e-9 and sane
+                and (prev_rss is None or rss < 0.9 * prev_rss)):
             best_score = score
             best = list(theta)
         else:
