@@ -49,6 +49,10 @@ def build_argparser():
     ap.add_argument('-u', dest='spatial_alpha', type=float, default=0.0)
     ap.add_argument('-L', dest='nulow', type=float, default=2.0)
     ap.add_argument('-H', dest='nuhigh', type=float, default=30.0)
+    ap.add_argument('-M', dest='mdl', action='store_true',
+                    help='evaluate AIC/MDL over polynomial orders 1..-P '
+                         'after the first tile and print the suggestion '
+                         '(reference -M, sagecal_master.cpp:992)')
     ap.add_argument('-O', dest='outcol', default='residual')
     ap.add_argument('-V', dest='verbose', action='store_true')
     return ap
@@ -132,6 +136,28 @@ def main(argv=None):
         ms.write_column(args.outcol, ti, xres)
         if writer:
             writer.write_tile(state)
+        if args.mdl and ti == 0:
+            # gather every band's J and score polynomial orders 1..Npoly
+            from ..consensus import mdl as mdl_mod
+            Jflat = (state.J.reshape(state.Mt, -1)
+                     [[state.chunk_off[c] for c in range(state.M)]])
+            Jb = torch.view_as_real(Jflat).reshape(1, state.M, -1).double()
+            if world > 1:
+                gath = [torch.zeros_like(Jb) for _ in range(world)]
+                dist.all_gather(gath, Jb)
+                Jall = torch.cat(gath, dim=0)
+            else:
+                Jall = Jb
+            best, scores = mdl_mod.minimum_description_length(
+                torch.view_as_complex(
+                    Jall.reshape(world, state.M, -1, 2).contiguous()),
+                rho.double(), f0s.double(), freq0_global,
+                polytype=args.polytype, Kstart=1,
+                Kfinish=min(args.npoly, world))
+            if rank == 0:
+                for k, (mv, av) in sorted(scores.items()):
+                    print(f"MDL: Npoly={k}: MDL={mv:.3f} AIC={av:.3f}")
+                print(f"MDL: suggested Npoly={best}")
         if rank == 0 or args.verbose:
             print(f"rank {rank} tile {ti}: res {res0:.6f} -> {res1:.6f} "
                   f"rho[0]={float(adm.rho[0]):.2f}")

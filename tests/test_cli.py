@@ -112,7 +112,7 @@ def _mpi_worker(rank, world, tmpdir):
                        '-s', os.path.join(tmpdir, 'sky.txt'),
                        '-c', os.path.join(tmpdir, 'cluster.txt'),
                        '-t', '4', '-A', '4', '-P', '2', '-j', '3',
-                       '-e', '2', '-g', '8', '-r', '2.0'])
+                       '-e', '2', '-g', '8', '-r', '2.0', '-M'])
         assert rc == 0
     finally:
         if dist.is_initialized():
