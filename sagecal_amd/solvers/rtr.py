@@ -171,6 +171,12 @@ def rtr_solve(prob, J0, maxiter=20, rsd_iters=2, delta0=None,
     """Batched RTR (rtr_solve_nocuda, rtr_solve.c:1208). Returns (J, info).
 
     prob: lm.LMProblem (weights -> robust RTR; admm -> RTR-ADMM)."""
+    import os as _os
+    if (_os.environ.get('SAGECAL_RTR_GRAPH') == '1'
+            and _os.environ.get('SAGECAL_NO_GRAPH') != '1'
+            and J0.is_cuda and prob.layout is not None):
+        return rtr_solve_graphed(prob, J0, maxiter=maxiter,
+                                 rsd_iters=rsd_iters)
     N = prob.N
     n = prob.nchunk
     dev = prob.x.device
@@ -259,3 +265,178 @@ def nsd_solve(prob, J0, maxiter=40, lr=None):
     cost = obj.cost(J)
     return J, {'init_cost': init_cost, 'final_cost': cost,
                'niter': maxiter}
+
+
+# ---------------------------------------------------------------------------
+# hipGraph capture of the RTR loop (rtr_solve body with the host polls
+# removed): fixed RSD warmup + fixed outer iterations + fixed-length tCG.
+# Gated behind SAGECAL_RTR_GRAPH=1 (off by default until measured on
+# hardware); the poll-free body itself is validated on CPU in
+# tests/test_rtr.py::test_rtr_body_matches_eager.
+# ---------------------------------------------------------------------------
+
+def _tcg_fixed(obj, J, grad, delta, maxit):
+    """_tcg without the host-side early-exit poll (capture-safe)."""
+    n = J.shape[0]
+    eta = torch.zeros_like(J)
+    r = grad.clone()
+    d = -r
+    r0 = _inner(r, r)
+    rnorm0 = r0.sqrt()
+    done = torch.zeros(n, dtype=torch.bool, device=J.device)
+    tol = rnorm0 * torch.minimum(rnorm0, torch.full_like(rnorm0, 0.01))
+    for it in range(maxit):
+        Hd = obj.hess_vec(J, d)
+        dHd = _inner(d, Hd)
+        alpha = _inner(r, r) / dHd.clamp_min(1e-30)
+        eta_new = eta + alpha[:, None, None, None] * d
+        enorm = _inner(eta_new, eta_new).sqrt()
+        hit = ((dHd <= 0) | (enorm > delta)) & ~done
+        ee = _inner(eta, eta)
+        ed = _inner(eta, d)
+        dd = _inner(d, d)
+        tau = (-ed + (ed ** 2 + dd * (delta ** 2 - ee)).clamp_min(0)
+               .sqrt()) / dd.clamp_min(1e-30)
+        bnd = eta + tau[:, None, None, None] * d
+        eta_new = torch.where(hit[:, None, None, None], bnd, eta_new)
+        eta = torch.where(done[:, None, None, None], eta, eta_new)
+        done = done | hit
+        r = r + alpha[:, None, None, None] * Hd
+        rn = _inner(r, r)
+        done = done | (rn.sqrt() <= tol)
+        beta = rn / r0.clamp_min(1e-30)
+        d = -r + beta[:, None, None, None] * d
+        r0 = rn
+    return eta
+
+
+def _rtr_body(x, coh, bb, N, nchunk, chunk_rows, w, layout, J0, iw,
+              maxiter, rsd_iters, tcg_maxit, prob_admm):
+    """Capture-safe RTR: no host reads, fixed iteration counts. Converged
+    chunks keep iterating (masked no-ops), like the graphed LM body."""
+
+    class _P:
+        pass
+    p = _P()
+    p.x, p.coh, p.bb, p.N = x, coh, bb, N
+    p.nchunk, p.chunk_rows, p.layout = nchunk, chunk_rows, layout
+    p.weights = w
+    p.admm = prob_admm
+    J = J0.clone()
+    obj = _Objective(p, iw)
+    obj.refresh(J)
+    cost = obj.cost(J)
+    init_cost = cost.clone()
+    for _ in range(rsd_iters):
+        g = obj.grad(J)
+        gn = _inner(g, g)
+        Hg = obj.hess_vec(J, g)
+        step = (gn / _inner(g, Hg).clamp_min(1e-30))
+        Jn = J - step[:, None, None, None].to(J.real.dtype) * g
+        cn = obj.cost(Jn)
+        better = cn < cost
+        J = torch.where(better[:, None, None, None], Jn, J)
+        cost = torch.where(better, cn, cost)
+        obj.refresh(J)
+    g = obj.grad(J)
+    delta = 0.1 * _inner(J, J).sqrt().clamp_min(1.0)
+    delta_bar = delta * 8
+    for _ in range(maxiter):
+        eta = _tcg_fixed(obj, J, g, delta, tcg_maxit)
+        Jn = J + eta
+        cn = obj.cost(Jn)
+        Heta = obj.hess_vec(J, eta)
+        mdec = -(_inner(g, eta) + 0.5 * _inner(eta, Heta))
+        rho = (cost - cn) / mdec.clamp_min(1e-30)
+        accept = (rho > 0.1) & (cn < cost)
+        J = torch.where(accept[:, None, None, None], Jn, J)
+        cost = torch.where(accept, cn, cost)
+        delta = torch.where(rho < 0.25, delta * 0.25,
+                            torch.where(rho > 0.75,
+                                        torch.minimum(delta * 2.0,
+                                                      delta_bar), delta))
+        obj.refresh(J)
+        g = obj.grad(J)
+    return J, cost, init_cost
+
+
+_rtr_graph_cache = {}
+
+
+def _station_iw(bb, N, device, rdtype):
+    import numpy as np
+    cnt = np.bincount(bb.cpu().reshape(-1).numpy(), minlength=N)
+    return torch.tensor(1.0 / np.maximum(cnt, 1)).to(device=device,
+                                                     dtype=rdtype)
+
+
+def rtr_solve_graphed(prob, J0, maxiter=20, rsd_iters=2, tcg_maxit=12):
+    """Graph-captured RTR (one replay per solve). Same caching scheme as
+    lm._lm_solve_graphed: static input buffers keyed by shape+layout."""
+    from ..ops import dispatch as ops_d
+    x, coh, bb, N = prob.x, prob.coh, prob.bb, prob.N
+    nchunk = prob.nchunk
+    dev = x.device
+    B = x.shape[0]
+    key = (B, nchunk, N, maxiter, rsd_iters, tcg_maxit,
+           ops_d.obj_token(prob.layout), prob.admm is not None)
+    ent = _rtr_graph_cache.get(key)
+    if ent is None:
+        ent = {'x': torch.empty_like(x), 'coh': torch.empty_like(coh),
+               'J0': torch.empty_like(J0),
+               'w': torch.ones(B, dtype=torch.float32, device=dev),
+               'rows': torch.zeros(B, dtype=torch.long, device=dev),
+               'bb': bb,
+               'iw': _station_iw(bb, N, dev, x.real.dtype)}
+        admm_s = None
+        if prob.admm is not None:
+            rho_a, Y_a, BZ_a = prob.admm
+            ent['admm'] = (torch.empty_like(rho_a), torch.empty_like(Y_a),
+                           torch.empty_like(BZ_a))
+            admm_s = ent['admm']
+        ent['admm_s'] = admm_s
+        ent['x'].copy_(x)
+        ent['coh'].copy_(coh)
+        ent['J0'].copy_(J0)
+        if prob.weights is not None:
+            ent['w'].copy_(prob.weights.to(torch.float32))
+        if prob.chunk_rows is not None:
+            ent['rows'].copy_(prob.chunk_rows)
+        if admm_s is not None:
+            for d_, s_ in zip(admm_s, prob.admm):
+                d_.copy_(s_)
+        st = torch.cuda.Stream()
+        st.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(st):
+            for _ in range(2):
+                _rtr_body(ent['x'], ent['coh'], ent['bb'], N, nchunk,
+                          ent['rows'], ent['w'], prob.layout, ent['J0'],
+                          ent['iw'], maxiter, rsd_iters, tcg_maxit, admm_s)
+        torch.cuda.current_stream().wait_stream(st)
+        gobj = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(gobj):
+            outs = _rtr_body(ent['x'], ent['coh'], ent['bb'], N, nchunk,
+                             ent['rows'], ent['w'], prob.layout, ent['J0'],
+                             ent['iw'], maxiter, rsd_iters, tcg_maxit,
+                             admm_s)
+        ent['graph'] = gobj
+        ent['outs'] = outs
+        _rtr_graph_cache[key] = ent
+    ent['x'].copy_(x)
+    ent['coh'].copy_(coh)
+    ent['J0'].copy_(J0)
+    if prob.weights is not None:
+        ent['w'].copy_(prob.weights.to(torch.float32))
+    else:
+        ent['w'].fill_(1.0)
+    if prob.chunk_rows is not None:
+        ent['rows'].copy_(prob.chunk_rows)
+    else:
+        ent['rows'].zero_()
+    if ent['admm_s'] is not None:
+        for d_, s_ in zip(ent['admm_s'], prob.admm):
+            d_.copy_(s_)
+    ent['graph'].replay()
+    J, cost, init_cost = ent['outs']
+    return J.clone(), {'init_cost': init_cost.clone(),
+                       'final_cost': cost.clone(), 'niter': maxiter}

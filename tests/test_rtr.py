@@ -114,3 +114,25 @@ def test_rtr_admm_pulls_to_target():
     U = manifold.polar_unitary(A)
     err = float((torch.einsum('sij,jk->sik', J[0], U) - BZ[0]).abs().mean())
     assert err < 0.05, err
+
+
+def test_rtr_body_matches_eager():
+    """The capture-safe poll-free RTR body (_rtr_body, used under graph
+    capture) must converge like the eager rtr_solve on the same
+    problem (run eagerly on CPU here; the graph mechanics themselves are
+    GPU-only)."""
+    from sagecal_amd.solvers import rtr as rtr_mod
+    x, coh, bb, Jt, N = _problem()
+    prob = lm_mod.LMProblem(x, coh, bb, N, 1, None)
+    rng = np.random.default_rng(7)
+    J0 = Jt + 0.1 * torch.tensor(rng.standard_normal(Jt.shape)
+                                 + 1j * rng.standard_normal(Jt.shape))
+    J_eager, info = rtr_mod.rtr_solve(prob, J0, maxiter=12)
+    iw = rtr_mod._station_iw(prob.bb, prob.N, prob.x.device,
+                             prob.x.real.dtype)
+    J_body, cost, init_cost = rtr_mod._rtr_body(
+        prob.x, prob.coh, prob.bb, prob.N, prob.nchunk, prob.chunk_rows,
+        prob.weights, prob.layout, J0, iw, 12, 2, 12, prob.admm)
+    assert float(cost.max()) <= 1.05 * float(
+        info['final_cost'].max()) + 1e-6
+    assert float(cost.max()) < 1e-6 * float(init_cost.max())
