@@ -229,6 +229,48 @@ class ConsensusADMM:
             self.alpha = self.spatial_alpha
             self._update_bii()
 
+    def diffuse_station_series(self):
+        """Per-station Jones-valued spatial model as a shapelet series
+        [N, G, 2, 2] with its scale (for this band): the FISTA model Zsp
+        evaluated with this band's polynomial basis. Coefficients come in
+        the image_basis normalization (4 pi^2/beta^2 prefactor, argument
+        2 pi l / beta), so the raw-phi series scale is beta/(2 pi)
+        (sagecal_master.cpp spatial model -> diffuse_predict.c hand-off)."""
+        if self.Zspat is None or self._Phi is None:
+            return None, None
+        from . import fista as fista_mod
+        M = self.state.M
+        N = self.Z.shape[2]
+        Npoly = self.Npoly
+        # refit to recover the coefficient matrix (Zspat stores Phi @ Zsp)
+        lam, mu_l1, order, fiters, cadence = self.spatial
+        Zb = torch.view_as_real(self.Z).reshape(M, -1)
+        Zb = torch.complex(Zb[:, 0::2], Zb[:, 1::2])
+        Zsp = fista_mod.update_spatialreg_fista(
+            Zb.cpu(), self._Phi, lam=lam, mu=mu_l1, maxiter=fiters)
+        G = Zsp.shape[1]
+        Zsp = Zsp.reshape(Npoly, N, 2, 2, G)
+        Bf = self.Bf.to(Zsp.dtype)
+        Zband = torch.einsum('p,pnijg->ngij', Bf, Zsp)   # [N, G, 2, 2]
+        beta = float(max(np.max(np.abs(self.centroids[0])),
+                         np.max(np.abs(self.centroids[1])), 1e-3))
+        C0 = 4.0 * np.pi ** 2 / beta ** 2
+        return Zband * C0, beta / (2.0 * np.pi)
+
+    def diffuse_coherencies(self, u, v, w, bb, Cm, beta_c, lmn, freq,
+                            fdelta):
+        """Diffuse-cluster coherencies with the spatial model applied
+        (recalculate_diffuse_coherencies, diffuse_predict.c:295):
+        Zp C Zq^H per baseline via Jones shapelet products."""
+        from .. import shapelet as shmod
+        Z, bz = self.diffuse_station_series()
+        if Z is None:
+            return None
+        ll, mm, nn1 = lmn
+        return shmod.recalculate_diffuse_coherencies(
+            u.cpu().double(), v.cpu().double(), w.cpu().double(),
+            bb.cpu(), Z, bz, Cm, beta_c, ll, mm, nn1, freq, fdelta)
+
     def global_solution(self):
         """J = B_f Z (use_global_solution path, sagecal_master:1064)."""
         return self._chunk_expand(poly.eval_poly_jones(self.Z, self.Bf))

@@ -137,3 +137,118 @@ def decompose_image(l, m, flux, n0, beta):
     y = torch.as_tensor(flux, dtype=torch.float64)
     sol = torch.linalg.lstsq(Bmat, y.unsqueeze(-1)).solution.squeeze(-1)
     return sol.numpy()
+
+
+# ---------------------------------------------------------------------------
+# Shapelet products (shapelet.c shapelet_product* / diffuse_predict.c):
+# the pointwise product of two Gauss-Hermite series with scales alpha,
+# beta is EXACTLY a series with scale gamma, 1/gamma^2 = 1/alpha^2 +
+# 1/beta^2, of order i+j — the machinery that applies a station-dependent
+# Jones-valued spatial model Z(l,m) to a diffuse shapelet sky:
+# Jp(l,m) C(l,m) Jq(l,m)^H per baseline, all three shapelet series.
+# ---------------------------------------------------------------------------
+
+def product_scale(alpha, beta):
+    """gamma with 1/gamma^2 = 1/alpha^2 + 1/beta^2."""
+    return 1.0 / math.sqrt(1.0 / alpha ** 2 + 1.0 / beta ** 2)
+
+
+def shapelet_product_tensor(L, M_, N_, alpha, beta, gamma):
+    """1-D product-projection tensor T [N_, L, M_]:
+    phi_i(x/alpha) phi_j(x/beta) = sum_k T[k, i, j] phi_k(x/gamma)
+    (exact when N_ > L-1 + M_-1 and gamma = product_scale(alpha, beta)).
+    Computed by Gauss-Hermite quadrature (integrand is a polynomial times
+    the combined Gaussian, so the quadrature is exact)."""
+    import numpy.polynomial.hermite as H
+    # int phi_i^a phi_j^b phi_k^g dx ; phi normalization per hermite_phi
+    # orthonormality: int phi_m^g phi_n^g dx = (gamma sqrt(pi)/2) delta
+    a2 = 0.5 / alpha ** 2
+    b2 = 0.5 / beta ** 2
+    g2 = 0.5 / gamma ** 2
+    s = math.sqrt(a2 + b2 + g2)
+    deg = L + M_ + N_ + 2
+    t, wq = H.hermgauss(deg)
+    x = t / s
+    pa = hermite_phi(torch.tensor(x / alpha), L).numpy()
+    pb = hermite_phi(torch.tensor(x / beta), M_).numpy()
+    pg = hermite_phi(torch.tensor(x / gamma), N_).numpy()
+    # strip the Gaussians (hermgauss supplies exp(-t^2) itself)
+    pa = pa * np.exp(0.5 * a2 * 2 * (x ** 2))[:, None]
+    pb = pb * np.exp(0.5 * b2 * 2 * (x ** 2))[:, None]
+    pg = pg * np.exp(0.5 * g2 * 2 * (x ** 2))[:, None]
+    scale = 1.0 / s
+    T = np.einsum('q,qi,qj,qk->kij', wq * scale, pa, pb, pg)
+    return torch.tensor(T / (gamma * math.sqrt(math.pi) / 2.0))
+
+
+def shapelet_product_jones(F, G, alpha, beta, n_out=None, conj_g=False):
+    """Jones-valued 2-D series product: F [Lf*Lf, 2, 2] (scale alpha,
+    column-major modes j2*Lf+j1) times G [Lg*Lg, 2, 2] (scale beta) ->
+    H [n_out*n_out, 2, 2] at scale product_scale(alpha, beta)
+    (shapelet_product_jones, shapelet.c API Dirac_radio.h:420-430).
+    conj_g: use G^H (per-mode conjugate transpose — valid when G's
+    image-plane series is Hermitian-symmetric in the Jones sense)."""
+    Lf = int(math.isqrt(F.shape[0]))
+    Lg = int(math.isqrt(G.shape[0]))
+    if n_out is None:
+        n_out = Lf + Lg - 1
+    gamma = product_scale(alpha, beta)
+    T = shapelet_product_tensor(Lf, Lg, n_out, alpha, beta, gamma).to(
+        torch.complex128)
+    Fm = F.reshape(Lf, Lf, 2, 2).to(torch.complex128)
+    Gm = G.reshape(Lg, Lg, 2, 2).to(torch.complex128)
+    if conj_g:
+        Gm = Gm.conj().transpose(-1, -2)
+    # modes indexed [n2][n1]: contract each axis with its own T
+    Hm = torch.einsum('kac,lbd,abij,cdjm->klim', T, T, Fm, Gm)
+    return Hm.reshape(n_out * n_out, 2, 2), gamma
+
+
+def eval_uv_jones(u, v, modes, beta, a=1.0, b=1.0):
+    """uv-plane evaluation of a COMPLEX Jones-valued mode series:
+    [B, 2, 2] = sum_k modes[k] . basis_k(-u, v) * 2 pi a b (the complex
+    generalization of shapelet_contrib's mode sum; u, v in wavelengths)."""
+    n0 = int(math.isqrt(modes.shape[0]))
+    basis = uv_mode_vectors(-u, v, beta, n0)       # [B, n2, n1]
+    m = modes.reshape(n0, n0, 2, 2).to(torch.complex128)
+    out = torch.einsum('bkl,klij->bij', basis.to(torch.complex128), m)
+    return 2.0 * math.pi * a * b * out
+
+
+def recalculate_diffuse_coherencies(u, v, w, bb, Z, beta_z, Cm, beta_c,
+                                    ll, mm, nn1, freq, fdelta):
+    """Per-baseline coherencies of a diffuse shapelet cluster with a
+    per-station Jones-valued spatial model applied
+    (recalculate_diffuse_coherencies, diffuse_predict.c:295):
+      V_pq = [ Zp(l,m) C(l,m) Zq(l,m)^H ]^(uv)(u_pq, v_pq) * phase * smear
+    u,v,w: [B] seconds; bb: [B,2] station pairs; Z: [N, Gz*Gz, 2, 2]
+    per-station spatial series (scale beta_z); Cm: [Gc*Gc, 2, 2] sky
+    coherency series (scale beta_c); (ll, mm, nn1): source direction;
+    freq, fdelta: Hz. Returns [B, 2, 2] complex64/128."""
+    N = Z.shape[0]
+    # stage 1: C_Zq = C x Zq^H per station
+    CZ = []
+    for q in range(N):
+        h, g1 = shapelet_product_jones(Cm, Z[q], beta_c, beta_z,
+                                       conj_g=True)
+        CZ.append(h)
+    CZ = torch.stack(CZ)
+    # stage 2: H_pq = Zp x C_Zq  per needed pair (unique pairs of bb)
+    pairs = {}
+    for p, q in {(int(p), int(q)) for p, q in bb.tolist()}:
+        h, g2 = shapelet_product_jones(Z[p], CZ[q], beta_z, g1)
+        pairs[(p, q)] = h
+    uf = u * freq
+    vf = v * freq
+    G = 2.0 * math.pi * (u * ll + v * mm + w * nn1)
+    ph = torch.remainder(G * freq, 2.0 * math.pi)
+    phc = torch.complex(torch.cos(ph), torch.sin(ph))
+    smf = G * (fdelta * 0.5)
+    sm = torch.where(smf.abs() > 1e-12, (torch.sin(smf) / smf).abs(),
+                     torch.ones_like(smf))
+    out = torch.zeros(len(u), 2, 2, dtype=torch.complex128)
+    for (p, q), h in pairs.items():
+        sel = (bb[:, 0] == p) & (bb[:, 1] == q)
+        if bool(sel.any()):
+            out[sel] = eval_uv_jones(uf[sel], vf[sel], h, g2)
+    return out * (phc * sm)[:, None, None]

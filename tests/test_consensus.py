@@ -237,3 +237,36 @@ def test_admm_four_bands_poly_spatial_gloo(tmp_path, poly_type):
         res0, res1, zdiff = map(float, txt)
         assert res1 < 0.2 * res0, f"rank {rank}: {res0} -> {res1}"
         assert zdiff < 1e-6, f"Z not replicated: {zdiff}"
+
+
+def test_diffuse_coherencies_from_spatial_model():
+    """End-to-end: ADMM with the FISTA spatial model produces a
+    per-station Jones shapelet series, and diffuse_coherencies returns
+    finite baseline coherencies of the right shape (diffuse_predict.c
+    analog)."""
+    from sagecal_amd.consensus.admm import ConsensusADMM
+    from sagecal_amd.solvers import sage
+    from sagecal_amd.constants import SM_LM_LBFGS
+    pack, ms, tile, bb, cohs, Jtrue, freqs_all, f0 = _band_problem(0, 1)
+    state = sage.CalState(pack, ms.N)
+    opts = sage.SageSolveOptions(max_emiter=2, max_iter=8,
+                                 solver_mode=SM_LM_LBFGS, mode='batched')
+    cent = (np.array([0.001 * i for i in range(pack.M)]),
+            np.array([0.0005 * i for i in range(pack.M)]))
+    adm = ConsensusADMM(state, freqs_all, f0, 0, 1, Npoly=1,
+                        rho=torch.full((pack.M,), 1.0),
+                        spatial=(0.01, 1e-5, 2, 20, 1),
+                        spatial_alpha=0.0, centroids=cent)
+    adm.run(cohs, tile, bb, opts, n_admm=3)
+    Z, bz = adm.diffuse_station_series()
+    assert Z is not None and Z.shape[0] == ms.N and Z.shape[2:] == (2, 2)
+    rng = np.random.default_rng(0)
+    n0, beta_c = 2, 1e-3
+    Cm = torch.zeros(n0 * n0, 2, 2, dtype=torch.complex128)
+    Cm[:, 0, 0] = torch.tensor(rng.standard_normal(n0 * n0))
+    Cm[:, 1, 1] = Cm[:, 0, 0]
+    out = adm.diffuse_coherencies(tile.u, tile.v, tile.w, bb, Cm, beta_c,
+                                  (1e-3, 0.0, 0.0), float(f0), 0.0)
+    assert out.shape == (len(tile.u), 2, 2)
+    assert torch.isfinite(torch.view_as_real(out)).all()
+    assert float(out.abs().max()) > 0

@@ -88,3 +88,80 @@ def test_predict_with_shapelet_source(tmp_path):
     pack2.shapelets = {}
     coh2 = R.predict_coh(pack2, u, v, w, 150e6, 150e6, 0.0, 0.0, np.pi / 4)
     assert float((coh - coh2).abs().max()) > 1e-3
+
+
+def test_jones_product_exact_in_image_plane():
+    """Zp(l,m) C(l,m) Zq(l,m)^H as a mode series: the Jones shapelet
+    product (shapelet_product_jones) must reproduce the pointwise product
+    of the factor series on a grid to machine precision."""
+    rng = np.random.default_rng(3)
+    Gc, Gz = 3, 2
+    bc, bz = 1.1, 0.8
+
+    def render(modes, beta, x, y):
+        n0 = int(np.sqrt(modes.shape[0]))
+        px = shapelet.hermite_phi(torch.tensor(x / beta), n0)
+        py = shapelet.hermite_phi(torch.tensor(y / beta), n0)
+        m = torch.tensor(modes).reshape(n0, n0, 2, 2)
+        return torch.einsum('bk,bl,klij->bij',
+                            py.to(torch.complex128),
+                            px.to(torch.complex128), m)
+
+    C = rng.standard_normal((Gc * Gc, 2, 2)) \
+        + 1j * rng.standard_normal((Gc * Gc, 2, 2))
+    Zp = rng.standard_normal((Gz * Gz, 2, 2)) \
+        + 1j * rng.standard_normal((Gz * Gz, 2, 2))
+    Zq = rng.standard_normal((Gz * Gz, 2, 2)) \
+        + 1j * rng.standard_normal((Gz * Gz, 2, 2))
+    CZq, g1 = shapelet.shapelet_product_jones(
+        torch.tensor(C), torch.tensor(Zq), bc, bz, conj_g=True)
+    H, g2 = shapelet.shapelet_product_jones(torch.tensor(Zp), CZq, bz, g1)
+    x = np.linspace(-2.0, 2.0, 9)
+    y = np.linspace(-2.0, 2.0, 9)
+    xx, yy = np.meshgrid(x, y, indexing='ij')
+    A = render(np.asarray(Zp), bz, xx.ravel(), yy.ravel())
+    B = render(C, bc, xx.ravel(), yy.ravel())
+    Dq = render(np.asarray(Zq), bz, xx.ravel(), yy.ravel())
+    want = A @ B @ Dq.conj().transpose(-1, -2)
+    got = render(H.numpy(), g2, xx.ravel(), yy.ravel())
+    err = float((want - got).abs().max() / want.abs().max())
+    assert err < 1e-10, err
+
+
+def test_diffuse_coherencies_identity_gain():
+    """With Z = identity for every station, the diffuse-predict path must
+    equal the plain shapelet coherency of the same sky series."""
+    rng = np.random.default_rng(5)
+    N, n0, beta = 4, 3, 2e-3
+    modes = rng.standard_normal(n0 * n0)
+    # sky coherency series: unpolarized C = modes x I/2 per mode
+    Cm = torch.zeros(n0 * n0, 2, 2, dtype=torch.complex128)
+    Cm[:, 0, 0] = torch.tensor(modes, dtype=torch.complex128) * 0.5
+    Cm[:, 1, 1] = torch.tensor(modes, dtype=torch.complex128) * 0.5
+    # identity spatial model: only the (0,0) mode, normalized so the
+    # rendered Z(l,m) ~ phi_0(x)phi_0(y) ... use a wide scale and scale
+    # the mode so Z ~ I over the source extent
+    Gz, bz = 1, 50.0 * beta
+    phi0 = float(shapelet.hermite_phi(torch.zeros(1), 1)[0, 0])
+    Z = torch.zeros(N, 1, 2, 2, dtype=torch.complex128)
+    Z[:, 0, 0, 0] = 1.0 / phi0 ** 2
+    Z[:, 0, 1, 1] = 1.0 / phi0 ** 2
+    pairs = [(p, q) for p in range(N) for q in range(p + 1, N)]
+    bb = torch.tensor(pairs)
+    B = len(pairs)
+    freq = 150e6
+    u = torch.tensor(rng.standard_normal(B) * 50 / freq)
+    v = torch.tensor(rng.standard_normal(B) * 50 / freq)
+    w = torch.zeros(B)
+    ll, mm, nn1 = 1e-3, -2e-3, 0.0
+    got = shapelet.recalculate_diffuse_coherencies(
+        u, v, w, bb, Z, bz, Cm, beta, ll, mm, nn1, freq, 0.0)
+    env = shapelet.shapelet_contrib(
+        u * freq, v * freq, w * freq, 1.0, 1.0, 0.0, 1, 0, 1, 0, False,
+        beta, n0, modes)
+    G = 2.0 * np.pi * (u * ll + v * mm + w * nn1)
+    ph = torch.complex(torch.cos(G * freq), torch.sin(G * freq))
+    want = (env * ph)[:, None, None] * 0.5 \
+        * torch.eye(2, dtype=torch.complex128)
+    err = float((got - want).abs().max() / want.abs().max())
+    assert err < 5e-3, err
