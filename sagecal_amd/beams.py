@@ -67,32 +67,47 @@ def array_beam(cfg, ra, dec, freqs, tmjd, station_ids=None, device='cpu'):
     nsta = len(cfg.element_enu) if isinstance(cfg.element_enu,
                                               (list, tuple)) else 1
     if station_ids is None:
-        station_ids = range(nsta)
+        station_ids = list(range(nsta))
+    station_ids = list(station_ids)
+    elem_list = [cfg.elements(s) for s in station_ids]
+    counts = {np.asarray(e).shape[0] for e in elem_list}
+    fq = torch.as_tensor(freqs, dtype=torch.float64, device=device)
+    # per-time geometry (shared across stations: one array origin)
+    dus, belows = [], []
+    for g in np.atleast_1d(gmst):
+        usrc, az, el = _direction_enu(ra, dec, cfg.lon, cfg.lat, g)
+        upnt, _, _ = _direction_enu(cfg.b_ra0, cfg.b_dec0, cfg.lon,
+                                    cfg.lat, g)
+        dus.append(torch.as_tensor(usrc - upnt, dtype=torch.float64,
+                                   device=device))     # [K, 3]
+        belows.append(torch.as_tensor(el < 0, device=device))
+    du = torch.stack(dus)                              # [T, K, 3]
+    below = torch.stack(belows)                        # [T, K]
+    if len(counts) == 1:
+        # equal element counts: ONE batched pass — phases
+        # [S,T,E,K] = elems[S,E,3] . du[T,K,3], then exp+mean over E and
+        # the frequency axis broadcast; fused device-wide tensor ops
+        # instead of per-(station,time,freq) python loops (the role of
+        # kernel_array_beam's 2-D grid, predict_model.cu:129).
+        elems = torch.as_tensor(np.stack(elem_list), dtype=torch.float64,
+                                device=device)         # [S, E, 3]
+        proj = torch.einsum('sex,tkx->stek', elems, du)
+        ph = proj.unsqueeze(-1) * (2.0 * math.pi / C_LIGHT) * fq
+        af = torch.complex(torch.cos(ph), torch.sin(ph)).mean(dim=2)
+        af = torch.where(below[None, :, :, None],
+                         torch.zeros_like(af), af)     # [S, T, K, F]
+        return af.permute(0, 2, 1, 3).contiguous()     # [S, K, T, F]
+    # jagged element counts: per-station batched over (T, K, F)
     out = []
-    for s in station_ids:
-        elems = torch.as_tensor(cfg.elements(s), dtype=torch.float64,
+    for elems_np in elem_list:
+        elems = torch.as_tensor(elems_np, dtype=torch.float64,
                                 device=device)
-        rows = []
-        for ti, g in enumerate(np.atleast_1d(gmst)):
-            usrc, az, el = _direction_enu(ra, dec, cfg.lon, cfg.lat, g)
-            upnt, _, el0 = _direction_enu(cfg.b_ra0, cfg.b_dec0, cfg.lon,
-                                          cfg.lat, g)
-            du = torch.as_tensor(usrc - upnt, dtype=torch.float64,
-                                 device=device)        # [K,3]
-            proj = elems @ du.T                        # [E, K]
-            ph = (2.0 * math.pi / C_LIGHT) * proj
-            # sum over elements per frequency
-            af = []
-            for f in freqs:
-                pr = ph * f
-                af.append(torch.complex(torch.cos(pr), torch.sin(pr))
-                          .mean(dim=0))
-            below = torch.as_tensor(el < 0, device=device)
-            aff = torch.stack(af, dim=-1)              # [K, F]
-            aff = torch.where(below[:, None], torch.zeros_like(aff), aff)
-            rows.append(aff)
-        out.append(torch.stack(rows, dim=1))           # [K, T, F]
-    return torch.stack(out)                            # [S, K, T, F]
+        proj = torch.einsum('ex,tkx->tek', elems, du)
+        ph = proj.unsqueeze(-1) * (2.0 * math.pi / C_LIGHT) * fq
+        af = torch.complex(torch.cos(ph), torch.sin(ph)).mean(dim=1)
+        af = torch.where(below[:, :, None], torch.zeros_like(af), af)
+        out.append(af.permute(1, 0, 2))                # [K, T, F]
+    return torch.stack(out)
 
 
 def tile_beam(cfg, ra, dec, freqs, tmjd, station_ids=None, device='cpu'):
