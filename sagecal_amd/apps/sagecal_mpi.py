@@ -120,10 +120,17 @@ def main(argv=None):
         solver_mode=args.solver_mode, robust_nulow=args.nulow,
         robust_nuhigh=args.nuhigh)
     writer = None
+    zwriter = None
     if args.solfile:
         writer = solutions.SolutionWriter(
             f"{args.solfile}.rank{rank}", ms.freq0, ms.fdelta,
             ms.tilesz * ms.tdelta / 60.0, ms.N, state.M, state.Mt)
+        if rank == 0:
+            # global Z solution file (the master's write,
+            # sagecal_master.cpp:1165): Z is replicated on every rank
+            zwriter = solutions.GlobalZWriter(
+                f"{args.solfile}.Z", freq0_global, ms.fdelta,
+                ms.tilesz * ms.tdelta / 60.0, ms.N, state.M, adm.Npoly)
     bb = ms.bb_tensor(device=device)
     for ti, tile in enumerate(ms.tiles()):
         cohs = sage.precalc_coherencies(pack, tile)
@@ -136,6 +143,8 @@ def main(argv=None):
         ms.write_column(args.outcol, ti, xres)
         if writer:
             writer.write_tile(state)
+        if zwriter:
+            zwriter.write_tile(adm.Z)
         if args.mdl and ti == 0:
             # gather every band's J and score polynomial orders 1..Npoly
             from ..consensus import mdl as mdl_mod
@@ -164,6 +173,8 @@ def main(argv=None):
     ms.save()
     if writer:
         writer.close()
+    if zwriter:
+        zwriter.close()
     if world > 1:
         dist.destroy_process_group()
     return 0

