@@ -1,0 +1,116 @@
+"""Property-based tests (hypothesis) for the core math transforms the
+solvers are built on: vecR/realify layouts, the damped-solve padding
+contract, polynomial-basis invariants, and the shapelet product-scale
+algebra."""
+import math
+
+import numpy as np
+import torch
+from hypothesis import given, settings, strategies as st
+
+from sagecal_amd.ops import reference as R
+from sagecal_amd import shapelet
+from sagecal_amd.consensus import poly
+
+
+@st.composite
+def jones(draw, maxn=6):
+    n = draw(st.integers(1, maxn))
+    seed = draw(st.integers(0, 2 ** 31 - 1))
+    rng = np.random.default_rng(seed)
+    return torch.tensor(rng.standard_normal((1, n, 2, 2))
+                        + 1j * rng.standard_normal((1, n, 2, 2)))
+
+
+@given(jones())
+@settings(max_examples=30, deadline=None)
+def test_vecR_roundtrip(J):
+    """vecR is the row-major interleaved re/im layout; inverting it
+    recovers the Jones matrices exactly."""
+    v = R.vecR(J)                          # [1, N, 8]
+    back = torch.view_as_complex(
+        v.reshape(J.shape[0], J.shape[1], 2, 2, 2).contiguous())
+    assert torch.equal(back, J)
+
+
+@given(st.integers(0, 2 ** 31 - 1), st.integers(1, 4))
+@settings(max_examples=20, deadline=None)
+def test_realify_is_complex_matmul(seed, n):
+    """realify(A) acting on vecR-stacked reals equals the complex
+    product: the 2x2-complex <-> 4x4-real embedding is a homomorphism."""
+    rng = np.random.default_rng(seed)
+    A = torch.tensor(rng.standard_normal((n, 2, 2))
+                     + 1j * rng.standard_normal((n, 2, 2)))
+    x = torch.tensor(rng.standard_normal((n, 2))
+                     + 1j * rng.standard_normal((n, 2)))
+    Ar = R.realify(A)                       # [n, 4, 4]
+    xr = torch.stack([x.real[..., 0], x.imag[..., 0],
+                      x.real[..., 1], x.imag[..., 1]], dim=-1)
+    yr = (Ar @ xr.unsqueeze(-1)).squeeze(-1)
+    y = (A @ x.unsqueeze(-1)).squeeze(-1)
+    want = torch.stack([y.real[..., 0], y.imag[..., 0],
+                        y.real[..., 1], y.imag[..., 1]], dim=-1)
+    assert torch.allclose(yr, want, atol=1e-12)
+
+
+@given(st.integers(0, 2 ** 31 - 1), st.integers(1, 4))
+@settings(max_examples=20, deadline=None)
+def test_antirealify_is_antilinear_map(seed, n):
+    """antirealify(A) acting on stacked reals equals x -> A conj(x)."""
+    rng = np.random.default_rng(seed)
+    A = torch.tensor(rng.standard_normal((n, 2, 2))
+                     + 1j * rng.standard_normal((n, 2, 2)))
+    x = torch.tensor(rng.standard_normal((n, 2))
+                     + 1j * rng.standard_normal((n, 2)))
+    Ar = R.antirealify(A)
+    xr = torch.stack([x.real[..., 0], x.imag[..., 0],
+                      x.real[..., 1], x.imag[..., 1]], dim=-1)
+    yr = (Ar @ xr.unsqueeze(-1)).squeeze(-1)
+    y = (A @ x.conj().unsqueeze(-1)).squeeze(-1)
+    want = torch.stack([y.real[..., 0], y.imag[..., 0],
+                        y.real[..., 1], y.imag[..., 1]], dim=-1)
+    assert torch.allclose(yr, want, atol=1e-12)
+
+
+@given(st.integers(2, 8), st.integers(1, 4), st.sampled_from([0, 1, 2, 3]))
+@settings(max_examples=25, deadline=None)
+def test_poly_basis_shape_and_bernstein_partition(F, Npoly, ptype):
+    """Basis is [F, Npoly]; Bernstein rows (type 2) sum to one (partition
+    of unity), and every type's first column spans a constant direction
+    so a frequency-independent gain is representable."""
+    freqs = np.linspace(140e6, 160e6, F)
+    B = poly.setup_polynomials(freqs, 150e6, Npoly, ptype)
+    assert B.shape == (F, Npoly)
+    assert torch.isfinite(B).all()
+    if ptype == 2:
+        assert torch.allclose(B.sum(dim=1),
+                              torch.ones(F, dtype=B.dtype), atol=1e-9)
+    # constant function representable: residual of ls-fit of ones is ~0
+    ones = torch.ones(F, 1, dtype=torch.float64)
+    sol = torch.linalg.lstsq(B.to(torch.float64), ones).solution
+    assert float((B.to(torch.float64) @ sol - ones).abs().max()) < 1e-6
+
+
+@given(st.floats(0.3, 3.0), st.floats(0.3, 3.0))
+@settings(max_examples=20, deadline=None)
+def test_product_scale_symmetric_and_contracting(a, b):
+    g = shapelet.product_scale(a, b)
+    assert abs(g - shapelet.product_scale(b, a)) < 1e-12
+    assert g < min(a, b)  # products always NARROW the Gaussian
+    # composing with an infinite-width factor is the identity
+    assert abs(shapelet.product_scale(a, 1e9) - a) < 1e-6
+
+
+@given(st.integers(0, 2 ** 31 - 1), st.integers(1, 3), st.integers(1, 3))
+@settings(max_examples=10, deadline=None)
+def test_shapelet_product_linearity(seed, Lf, Lg):
+    """The Jones series product is bilinear: (aF) x G == a (F x G)."""
+    rng = np.random.default_rng(seed)
+    F = torch.tensor(rng.standard_normal((Lf * Lf, 2, 2))
+                     + 1j * rng.standard_normal((Lf * Lf, 2, 2)))
+    G = torch.tensor(rng.standard_normal((Lg * Lg, 2, 2))
+                     + 1j * rng.standard_normal((Lg * Lg, 2, 2)))
+    H1, g1 = shapelet.shapelet_product_jones(2.5 * F, G, 0.9, 1.2)
+    H2, g2 = shapelet.shapelet_product_jones(F, G, 0.9, 1.2)
+    assert g1 == g2
+    assert torch.allclose(H1, 2.5 * H2, atol=1e-10)
