@@ -44,7 +44,12 @@ def build_argparser():
                     help='0 calibrate; 1 simulate; 2 sim+add; 3 sim+sub')
     ap.add_argument('-z', dest='ignfile', help='ignore-cluster file')
     ap.add_argument('-b', dest='dochan', type=int, default=0)
-    ap.add_argument('-B', dest='dobeam', type=int, default=0)
+    ap.add_argument('-B', dest='dobeam', type=int, default=0,
+                    help='1: apply the station array beam in predict '
+                         '(MS must carry element_enu; predict_withbeam.c)')
+    ap.add_argument('-W', dest='whiten', type=int, default=0,
+                    help='1: pre-whiten data with the NCP uv taper '
+                         '(whiten_data, updatenu.c)')
     ap.add_argument('-E', dest='gpupredict', type=int, default=None,
                     help='1: use GPU (default: auto)')
     ap.add_argument('-x', dest='min_uvcut', type=float, default=0.0)
@@ -91,6 +96,29 @@ def load_context(args):
     return ms, pack, clusters, device, dtype
 
 
+def _predict_with_beam(ms, pack, tile, ti, args):
+    """Coherencies with the station array beam applied (-B 1;
+    predict_visibilities_multifreq_withbeam role). The MS must carry the
+    element layout: NpzMS key `element_enu` [N, E, 3] (m) plus optional
+    `lon`/`lat` (rad) — the analog of the LOFAR ANTENNA_FIELD tables the
+    reference reads (data.cpp:268-288)."""
+    from .. import beams
+    z = getattr(ms, '_z', {})
+    if 'element_enu' not in z:
+        raise SystemExit(
+            "-B 1 requires element layouts in the MS (key 'element_enu')")
+    lon = float(z['lon']) if 'lon' in z else 0.0
+    lat = float(z['lat']) if 'lat' in z else 0.92
+    cfg = beams.ArrayConfig(list(np.asarray(z['element_enu'])), lon, lat,
+                            ms.ra0, ms.dec0)
+    T = ms.tilesz
+    t0 = float(z['tmjd0']) if 'tmjd0' in z else 56789.0
+    tmjd = t0 + (ti * T + np.arange(T) + 0.5) * ms.tdelta / 86400.0
+    return beams.predict_coh_withbeam(
+        pack, tile.u, tile.v, tile.w, tile.freq0, tile.freq0, tile.fdelta,
+        tile.tdelta, tile.dec0, cfg, tmjd, ms.bb_tensor(), ms.Nbase, T)
+
+
 def uv_flags(tile, args):
     """Baseline uv-cut flags (predict.c flag=2 semantics for -x/-y)."""
     uvlen = torch.sqrt(tile.u ** 2 + tile.v ** 2) * tile.freq0
@@ -122,7 +150,14 @@ def run_calibration(args):
     for ti, tile in enumerate(ms.tiles()):
         t0 = time.time()
         flags = uv_flags(tile, args)
-        cohs = sage.precalc_coherencies(pack, tile)
+        if args.whiten:
+            from ..utils import taper
+            tile.x, _ = taper.whiten_data(tile.x, tile.u, tile.v,
+                                          tile.freq0)
+        if args.dobeam:
+            cohs = _predict_with_beam(ms, pack, tile, ti, args)
+        else:
+            cohs = sage.precalc_coherencies(pack, tile)
         if device != 'cpu':
             cohs = cohs.to(torch.complex64)
         bb = ms.bb_tensor(device=device)
