@@ -70,7 +70,8 @@ def build_argparser():
                     help='>0: stochastic calibration epochs')
     ap.add_argument('-M', dest='minibatches', type=int, default=1)
     ap.add_argument('-w', dest='minibands', type=int, default=1)
-    ap.add_argument('-A', dest='nadmm', type=int, default=1)
+    ap.add_argument('-A', dest='nadmm', type=int, default=1,
+                    help='consensus updates per epoch (stochastic -w>1)')
     ap.add_argument('-P', dest='npoly', type=int, default=2)
     ap.add_argument('-Q', dest='polytype', type=int, default=0)
     ap.add_argument('-r', dest='admm_rho', type=float, default=5.0)
@@ -85,8 +86,9 @@ def load_context(args):
         else bool(args.gpupredict)
     device = 'cuda:0' if use_gpu else 'cpu'
     dtype = torch.float32 if use_gpu else torch.float64
+    torch.set_num_threads(max(1, args.nthreads))
     ms = msdata.open_ms(args.ms, tilesz=args.tilesz, device=device,
-                      dtype=dtype)
+                        dtype=dtype, data_col=args.incol)
     ignore = skymod.read_ignore_file(args.ignfile) if args.ignfile else ()
     clusters = skymod.read_sky_cluster(args.sky, args.cluster, ms.ra0,
                                        ms.dec0, ms.freq0, fmt=args.format,
@@ -141,6 +143,7 @@ def run_calibration(args):
         solver_mode=args.solver_mode, robust_nulow=args.nulow,
         robust_nuhigh=args.nuhigh, lbfgs_iters=args.max_lbfgs if
         args.max_lbfgs > 0 else 0)
+    opts.lbfgs_m = args.lbfgs_m
     writer = None
     if args.solfile:
         writer = solutions.SolutionWriter(
@@ -220,6 +223,8 @@ def run_calibration(args):
         if not np.isfinite(res1) or (res0 > 0 and res1 > 5 * res0):
             state.J = pinit.clone()
         mean_nu = float(state.nu.mean())
+        if args.verbose:
+            print(f"tile {ti}: solve took {time.time() - t0:.2f} s")
         print(f"tile {ti}: residual {res0:.6f} -> {res1:.6f}, mean nu "
               f"{mean_nu:.1f} ({time.time() - t0:.1f}s)")
     ms.save()
@@ -277,9 +282,12 @@ def run_stochastic(args):
     bb = ms.bb_tensor(device=device)
     for ti, tile in enumerate(ms.tiles()):
         for ep in range(args.epochs):
-            cal.epoch(tile, bb, nmb=args.minibatches,
-                      lbfgs_iters=args.max_lbfgs,
-                      robust_nu=(args.nulow + args.nuhigh) / 2)
+            # -A extra consensus sweeps per epoch
+            # (minibatch_consensus_mode.cpp admm loop)
+            for _ in range(max(1, args.nadmm)):
+                cal.epoch(tile, bb, nmb=args.minibatches,
+                          lbfgs_iters=args.max_lbfgs,
+                          robust_nu=(args.nulow + args.nuhigh) / 2)
         xres = cal.residuals(tile, bb)
         ms.write_column(args.outcol, ti, xres)
         print(f"tile {ti}: stochastic epochs {args.epochs}, "

@@ -118,7 +118,8 @@ def main(argv=None):
     opts = sage.SageSolveOptions(
         max_emiter=args.max_emiter, max_iter=args.max_iter,
         solver_mode=args.solver_mode, robust_nulow=args.nulow,
-        robust_nuhigh=args.nuhigh)
+        robust_nuhigh=args.nuhigh,
+        lbfgs_iters=max(args.max_lbfgs, 0))
     writer = None
     zwriter = None
     if args.solfile:

@@ -217,8 +217,9 @@ class NpzMS:
     """
 
     def __init__(self, path, tilesz=None, device='cpu',
-                 dtype=torch.float64):
+                 dtype=torch.float64, data_col='data'):
         self.path = path
+        self.data_col = data_col
         z = np.load(path)
         self.N = int(z['N'])
         self.Nbase = self.N * (self.N - 1) // 2
@@ -282,7 +283,7 @@ class NpzMS:
                          device=self.device)
         w = torch.tensor(self._z['w'][r0:r1] / C_LIGHT, dtype=self.dtype,
                          device=self.device)
-        xo = torch.tensor(self._z['data'][r0:r1], dtype=cdtype,
+        xo = torch.tensor(self._z[self.data_col][r0:r1], dtype=cdtype,
                           device=self.device).permute(1, 0, 2, 3).contiguous()
         flags = torch.tensor(self._z['flags'][r0:r1], dtype=torch.bool,
                              device=self.device)
@@ -308,13 +309,17 @@ class NpzMS:
         np.savez_compressed(path or self.path, **out)
 
 
-def open_ms(path, tilesz=10, device='cpu', dtype=torch.float64):
+def open_ms(path, tilesz=10, device='cpu', dtype=torch.float64,
+            data_col='data'):
     """Open a measurement set by suffix: .npz -> NpzMS (the container used
     in casacore-free environments), anything else -> CasaMS
     (python-casacore table on disk, e.g. a LOFAR .MS directory)."""
     if str(path).endswith('.npz'):
-        return NpzMS(path, tilesz=tilesz, device=device, dtype=dtype)
-    return CasaMS(path, tilesz=tilesz, device=device, dtype=dtype)
+        return NpzMS(path, tilesz=tilesz, device=device, dtype=dtype,
+                     data_col=data_col)
+    return CasaMS(path, tilesz=tilesz, device=device, dtype=dtype,
+                  data_col='DATA' if data_col == 'data'
+                  else data_col.upper())
 
 
 def make_synthetic_npz(path, N=8, tilesz=4, Ntime=4, Nchan=2, pack=None,

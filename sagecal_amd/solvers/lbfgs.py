@@ -175,7 +175,8 @@ def polish(state, cohs, tile, bb, opts):
         return c, g
 
     v0 = _pack_params(state.J)
-    v1, _, _ = lbfgs_fit(fg, v0, maxiter=opts.lbfgs_iters, m=7)
+    v1, _, _ = lbfgs_fit(fg, v0, maxiter=opts.lbfgs_iters,
+                         m=getattr(opts, 'lbfgs_m', 7))
     state.J = _unpack_params(v1, Mt, N, state.J.dtype)
 
 
