@@ -225,7 +225,7 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts,
     # (ops.obj_token); a fresh token means a new tile -> drop old buffers
     new_tile = getattr(cohs, '_sagecal_token', None) is None
     tok = ops.obj_token(cohs)
-    if new_tile:
+    if new_tile or len(cache) > 64:
         cache.clear()
     key = (tuple(group), tok, B)
     ent = cache.get(key)
