@@ -55,6 +55,11 @@ class SageSolveOptions:
         self.em_group = em_group or (1 if mode == 'sequential' else 2)
         self.linsolv = linsolv
         self.nsubsets = nsubsets  # >0: ordered-subsets acceleration
+        # OS solver modes imply subsets (oslevmar_*, reference modes 1/2)
+        from ..constants import SM_OSLM_LBFGS, SM_OSLM_OSRLM_RLBFGS
+        if nsubsets == 0 and solver_mode in (SM_OSLM_LBFGS,
+                                             SM_OSLM_OSRLM_RLBFGS):
+            self.nsubsets = 2
         # joint cross-cluster LM refinement iterations after the EM sweeps
         # (quadratic local convergence; not in the reference, which only has
         # the LBFGS polish)
@@ -159,7 +164,9 @@ def total_model(state, cohs, bb, T, Nbase):
 def _inner_solve(prob, J, opts, maxiter):
     """Dispatch one deterministic inner solve by solver mode, honoring the
     reference's LMCUT heuristic (sagecalmain.h:24: N <= 40 -> RTR/NSD
-    replaced by LM)."""
+    replaced by LM) and ordered-subsets acceleration for the OS modes
+    (oslevmar_*; the graphed robust path solves full-batch instead —
+    a strictly more accurate iteration at the same cost there)."""
     mode = opts.solver_mode
     if prob.N > LMCUT and mode in (SM_RTR_OSLM_LBFGS,
                                    SM_RTR_OSRLM_RLBFGS):
@@ -168,6 +175,9 @@ def _inner_solve(prob, J, opts, maxiter):
     if prob.N > LMCUT and mode == SM_NSD_RLBFGS:
         from . import rtr as rtr_mod
         return rtr_mod.nsd_solve(prob, J, maxiter=2 * maxiter)[0]
+    if getattr(opts, 'nsubsets', 0) > 1:
+        return lm_mod.os_lm_solve(prob, J, maxiter=maxiter,
+                                  nsubsets=opts.nsubsets)[0]
     return lm_mod.lm_solve(prob, J, maxiter=maxiter)[0]
 
 
@@ -178,10 +188,11 @@ def robust_lm(prob, J0, nu0, opts):
     import os as _os
     nu = nu0 if nu0 > 0 else 2.0
     inner = max(3, opts.max_iter // max(1, opts.robust_outer))
-    if (opts.robust_outer == 1 and prob.layout is not None
-            and prob.x.is_cuda and opts.solver_mode != 5 or
-            (opts.robust_outer == 1 and prob.layout is not None
-             and prob.x.is_cuda and prob.N <= 40))             and _os.environ.get('SAGECAL_NO_GRAPH') != '1':
+    graph_ok = (opts.robust_outer == 1 and prob.layout is not None
+                and prob.x.is_cuda
+                and (opts.solver_mode != 5 or prob.N <= 40)
+                and _os.environ.get('SAGECAL_NO_GRAPH') != '1')
+    if graph_ok:
         # single-IRLS-pass schedule: whole solve as one graph replay; the
         # nu grid update runs on the host from the captured mean(log w - w)
         J, logsumw, _ = lm_mod.robust_lm_graphed(prob, J0, nu, inner)

@@ -120,3 +120,20 @@ def test_residuals_multifreq():
     assert xres.shape == tile.xo.shape
     # residual much smaller than data
     assert float(xres.abs().mean()) < 0.05 * float(tile.xo.abs().mean())
+
+
+def test_os_solver_modes_converge():
+    """Reference modes 1 (OS-LM) and 2 (OS robust LM) route through
+    ordered-subsets acceleration (oslevmar_*) and still converge."""
+    from sagecal_amd.constants import SM_OSLM_LBFGS, SM_OSLM_OSRLM_RLBFGS
+    for mode in (SM_OSLM_LBFGS, SM_OSLM_OSRLM_RLBFGS):
+        ms, pack = setup_ms(M=3, noise=1e-3, seed=4)
+        tile = ms.load_tile(0)
+        bb = ms.bb_tensor()
+        state = sage.CalState(pack, ms.N)
+        cohs = sage.precalc_coherencies(pack, tile)
+        opts = sage.SageSolveOptions(max_emiter=4, max_iter=12,
+                                     solver_mode=mode, robust_outer=2)
+        assert opts.nsubsets == 2
+        res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
+        assert res1 < 0.2 * res0, f"mode {mode}: {res0} -> {res1}"
