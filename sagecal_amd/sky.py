@@ -145,7 +145,8 @@ def read_sky_model(path, fmt=0, modes_dir=None):
                     continue
                 mf = os.path.join(d, name + '.fits.modes')
                 if os.path.exists(mf):
-                    src.sh_n0, src.sh_beta, src.sh_coeff =                         shmod.read_modes_file(mf)
+                    src.sh_n0, src.sh_beta, src.sh_coeff = \
+                        shmod.read_modes_file(mf)
                     break
             if src.eX == 0:
                 src.eX = 1.0
