@@ -261,7 +261,7 @@ def read_sky_cluster(sky_path, cluster_path, ra0, dec0, freq0, fmt=0,
     sources = read_sky_model(sky_path, fmt=fmt)
     if jd is not None:
         from . import coords
-        for s in sources:
+        for s in sources.values():
             s.ra, s.dec = coords.precess_radec(s.ra, s.dec, jd)
         ra0, dec0 = coords.precess_radec(ra0, dec0, jd)
     clist = read_cluster_file(cluster_path)
