@@ -4,6 +4,8 @@ Every test is marked gpu and compares the gfx950 kernel output against the
 same op computed by the plain PyTorch reference implementation (fp64 on
 CPU), with fp32-appropriate tolerances.
 """
+import os
+
 import numpy as np
 import pytest
 import torch
@@ -278,3 +280,41 @@ def test_chol_mw_matches_single_kernel():
     assert int(i1.sum()) == 0 and int(i2.sum()) == 0
     err = (x1 - x2).abs().max() / x1.abs().max()
     assert float(err) < 1e-4, f"mw vs single rel err {float(err)}"
+
+
+@pytest.mark.skipif(os.environ.get('SAGECAL_RTR_GRAPH') != '1',
+                    reason='RTR graph path is opt-in '
+                           '(SAGECAL_RTR_GRAPH=1); round-2 validation')
+def test_rtr_graphed_matches_eager():
+    """Graph-captured RTR vs the eager solver on the same problem
+    (enable with SAGECAL_RTR_GRAPH=1)."""
+    from sagecal_amd.solvers import rtr as rtr_mod, lm as lm_mod
+    from sagecal_amd.ops.hip_host import BaselineLayout
+    from sagecal_amd.ops import reference as R
+    dev = 'cuda:0'
+    rng = np.random.default_rng(2)
+    N, T = 64, 4
+    pairs = [(p, q) for p in range(N) for q in range(p + 1, N)]
+    Nbase = len(pairs)
+    bb = torch.tensor(pairs * T, device=dev)
+    B = Nbase * T
+    s = torch.tensor(rng.standard_normal(B) + 1j * rng.standard_normal(B))
+    coh = (s[:, None, None] * torch.eye(2, dtype=torch.complex128)) \
+        .to(device=dev, dtype=torch.complex64)
+    Jt = torch.tensor(np.eye(2)[None, None] + 0.2 * (
+        rng.standard_normal((1, N, 2, 2))
+        + 1j * rng.standard_normal((1, N, 2, 2)))).to(
+            device=dev, dtype=torch.complex64)
+    x = R.apply_jones(coh.cpu().to(torch.complex128),
+                      Jt.cpu().to(torch.complex128), bb.cpu()).to(
+                          device=dev, dtype=torch.complex64)
+    lay = BaselineLayout(bb, Nbase, T, 1, N, dev)
+    prob = lm_mod.LMProblem(x, coh, bb, N, 1, None, layout=lay)
+    J0 = Jt + 0.05 * torch.randn_like(Jt.real).to(Jt.dtype)
+    os.environ['SAGECAL_RTR_GRAPH'] = '0'
+    J_e, info_e = rtr_mod.rtr_solve(prob, J0.clone(), maxiter=10)
+    os.environ['SAGECAL_RTR_GRAPH'] = '1'
+    J_g, info_g = rtr_mod.rtr_solve_graphed(prob, J0.clone(), maxiter=10)
+    ce = float(info_e['final_cost'].max())
+    cg = float(info_g['final_cost'].max())
+    assert cg <= 1.2 * ce + 1e-6, (ce, cg)
