@@ -17,7 +17,6 @@ import sys
 import numpy as np
 
 from ..utils import fits as fitsio
-from .. import coords
 
 FWHM = 2.0 * np.sqrt(2.0 * np.log(2.0))
 

@@ -12,7 +12,6 @@ import sys
 import numpy as np
 
 from .. import msdata
-from ..constants import C_LIGHT
 
 
 def main(argv=None):

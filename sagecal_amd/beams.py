@@ -286,7 +286,6 @@ def _single_source_coh(pack, gi, u, v, w, freq, freq0, fdelta, tdelta,
     class _P:
         pass
     one = _P()
-    import copy
     for f_ in R.SourcePack.FIELDS:
         setattr(one, f_, getattr(pack, f_)[gi:gi + 1])
     one.stype = pack.stype[gi:gi + 1]

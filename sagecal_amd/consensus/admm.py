@@ -16,7 +16,6 @@ import torch.distributed as dist
 
 from . import poly
 from ..solvers import sage
-from ..ops import dispatch as ops
 
 
 class ConsensusADMM:

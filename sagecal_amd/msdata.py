@@ -14,12 +14,10 @@ Layouts (time-major, matching data.cpp:604 row order):
   x (channel-averaged DATA) complex [rows, 2, 2]; xo full channels
   [Nchan, rows, 2, 2].
 """
-import math
 import numpy as np
 import torch
 
 from .constants import C_LIGHT
-from . import coords
 
 
 def baseline_pairs(N):

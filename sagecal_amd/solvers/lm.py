@@ -12,7 +12,6 @@ Cholesky per LM iteration, sized so a 256-CU GPU is actually filled.
 Ordered-subsets (OS-LM, oslmfit.c) acceleration: each outer pass runs LM on
 a random contiguous fraction of the baselines then a final full pass.
 """
-import math
 import os
 import torch
 
