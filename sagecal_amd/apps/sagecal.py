@@ -295,21 +295,32 @@ def run_stochastic(args):
     ms.save()
 
 
-def main(argv=None):
-    args = build_argparser().parse_args(argv)
-    if not args.ms and args.mslist:
-        with open(args.mslist) as f:
-            names = [l.strip() for l in f if l.strip()]
-        args.ms = names[0]
-    if not args.ms:
-        print("need -d MS or -f MSlist", file=sys.stderr)
-        return 1
+def _run_one(args):
     if args.dosim > 0:
         run_simulation(args)
     elif args.epochs > 0:
         run_stochastic(args)
     else:
         run_calibration(args)
+
+
+def main(argv=None):
+    args = build_argparser().parse_args(argv)
+    names = [args.ms] if args.ms else []
+    if args.mslist:
+        with open(args.mslist) as f:
+            names += [l.strip() for l in f if l.strip()]
+    if not names:
+        print("need -d MS or -f MSlist", file=sys.stderr)
+        return 1
+    base_sol = args.solfile
+    # -f MSlist: every MS processed in sequence (main.cpp loops the
+    # mslist the same way; per-MS solution files keep the outputs apart)
+    for i, name in enumerate(names):
+        args.ms = name
+        if base_sol and len(names) > 1:
+            args.solfile = f"{base_sol}.ms{i}"
+        _run_one(args)
     return 0
 
 

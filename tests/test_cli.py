@@ -301,3 +301,34 @@ def test_cli_precesses_with_ms_epoch(tmp_path):
     assert rc == 0
     out = np.load(msf)
     assert np.abs(out['resp']).mean() < 0.4 * np.abs(out['data']).mean()
+
+
+def test_cli_mslist_processes_all(tmp_path):
+    """-f MSlist runs the calibration on EVERY listed MS (main.cpp MS
+    loop), writing residuals into each."""
+    from sagecal_amd.apps import sagecal as app
+    (tmp_path / 'sky.txt').write_text(SKY)
+    (tmp_path / 'cluster.txt').write_text(CLUSTER)
+    clusters = sky.read_sky_cluster(str(tmp_path / 'sky.txt'),
+                                    str(tmp_path / 'cluster.txt'),
+                                    0.0, np.pi / 4, 150e6)
+    pack = SourcePack(clusters)
+    names = []
+    for i in range(2):
+        msf = str(tmp_path / f'm{i}.npz')
+        msdata.make_synthetic_npz(msf, N=8, tilesz=4, Ntime=4, Nchan=2,
+                                  pack=pack, noise_sigma=1e-3, seed=10 + i,
+                                  ra0=0.0, dec0=np.pi / 4)
+        names.append(msf)
+    (tmp_path / 'list.txt').write_text('\n'.join(names))
+    rc = app.main(['-f', str(tmp_path / 'list.txt'),
+                   '-s', str(tmp_path / 'sky.txt'),
+                   '-c', str(tmp_path / 'cluster.txt'), '-t', '4',
+                   '-e', '2', '-g', '8', '-j', '3', '-l', '0',
+                   '-O', 'resm'])
+    assert rc == 0
+    for msf in names:
+        out = np.load(msf)
+        assert 'resm' in out.files, msf
+        assert np.abs(out['resm']).mean() < \
+            0.4 * np.abs(out['data']).mean(), msf
