@@ -12,6 +12,7 @@ clustering of the fitted sources into calibration directions (cluster.c)
 (the reference's create_clusters.py role included).
 """
 import argparse
+import os
 import sys
 
 import numpy as np
@@ -227,9 +228,69 @@ def _fmt_radec(ra, dec):
     return f"{hh} {mm} {ss:.4f} {sgn * dd} {dm} {dss:.4f}"
 
 
+def fit_spectra(imgs, freqs, hdr, islands, maxfits, criterion, psf_pix,
+                freq0=None):
+    """Multi-frequency island fitting (buildmultisky.c): positions and
+    shapes come from a joint fit of the MEAN image; per-plane fluxes are
+    then linear least squares with the shapes FIXED, and each component
+    gets a spectral index from a log-log fit. Returns sources with
+    'si' filled and flux referenced to freq0 (default: mean freq)."""
+    freqs = np.asarray(freqs, dtype=float)
+    f0 = float(freq0 or freqs.mean())
+    mean_img = np.mean(imgs, axis=0)
+    out = []
+    for ys, xs in islands:
+        comps = fit_island_multi(mean_img, ys, xs, maxfits, criterion)
+        if not comps:
+            continue
+        # design matrix of unit-amplitude components on this island
+        ysf, xsf = ys.astype(float), xs.astype(float)
+        cols = []
+        for c in comps:
+            th = [1.0, c['cy'], c['cx'], c['sy'], c['sx'], c['pa']]
+            cols.append(_gauss_eval(th, ysf, xsf, 1))
+        A = np.stack(cols, axis=1)                    # [npix, k]
+        AtA = A.T @ A + 1e-12 * np.eye(A.shape[1])
+        fluxes = []
+        for img in imgs:                               # per-plane amps
+            amp = np.linalg.solve(AtA, A.T @ img[ys, xs])
+            fluxes.append(amp * np.array(
+                [2.0 * np.pi * c['sy'] * c['sx'] for c in comps]))
+        fluxes = np.stack(fluxes)                      # [F, k]
+        lf = np.log(freqs / f0)
+        for k, c in enumerate(comps):
+            fk = fluxes[:, k]
+            pos = fk > 1e-12
+            if pos.sum() >= 2 and len(freqs) >= 2:
+                # weighted linear fit of log flux vs log freq
+                w = fk[pos]
+                X = np.stack([np.ones(pos.sum()), lf[pos]], axis=1)
+                sol, *_ = np.linalg.lstsq(X * w[:, None],
+                                          np.log(fk[pos]) * w, rcond=None)
+                s0, si = float(np.exp(sol[0])), float(sol[1])
+            else:
+                s0, si = float(fk.mean()), 0.0
+            ra, dec = fitsio.pix_to_radec(hdr, c['cx'], c['cy'])
+            d2r = np.pi / 180.0
+            pscale = abs(hdr['CDELT1']) * d2r
+            sig_maj, sig_min = max(c['sy'], c['sx']), min(c['sy'], c['sx'])
+            unresolved = sig_maj <= max(0.7, 1.05 * psf_pix / FWHM)
+            out.append(dict(
+                stype='P' if unresolved else 'G', ra=float(ra),
+                dec=float(dec), flux=s0, si=si,
+                eX=0.0 if unresolved else float(sig_maj * pscale * FWHM),
+                eY=0.0 if unresolved else float(sig_min * pscale * FWHM),
+                eP=0.0 if unresolved else float(c['pa'])))
+    return out, f0
+
+
 def main(argv=None):
     ap = argparse.ArgumentParser(prog='buildsky')
-    ap.add_argument('-f', dest='fits', required=True, help='FITS image')
+    ap.add_argument('-f', dest='fits', help='FITS image')
+    ap.add_argument('-d', dest='fitsdir',
+                    help='directory of per-frequency FITS planes '
+                         '(multi-frequency fit with spectral indices, '
+                         'buildmultisky.c)')
     ap.add_argument('-m', dest='mask', help='mask FITS (Duchamp style)')
     ap.add_argument('-t', dest='threshold', type=float,
                     help='island threshold (default: 5 sigma)')
@@ -259,10 +320,28 @@ def main(argv=None):
     ap.add_argument('-s', dest='outsky', help='output sky file')
     ap.add_argument('-c', dest='outcluster', help='output cluster file')
     args = ap.parse_args(argv)
+    if not args.fits and not args.fitsdir:
+        ap.error('need -f image.fits or -d fits_directory')
 
-    img, hdr = fitsio.read_fits_image(args.fits)
-    if args.negative:
-        img = -img
+    if args.fitsdir:
+        import glob as _glob
+        files = sorted(_glob.glob(os.path.join(args.fitsdir, '*.fits')))
+        if len(files) < 1:
+            print('buildsky: no FITS in directory', file=sys.stderr)
+            return 1
+        planes, freqs = [], []
+        hdr = None
+        for fn in files:
+            im, h = fitsio.read_fits_image(fn)
+            planes.append(-im if args.negative else im)
+            freqs.append(float(h.get('RESTFRQ', 150e6)))
+            hdr = hdr or h
+        img = np.mean(planes, axis=0)
+    else:
+        img, hdr = fitsio.read_fits_image(args.fits)
+        if args.negative:
+            img = -img
+        planes, freqs = None, None
     mask = None
     if args.mask:
         mask, _ = fitsio.read_fits_image(args.mask)
@@ -275,10 +354,20 @@ def main(argv=None):
         pscale = abs(hdr['CDELT1']) * d2r
         psf_pix = args.bmaj * as2rad / pscale
         beam_rad = 0.5 * (args.bmaj + args.bmin) * as2rad
+    if args.sidelobe_cut > 0:
+        islands = [(ys, xs) for ys, xs in islands
+                   if img[ys, xs].max() >= args.sidelobe_cut]
+    if planes is not None:
+        srcs, f0 = fit_spectra(planes, freqs, hdr, islands, args.maxfits,
+                               args.criterion, psf_pix)
+        if args.negative:
+            for s_ in srcs:
+                s_['flux'] = -s_['flux']
+        freq = f0
+        _finish(args, srcs, freq)
+        return 0
     srcs = []
     for ys, xs in islands:
-        if args.sidelobe_cut > 0 and img[ys, xs].max() < args.sidelobe_cut:
-            continue
         comps = fit_island(img, hdr, ys, xs, args.maxfits, args.criterion,
                            psf_pix)
         if args.rescale:
@@ -298,7 +387,12 @@ def main(argv=None):
         for s in srcs:
             s['flux'] = -s['flux']
     freq = hdr.get('RESTFRQ', 150e6)
-    outsky = args.outsky or args.fits + '.sky.txt'
+    _finish(args, srcs, freq)
+    return 0
+
+
+def _finish(args, srcs, freq):
+    outsky = args.outsky or (args.fits or args.fitsdir) + '.sky.txt'
     outcl = args.outcluster or outsky + '.cluster'
     names = []
     with open(outsky, 'w') as f:
@@ -306,8 +400,9 @@ def main(argv=None):
         for i, s in enumerate(srcs):
             name = f"{s['stype']}{i}C{i}"
             names.append(name)
+            si = s.get('si', 0.0)
             f.write(f"{name} {_fmt_radec(s['ra'], s['dec'])} "
-                    f"{s['flux']:.6f} 0 0 0 0 0 "
+                    f"{s['flux']:.6f} 0 0 0 {si:.6f} 0 "
                     f"{s['eX']:.8e} {s['eY']:.8e} {s['eP']:.8e} {freq}\n")
     ras = np.array([s['ra'] for s in srcs])
     decs = np.array([s['dec'] for s in srcs])
@@ -320,7 +415,6 @@ def main(argv=None):
             f.write(f"{q + 1} 1 " + ' '.join(members) + "\n")
     print(f"buildsky: {len(srcs)} sources -> {outsky}, "
           f"{len(set(assign))} clusters -> {outcl}")
-    return 0
 
 
 if __name__ == '__main__':
