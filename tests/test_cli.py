@@ -332,3 +332,37 @@ def test_cli_mslist_processes_all(tmp_path):
         assert 'resm' in out.files, msf
         assert np.abs(out['resm']).mean() < \
             0.4 * np.abs(out['data']).mean(), msf
+
+
+def test_annotate_and_convert(tmp_path):
+    """annotate produces a DS9 region per source; convert_skymodel LSM ->
+    BBS -> LSM round-trips positions/fluxes/spectral index."""
+    from sagecal_amd.apps import annotate, convert_skymodel
+    from sagecal_amd import sky as skymod
+    (tmp_path / 'sky.txt').write_text(SKY)
+    (tmp_path / 'cluster.txt').write_text(CLUSTER)
+    reg = str(tmp_path / 'out.reg')
+    rc = annotate.main(['-s', str(tmp_path / 'sky.txt'),
+                        '-c', str(tmp_path / 'cluster.txt'),
+                        '-o', reg, '-n'])
+    assert rc == 0
+    lines = open(reg).read()
+    srcs = skymod.read_sky_model(str(tmp_path / 'sky.txt'))
+    assert lines.count('circle') + lines.count('ellipse') == len(srcs)
+    assert 'fk5' in lines
+    # LSM -> BBS -> LSM
+    bbs = str(tmp_path / 'model.bbs')
+    lsm2 = str(tmp_path / 'sky2.txt')
+    assert convert_skymodel.main(['-l', '-i', str(tmp_path / 'sky.txt'),
+                                  '-o', bbs]) == 0
+    assert convert_skymodel.main(['-b', '-i', bbs, '-o', lsm2]) == 0
+    s2 = skymod.read_sky_model(lsm2)
+    assert len(s2) == len(srcs)
+    a = sorted(srcs.values(), key=lambda s: s.sI)
+    b = sorted(s2.values(), key=lambda s: s.sI)
+    for x, y in zip(a, b):
+        dra = (x.ra - y.ra + np.pi) % (2 * np.pi) - np.pi
+        assert abs(dra) < 1e-6
+        assert abs(x.dec - y.dec) < 1e-6
+        assert abs(x.sI - y.sI) < 1e-5
+        assert abs(x.spec_idx - y.spec_idx) < 1e-3
