@@ -176,6 +176,19 @@ def main(argv=None):
         writer.close()
     if zwriter:
         zwriter.close()
+    if rank == 0 and args.solfile and args.spatial:
+        Zsp = adm.spatial_coefficients()
+        if Zsp is not None:
+            # spatial_<solfile>: rows p of the [P, 2G] real coefficient
+            # matrix (sagecal_master.cpp spatial write)
+            import os as _os
+            d, b = _os.path.split(args.solfile)
+            with open(_os.path.join(d, f"spatial_{b}"), 'w') as fh:
+                fh.write("# spatial model coefficients: P x 2G\n")
+                for p in range(Zsp.shape[0]):
+                    fh.write(str(p) + ''.join(
+                        f" {v.real:e} {v.imag:e}" for v in
+                        Zsp[p].tolist()) + "\n")
     if world > 1:
         dist.destroy_process_group()
     return 0

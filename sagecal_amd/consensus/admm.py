@@ -228,6 +228,21 @@ class ConsensusADMM:
             self.alpha = self.spatial_alpha
             self._update_bii()
 
+    def spatial_coefficients(self):
+        """FISTA spatial-model coefficient matrix Zsp [P, G] (P = the
+        flattened Npoly*N*4 complex parameters, G = basis modes) — the
+        payload of the master's spatial_<solfile> write
+        (sagecal_master.cpp:1176-1186)."""
+        if self.Zspat is None or self._Phi is None:
+            return None
+        from . import fista as fista_mod
+        M = self.state.M
+        lam, mu_l1, order, fiters, cadence = self.spatial
+        Zb = torch.view_as_real(self.Z).reshape(M, -1)
+        Zb = torch.complex(Zb[:, 0::2], Zb[:, 1::2])
+        return fista_mod.update_spatialreg_fista(
+            Zb.cpu(), self._Phi, lam=lam, mu=mu_l1, maxiter=fiters)
+
     def diffuse_station_series(self):
         """Per-station Jones-valued spatial model as a shapelet series
         [N, G, 2, 2] with its scale (for this band): the FISTA model Zsp

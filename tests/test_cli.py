@@ -113,6 +113,7 @@ def _mpi_worker(rank, world, tmpdir):
                        '-c', os.path.join(tmpdir, 'cluster.txt'),
                        '-t', '4', '-A', '4', '-P', '2', '-j', '3',
                        '-e', '2', '-g', '8', '-r', '2.0', '-M',
+                       '-X', '0.01,1e-4,1,10,2', '-u', '0.0',
                        '-p', os.path.join(tmpdir, 'sol.txt')])
         assert rc == 0
     finally:
@@ -152,6 +153,8 @@ def test_sagecal_mpi_two_bands(tmp_path):
     assert tiles and tiles[0].shape[-2:] == (2, 2)
     hdrz, ztiles = solutions.read_solutions(str(tmp_path / 'sol.txt.Z'))
     assert ztiles, "global Z solution file missing/empty"
+    spf = tmp_path / 'spatial_sol.txt'
+    assert spf.exists() and len(spf.read_text().splitlines()) > 2
     # Npoly=2: the Z file carries twice the J file's values per tile
     # (read_solutions chunks rows in 8N groups, so Npoly groups come back
     # as Npoly entries)
