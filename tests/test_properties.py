@@ -116,3 +116,45 @@ def test_shapelet_product_linearity(seed, Lf, Lg):
     H2, g2 = shapelet.shapelet_product_jones(F, G, 0.9, 1.2)
     assert g1 == g2
     assert torch.allclose(H1, 2.5 * H2, atol=1e-10)
+
+
+@given(st.integers(0, 2 ** 31 - 1))
+@settings(max_examples=10, deadline=None)
+def test_lm_monotone_cost_reduction(seed):
+    """LM on a random solvable (noiseless) calibration problem never
+    increases the cost and reaches a meaningful reduction."""
+    from sagecal_amd.solvers import lm as lm_mod
+    rng = np.random.default_rng(seed)
+    N, T = 6, 2
+    pairs = [(p, q) for p in range(N) for q in range(p + 1, N)]
+    bb = torch.tensor(pairs * T)
+    B = len(pairs) * T
+    coh = torch.tensor(rng.standard_normal((B, 2, 2))
+                       + 1j * rng.standard_normal((B, 2, 2)))
+    Jt = torch.tensor(np.eye(2)[None, None] + 0.3 * (
+        rng.standard_normal((1, N, 2, 2))
+        + 1j * rng.standard_normal((1, N, 2, 2))))
+    x = R.apply_jones(coh, Jt, bb)
+    prob = lm_mod.LMProblem(x, coh, bb, N, 1, None)
+    J0 = torch.eye(2, dtype=torch.complex128)[None, None].expand(
+        1, N, 2, 2).clone()
+    J, info = lm_mod.lm_solve(prob, J0, maxiter=25)
+    assert float(info['final_cost'][0]) <= float(info['init_cost'][0])
+    assert float(info['final_cost'][0]) < 0.2 * float(
+        info['init_cost'][0]) + 1e-12
+
+
+@given(st.integers(0, 2 ** 31 - 1), st.floats(2.5, 25.0))
+@settings(max_examples=15, deadline=None)
+def test_student_t_weights_bounded(seed, nu):
+    """IRLS Student's-t weights live in (0, (nu+8)/nu] and weight
+    outliers strictly below clean points."""
+    from sagecal_amd.ops import reference as RR
+    rng = np.random.default_rng(seed)
+    r = torch.tensor(rng.standard_normal((50, 2, 2))
+                     + 1j * rng.standard_normal((50, 2, 2))) * 0.1
+    r[0] *= 100.0                        # one outlier
+    w = RR.update_weights(r, nu, p=8)
+    assert float(w.min()) > 0
+    assert float(w.max()) <= (nu + 8.0) / nu + 1e-9
+    assert float(w[0]) < float(w[1:].min())
