@@ -369,3 +369,22 @@ def test_annotate_and_convert(tmp_path):
         assert abs(x.dec - y.dec) < 1e-6
         assert abs(x.sI - y.sI) < 1e-5
         assert abs(x.spec_idx - y.spec_idx) < 1e-3
+
+
+def test_create_clusters(tmp_path):
+    from sagecal_amd.apps import create_clusters
+    from sagecal_amd import sky as skymod
+    (tmp_path / 'sky.txt').write_text(SKY)
+    out = str(tmp_path / 'cl2.txt')
+    rc = create_clusters.main(['-s', str(tmp_path / 'sky.txt'),
+                               '-c', out, '-Q', '2', '-t', '2'])
+    assert rc == 0
+    clist = skymod.read_cluster_file(out)
+    assert len(clist) == 2
+    allnames = [n for _, _, ns in clist for n in ns]
+    assert sorted(allnames) == ['P1C1', 'P2C1', 'P3C2']
+    assert all(nc == 2 for _, nc, _ in clist)
+    # the written file drives a calibration end-to-end
+    clusters = skymod.read_sky_cluster(str(tmp_path / 'sky.txt'), out,
+                                       0.0, np.pi / 4, 150e6)
+    assert sum(c.nsrc for c in clusters) == 3
