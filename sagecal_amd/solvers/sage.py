@@ -152,10 +152,16 @@ def _model_cluster(state, ci, coh_ci, bb, T, Nbase, B, lay=None):
     return ops.apply_jones(coh_ci, Jc, bb, rows_g, lay), rows
 
 
-def total_model(state, cohs, bb, T, Nbase):
+def total_model(state, cohs, bb, T, Nbase, skip=()):
+    """Sum of J_p C J_q^H over clusters; `skip` indices are left out
+    (negative-cluster-id semantics: residual.c:74 subtracts only
+    clusters with id >= 0 — target-field clusters keep their flux in
+    the residual)."""
     B = cohs.shape[1]
     V = torch.zeros_like(cohs[0])
     for ci in range(state.M):
+        if ci in skip:
+            continue
         Vc, _ = _model_cluster(state, ci, cohs[ci], bb, T, Nbase, B)
         V = V + Vc
     return V
@@ -395,10 +401,14 @@ def calculate_residuals_multifreq(state, pack, tile, bb, ccid=None, rho=0.0,
     B = tile.x.shape[0]
     fdelta_ch = tile.fdelta / len(tile.freqs)
     out = torch.empty_like(tile.xo)
+    # negative cluster ids: solved during calibration but NOT subtracted
+    # from the residual (residual.c:74-75 target-field convention)
+    ids = getattr(pack, 'cluster_ids', list(range(state.M)))
+    skip = {i for i, c in enumerate(ids) if c < 0}
     for fi, f in enumerate(tile.freqs):
         cohs = ops.predict_coh(pack, tile.u, tile.v, tile.w, float(f),
                                tile.freq0, fdelta_ch, tile.tdelta, tile.dec0)
-        V = total_model(state, cohs, bb, T, Nbase)
+        V = total_model(state, cohs, bb, T, Nbase, skip=skip)
         out[fi] = tile.xo[fi] - V
     if ccid is not None:
         ids = getattr(pack, 'cluster_ids', list(range(state.M)))

@@ -137,3 +137,36 @@ def test_os_solver_modes_converge():
         assert opts.nsubsets == 2
         res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
         assert res1 < 0.2 * res0, f"mode {mode}: {res0} -> {res1}"
+
+
+def test_negative_cluster_id_not_subtracted():
+    """Clusters with a negative id are solved but NOT subtracted from the
+    residual (residual.c:74, the 3c196 target-field convention): the
+    residual retains that cluster's flux while positive clusters are
+    removed."""
+    srcs, clist = sky.make_synthetic_sky(M=2, nsrc_per_cluster=3, seed=2)
+    # flip the first cluster's id negative
+    clist = [(-1 if i == 0 else cid, nc, names)
+             for i, (cid, nc, names) in enumerate(clist)]
+    clusters = sky.build_clusters(srcs, clist, 0.0, np.pi / 4, 150e6)
+    pack = SourcePack(clusters)
+    ms = msdata.SyntheticMS(N=8, tilesz=4, Ntime=4, Nchan=1, pack=pack,
+                            noise_sigma=1e-4, seed=2, bandwidth=60e3)
+    tile = ms.load_tile(0)
+    bb = ms.bb_tensor()
+    state = sage.CalState(pack, ms.N)
+    cohs = sage.precalc_coherencies(pack, tile)
+    opts = sage.SageSolveOptions(max_emiter=6, max_iter=12,
+                                 solver_mode=SM_LM_LBFGS)
+    sage.sagefit(state, cohs, tile, bb, opts)
+    xres = sage.calculate_residuals_multifreq(state, pack, tile, bb)
+    # residual must still hold ~the negative cluster's model power
+    Vneg = sage.total_model(state, cohs, bb, tile.tilesz, tile.Nbase,
+                            skip={1})
+    keep = float(Vneg.abs().pow(2).mean())
+    res = float(xres[0].abs().pow(2).mean())
+    assert res > 0.5 * keep, (res, keep)
+    # and with no skip everything subtracts to near the noise floor
+    Vall = sage.total_model(state, cohs, bb, tile.tilesz, tile.Nbase)
+    full_res = float((tile.xo[0] - Vall).abs().pow(2).mean())
+    assert full_res < 0.05 * res
