@@ -24,7 +24,9 @@ import torch.distributed as dist
 def build_argparser():
     ap = argparse.ArgumentParser(prog='sagecal-mpi')
     ap.add_argument('-f', dest='mslist', required=True,
-                    help='file listing per-band MS paths (one per rank)')
+                    help='file listing per-band MS paths (one per rank), '
+                         'or a glob pattern like "*.ms" (the reference '
+                         'master discovers MSs by pattern)')
     ap.add_argument('-s', dest='sky', required=True)
     ap.add_argument('-c', dest='cluster', required=True)
     ap.add_argument('-p', dest='solfile', help='per-rank solutions file')
@@ -49,6 +51,18 @@ def build_argparser():
     ap.add_argument('-u', dest='spatial_alpha', type=float, default=0.0)
     ap.add_argument('-L', dest='nulow', type=float, default=2.0)
     ap.add_argument('-H', dest='nuhigh', type=float, default=30.0)
+    ap.add_argument('-n', dest='nthreads', type=int, default=6)
+    ap.add_argument('-m', dest='lbfgs_m', type=int, default=7)
+    ap.add_argument('-x', dest='min_uvcut', type=float, default=0.0)
+    ap.add_argument('-y', dest='max_uvcut', type=float, default=1e9)
+    ap.add_argument('-k', dest='ccid', type=int, default=-99999,
+                    help='correct residuals with this cluster id')
+    ap.add_argument('-o', dest='rho_corr', type=float, default=1e-9)
+    ap.add_argument('-K', dest='nskip', type=int, default=0,
+                    help='skip this many solution intervals first')
+    ap.add_argument('-T', dest='nend', type=int, default=0,
+                    help='>0: stop after this solution interval')
+    ap.add_argument('-W', dest='whiten', type=int, default=0)
     ap.add_argument('-M', dest='mdl', action='store_true',
                     help='evaluate AIC/MDL over polynomial orders 1..-P '
                          'after the first tile and print the suggestion '
@@ -77,8 +91,16 @@ def main(argv=None):
     from ..solvers import sage
     from ..consensus.admm import ConsensusADMM
 
-    with open(args.mslist) as f:
-        names = [l.strip() for l in f if l.strip()]
+    import glob as _glob
+    if os.path.isfile(args.mslist):
+        with open(args.mslist) as f:
+            names = [l.strip() for l in f if l.strip()]
+    else:
+        names = sorted(_glob.glob(args.mslist))
+        if not names:
+            print(f"no MS matches {args.mslist}", file=sys.stderr)
+            return 1
+    torch.set_num_threads(max(1, args.nthreads))
     my_ms = names[rank % len(names)] if world <= len(names) else \
         names[rank]
     ms = msdata.open_ms(my_ms, tilesz=args.tilesz, device=device,
@@ -120,6 +142,7 @@ def main(argv=None):
         solver_mode=args.solver_mode, robust_nulow=args.nulow,
         robust_nuhigh=args.nuhigh,
         lbfgs_iters=max(args.max_lbfgs, 0))
+    opts.lbfgs_m = args.lbfgs_m
     writer = None
     zwriter = None
     if args.solfile:
@@ -134,13 +157,29 @@ def main(argv=None):
                 ms.tilesz * ms.tdelta / 60.0, ms.N, state.M, adm.Npoly)
     bb = ms.bb_tensor(device=device)
     for ti, tile in enumerate(ms.tiles()):
+        if ti < args.nskip:
+            continue
+        if args.nend > 0 and ti >= args.nend:
+            break
+        if args.whiten:
+            from ..utils import taper
+            tile.x, _ = taper.whiten_data(tile.x, tile.u, tile.v,
+                                          tile.freq0)
+        # uv cuts flag baselines out of the solve (predict.c flag=2)
+        uvlen = torch.sqrt(tile.u ** 2 + tile.v ** 2) * tile.freq0
+        flags = tile.flags | (uvlen < args.min_uvcut) | \
+            (uvlen > args.max_uvcut)
         cohs = sage.precalc_coherencies(pack, tile)
         if cohs.dtype != cdtype:
             cohs = cohs.to(cdtype)
-        res0, res1 = adm.run(cohs, tile, bb, opts, n_admm=args.nadmm)
+        res0, res1 = adm.run(cohs, tile, bb, opts, n_admm=args.nadmm,
+                             flags=flags)
         if args.use_global:
             state.J = adm.global_solution()
-        xres = sage.calculate_residuals_multifreq(state, pack, tile, bb)
+        ccid = args.ccid if args.ccid != -99999 else None
+        xres = sage.calculate_residuals_multifreq(state, pack, tile, bb,
+                                                  ccid=ccid,
+                                                  rho=args.rho_corr)
         ms.write_column(args.outcol, ti, xres)
         if writer:
             writer.write_tile(state)

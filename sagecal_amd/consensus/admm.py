@@ -179,7 +179,7 @@ class ConsensusADMM:
         self.Yhat_prev = Yhat.clone()
         self.J_prev = self.state.J.clone()
 
-    def run(self, cohs, tile, bb, opts, n_admm=10):
+    def run(self, cohs, tile, bb, opts, n_admm=10, flags=None):
         """The per-tile ADMM loop (sagecal_master.cpp:731-1060 semantics).
         Returns (res0, res1) of the final local solve."""
         res0 = res1 = None
@@ -189,7 +189,7 @@ class ConsensusADMM:
             admm_terms = None
             if it > 0:
                 admm_terms = (self.rho.to(self.dev), self.Y, BZ_old)
-            r0, r1 = sage.sagefit(st, cohs, tile, bb, opts,
+            r0, r1 = sage.sagefit(st, cohs, tile, bb, opts, flags=flags,
                                   admm_terms=admm_terms)
             if res0 is None:
                 res0 = r0
