@@ -45,8 +45,9 @@ def build_argparser():
     ap.add_argument('-z', dest='ignfile', help='ignore-cluster file')
     ap.add_argument('-b', dest='dochan', type=int, default=0)
     ap.add_argument('-B', dest='dobeam', type=int, default=0,
-                    help='1: apply the station array beam in predict '
-                         '(MS must carry element_enu; predict_withbeam.c)')
+                    help='beam in predict: 1 array, 2 array+element, '
+                         '3 element (MS must carry element_enu; '
+                         'predict_withbeam.c / DOBEAM_* modes)')
     ap.add_argument('-W', dest='whiten', type=int, default=0,
                     help='1: pre-whiten data with the NCP uv taper '
                          '(whiten_data, updatenu.c)')
@@ -118,7 +119,8 @@ def _predict_with_beam(ms, pack, tile, ti, args):
     tmjd = t0 + (ti * T + np.arange(T) + 0.5) * ms.tdelta / 86400.0
     return beams.predict_coh_withbeam(
         pack, tile.u, tile.v, tile.w, tile.freq0, tile.freq0, tile.fdelta,
-        tile.tdelta, tile.dec0, cfg, tmjd, ms.bb_tensor(), ms.Nbase, T)
+        tile.tdelta, tile.dec0, cfg, tmjd, ms.bb_tensor(), ms.Nbase, T,
+        mode=min(args.dobeam, 3))
 
 
 def uv_flags(tile, args):

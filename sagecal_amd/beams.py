@@ -250,20 +250,41 @@ def predict_coh_withbeam(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0,
         else torch.complex64
     out = torch.zeros(M, B, 2, 2, dtype=cdtype, device=u.device)
     ra, dec = _pack_radec(pack, dec0)
-    af = array_beam(cfg, ra, dec, [freq], tmjd)  # [S, K, T, 1]
-    af = af[..., 0]                              # [S, K, T]
     t_idx = torch.arange(B) // Nbase
     p = bb[:, 0]
     q = bb[:, 1]
+    use_array = mode in (1, 2)
+    use_elem = mode in (2, 3)
+    if use_array:
+        af = array_beam(cfg, ra, dec, [freq], tmjd)  # [S, K, T, 1]
+        af = af[..., 0]                              # [S, K, T]
+    if use_elem:
+        # element E-Jones per (source, time): one dipole pattern per
+        # array origin (kernel_element_beam; DOBEAM_ELEMENT/FULL modes)
+        if coeffs is None:
+            coeffs = make_synthetic_element_coeffs(freqs=(freq,))
+        gmst = coords.jd_to_gmst(np.asarray(np.atleast_1d(tmjd))
+                                 + 2400000.5)
+        Es = []
+        for g in np.atleast_1d(gmst):
+            _, az, el = _direction_enu(ra, dec, cfg.lon, cfg.lat, g)
+            E = element_beam(coeffs, az, np.maximum(el, 0.0), freq)
+            E[torch.from_numpy(el < 0)] = 0
+            Es.append(E)
+        Ej = torch.stack(Es, dim=1).to(cdtype)       # [K, T, 2, 2]
     for ci in range(M):
         s0, s1 = int(pack.cluster_off[ci]), int(pack.cluster_off[ci + 1])
         for gi in range(s0, s1):
             sub = _single_source_coh(pack, gi, u, v, w, freq, freq0,
                                      fdelta, tdelta, dec0)   # [B,2,2]
-            gp = af[p, gi, t_idx]          # [B] complex
-            gq = af[q, gi, t_idx]
-            wgt = (gp * gq.conj()).to(sub.dtype)
-            out[ci] += wgt[:, None, None] * sub
+            if use_elem:
+                Eb = Ej[gi, t_idx]                   # [B, 2, 2]
+                sub = Eb @ sub @ Eb.conj().transpose(-1, -2)
+            if use_array:
+                gp = af[p, gi, t_idx]                # [B] complex
+                gq = af[q, gi, t_idx]
+                sub = (gp * gq.conj()).to(sub.dtype)[:, None, None] * sub
+            out[ci] += sub
     return out
 
 
