@@ -46,8 +46,9 @@ def build_argparser():
     ap.add_argument('-b', dest='dochan', type=int, default=0)
     ap.add_argument('-B', dest='dobeam', type=int, default=0,
                     help='beam in predict: 1 array, 2 array+element, '
-                         '3 element (MS must carry element_enu; '
-                         'predict_withbeam.c / DOBEAM_* modes)')
+                         '3 element; 4/5/6 same but per channel '
+                         '(MS must carry element_enu; predict_withbeam.c '
+                         '/ DOBEAM_* + _WB modes)')
     ap.add_argument('-W', dest='whiten', type=int, default=0,
                     help='1: pre-whiten data with the NCP uv taper '
                          '(whiten_data, updatenu.c)')
@@ -117,10 +118,24 @@ def _predict_with_beam(ms, pack, tile, ti, args):
     T = ms.tilesz
     t0 = float(z['tmjd0']) if 'tmjd0' in z else 56789.0
     tmjd = t0 + (ti * T + np.arange(T) + 0.5) * ms.tdelta / 86400.0
-    return beams.predict_coh_withbeam(
-        pack, tile.u, tile.v, tile.w, tile.freq0, tile.freq0, tile.fdelta,
-        tile.tdelta, tile.dec0, cfg, tmjd, ms.bb_tensor(), ms.Nbase, T,
-        mode=min(args.dobeam, 3))
+    mode = args.dobeam
+    if mode <= 3:
+        return beams.predict_coh_withbeam(
+            pack, tile.u, tile.v, tile.w, tile.freq0, tile.freq0,
+            tile.fdelta, tile.tdelta, tile.dec0, cfg, tmjd,
+            ms.bb_tensor(), ms.Nbase, T, mode=mode)
+    # -B 4/5/6: per-channel beam (DOBEAM_*_WB): evaluate the beam at each
+    # channel frequency and average, like the discrete channel model
+    mode -= 3
+    fdelta_ch = tile.fdelta / len(tile.freqs)
+    acc = None
+    for f in tile.freqs:
+        c = beams.predict_coh_withbeam(
+            pack, tile.u, tile.v, tile.w, float(f), tile.freq0,
+            fdelta_ch, tile.tdelta, tile.dec0, cfg, tmjd,
+            ms.bb_tensor(), ms.Nbase, T, mode=mode)
+        acc = c if acc is None else acc + c
+    return acc / len(tile.freqs)
 
 
 def uv_flags(tile, args):

@@ -269,6 +269,14 @@ def test_cli_beam_and_whiten(tmp_path):
     out = np.load(msf)
     assert 'resb' in out.files
     assert np.isfinite(out['resb']).all()
+    # per-channel full beam (-B 5 = array+element WB)
+    rc = app.main(['-d', msf, '-s', str(tmp_path / 'sky.txt'),
+                   '-c', str(tmp_path / 'cluster.txt'), '-t', '4',
+                   '-e', '1', '-g', '6', '-j', '3', '-l', '0',
+                   '-B', '5', '-O', 'resb5'])
+    assert rc == 0
+    out = np.load(msf)
+    assert np.isfinite(out['resb5']).all()
 
 
 def test_cli_precesses_with_ms_epoch(tmp_path):
