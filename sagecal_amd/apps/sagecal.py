@@ -77,6 +77,9 @@ def build_argparser():
     ap.add_argument('-P', dest='npoly', type=int, default=2)
     ap.add_argument('-Q', dest='polytype', type=int, default=0)
     ap.add_argument('-r', dest='admm_rho', type=float, default=5.0)
+    ap.add_argument('-R', dest='randomize', type=int, default=0,
+                    help='1: alternate EM sweeps reallocate LM iterations '
+                         'to high-error groups (lmfit.c weighted_iter)')
     ap.add_argument('-V', dest='verbose', action='store_true')
     return ap
 
@@ -159,7 +162,7 @@ def run_calibration(args):
         max_emiter=args.max_emiter, max_iter=args.max_iter,
         solver_mode=args.solver_mode, robust_nulow=args.nulow,
         robust_nuhigh=args.nuhigh, lbfgs_iters=args.max_lbfgs if
-        args.max_lbfgs > 0 else 0)
+        args.max_lbfgs > 0 else 0, randomize=bool(args.randomize))
     opts.lbfgs_m = args.lbfgs_m
     writer = None
     if args.solfile:

@@ -38,7 +38,7 @@ class SageSolveOptions:
                  robust_nulow=NU_LOW, robust_nuhigh=NU_HIGH,
                  robust_outer=3, lbfgs_iters=0, lbfgs_minibatch=0,
                  mode='batched', em_group=None, linsolv=0, nsubsets=0,
-                 joint_iters=0):
+                 joint_iters=0, randomize=False):
         self.max_emiter = max_emiter
         self.max_iter = max_iter
         self.solver_mode = solver_mode
@@ -55,6 +55,11 @@ class SageSolveOptions:
         self.em_group = em_group or (1 if mode == 'sequential' else 2)
         self.linsolv = linsolv
         self.nsubsets = nsubsets  # >0: ordered-subsets acceleration
+        # -R: alternate EM sweeps reallocate LM iterations toward the
+        # groups whose cost dropped most (lmfit.c:871-879 weighted_iter;
+        # off by default — the batched design converges without it, and
+        # constant iteration counts keep hipGraph reuse perfect)
+        self.randomize = randomize
         # OS solver modes imply subsets (oslevmar_*, reference modes 1/2)
         from ..constants import SM_OSLM_LBFGS, SM_OSLM_OSRLM_RLBFGS
         if nsubsets == 0 and solver_mode in (SM_OSLM_LBFGS,
@@ -227,7 +232,7 @@ def robust_lm(prob, J0, nu0, opts):
 
 
 def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts,
-                 admm_terms=None):
+                 admm_terms=None, itermax=None):
     """Solve a group of clusters as ONE batched LM problem (block-diagonal
     across clusters via the chunk axis), then update the running residual
     incrementally: res += sum(V_old - V_new) over the group."""
@@ -288,17 +293,22 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts,
         admm = (rho_c, Y[sel], BZ[sel])
     prob = lm_mod.LMProblem(xcat, ccat, bbcat, state.N, nch_tot, rcat,
                             layout=lay, admm=admm)
+    mi = itermax if itermax is not None else opts.max_iter
     if opts.robust:
         nus = float(torch.stack([state.nu[ci] for ci in group]).mean())
+        if itermax is not None:
+            import copy as _copy
+            opts = _copy.copy(opts)
+            opts.max_iter = mi
         Jn, nu_new = robust_lm(prob, J0, nus, opts)
         for ci in group:
             state.nu[ci] = nu_new
     else:
         if opts.nsubsets > 1:
-            Jn, _ = lm_mod.os_lm_solve(prob, J0, maxiter=opts.max_iter,
+            Jn, _ = lm_mod.os_lm_solve(prob, J0, maxiter=mi,
                                        nsubsets=opts.nsubsets)
         else:
-            Jn = _inner_solve(prob, J0, opts, opts.max_iter)
+            Jn = _inner_solve(prob, J0, opts, mi)
     Jprev = [state.cluster_J(ci).clone() for ci in group]
     off = 0
     for gi, ci in enumerate(group):
@@ -354,10 +364,31 @@ def sagefit(state, cohs, tile, bb, opts, flags=None, admm_terms=None):
     G = opts.em_group
     groups = [list(range(g, min(g + G, state.M)))
               for g in range(0, state.M, G)]
+    nerr = [1.0 / len(groups)] * len(groups)
+    weighted = False
     for em in range(opts.max_emiter):
-        for group in groups:
+        red = []
+        for gi, group in enumerate(groups):
+            itermax = None
+            if weighted:
+                # lmfit.c:880 formula, quantized to a few buckets so the
+                # captured-graph cache stays small
+                raw = (0.2 * nerr[gi] * len(groups) * opts.max_iter
+                       + 0.8 * opts.max_iter)
+                buckets = [max(2, int(0.8 * opts.max_iter)),
+                           opts.max_iter,
+                           int(1.5 * opts.max_iter)]
+                itermax = min(buckets, key=lambda b: abs(b - raw))
+            r_before = float((res[valid].abs() ** 2).sum())
             res = _solve_group(state, group, res, cohs, bb, T, Nbase, B,
-                               opts, admm_terms)
+                               opts, admm_terms, itermax=itermax)
+            red.append(max(r_before
+                           - float((res[valid].abs() ** 2).sum()), 0.0))
+        tot = sum(red)
+        if tot > 0:
+            nerr = [r / tot for r in red]
+        if opts.randomize:
+            weighted = not weighted
         # divergence guard (fullbatch_mode.cpp:622-632 resets on blow-up):
         rn = resnorm(res)
         if not (rn == rn) or rn > 5.0 * res_0:

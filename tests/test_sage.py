@@ -170,3 +170,19 @@ def test_negative_cluster_id_not_subtracted():
     Vall = sage.total_model(state, cohs, bb, tile.tilesz, tile.Nbase)
     full_res = float((tile.xo[0] - Vall).abs().pow(2).mean())
     assert full_res < 0.05 * res
+
+
+def test_weighted_iteration_allocation_converges():
+    """-R 1 (lmfit.c weighted_iter): alternate EM sweeps give groups
+    iteration budgets proportional to their cost reduction; convergence
+    is preserved."""
+    ms, pack = setup_ms(M=3, noise=1e-3, seed=6)
+    tile = ms.load_tile(0)
+    bb = ms.bb_tensor()
+    state = sage.CalState(pack, ms.N)
+    cohs = sage.precalc_coherencies(pack, tile)
+    opts = sage.SageSolveOptions(max_emiter=6, max_iter=10,
+                                 solver_mode=SM_LM_LBFGS, em_group=1,
+                                 randomize=True)
+    res0, res1 = sage.sagefit(state, cohs, tile, bb, opts)
+    assert res1 < 0.05 * res0
