@@ -77,6 +77,9 @@ def build_argparser():
     ap.add_argument('-P', dest='npoly', type=int, default=2)
     ap.add_argument('-Q', dest='polytype', type=int, default=0)
     ap.add_argument('-r', dest='admm_rho', type=float, default=5.0)
+    ap.add_argument('-U', dest='use_global', type=int, default=0,
+                    help='1: residuals from the consensus polynomial '
+                         'solution (stochastic -w > 1)')
     ap.add_argument('-R', dest='randomize', type=int, default=0,
                     help='1: alternate EM sweeps reallocate LM iterations '
                          'to high-error groups (lmfit.c weighted_iter)')
@@ -308,7 +311,7 @@ def run_stochastic(args):
                 cal.epoch(tile, bb, nmb=args.minibatches,
                           lbfgs_iters=args.max_lbfgs,
                           robust_nu=(args.nulow + args.nuhigh) / 2)
-        xres = cal.residuals(tile, bb)
+        xres = cal.residuals(tile, bb, use_global=bool(args.use_global))
         ms.write_column(args.outcol, ti, xres)
         print(f"tile {ti}: stochastic epochs {args.epochs}, "
               f"res {float(xres.abs().pow(2).mean()):.6f}")

@@ -221,8 +221,20 @@ class MinibatchConsensusCalibration:
                     [Zs[r][ci, p] for r in range(self.world)]).mean(dim=0)
         return out
 
-    def residuals(self, tile, bb):
-        """Per-channel residuals using each channel's mini-band solution."""
+    def global_band_J(self, bi):
+        """Consensus solution for mini-band bi: J = sum_p B[bi,p] Z_p
+        (the -U use-global path, minibatch_consensus_mode.cpp)."""
+        flat = poly_mod.eval_poly_jones(self.Z, self.B[bi]).to(self.dtype)
+        out = torch.zeros(self.Mt, self.N, 2, 2, dtype=self.dtype,
+                          device=self.device)
+        for ci in range(self.M):
+            o = self.chunk_off[ci]
+            out[o:o + self.nchunks[ci]] = flat[ci]
+        return out
+
+    def residuals(self, tile, bb, use_global=False):
+        """Per-channel residuals using each channel's mini-band solution
+        (or the consensus polynomial solution when use_global)."""
         out = torch.empty_like(tile.xo)
         T, Nbase = tile.tilesz, tile.Nbase
         fdelta_ch = tile.fdelta / len(tile.freqs)
@@ -236,7 +248,8 @@ class MinibatchConsensusCalibration:
                 cohs = cohs.to(self.dtype)
             V = torch.zeros(tile.x.shape, dtype=self.dtype,
                             device=self.device)
-            J = self.states[bi].J
+            J = self.global_band_J(bi) if use_global \
+                else self.states[bi].J
             for ci in range(self.M):
                 rows = R.chunk_rows_for(ci, self.nchunks, T, Nbase,
                                         tile.x.shape[0], self.device)

@@ -131,3 +131,20 @@ def test_federated_two_rank_gloo(tmp_path):
             tmp_path / f'fed{rank}.txt').read_text().split())
         assert res1 < 0.1 * res0, f"rank {rank}: {res0} -> {res1}"
         assert zdiff < 0.05, f"federated Z diverged: {zdiff}"
+
+
+def test_global_consensus_residuals():
+    """-U 1: residuals from the consensus polynomial B Z instead of the
+    per-band states; both reduce the data."""
+    from sagecal_amd.solvers.stochastic import MinibatchConsensusCalibration
+    pack, ms = _bandpass_ms()
+    tile = ms.load_tile(0)
+    bb = ms.bb_tensor()
+    cal = MinibatchConsensusCalibration(pack, ms.N, ms.freqs, nsolbw=2,
+                                        Npoly=2, rho=0.5)
+    res_before = float(tile.xo.abs().pow(2).mean())
+    for epoch in range(6):
+        cal.epoch(tile, bb, nmb=2, lbfgs_iters=8, robust_nu=10.0)
+    xg = cal.residuals(tile, bb, use_global=True)
+    res_g = float(xg.abs().pow(2).mean())
+    assert res_g < 0.15 * res_before, (res_before, res_g)
