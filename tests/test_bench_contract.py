@@ -82,3 +82,25 @@ def test_bench_two_rank_launch():
     d = json.loads(line)
     _check(d)
     assert d['config']['parallelism'].endswith('2')
+
+
+def test_bench_eight_rank_launch():
+    """The driver's 8-GPU scale run shape, on CPU/gloo: 8 ranks, one
+    band each, fused consensus all-reduce; rank 0 prints the aggregate
+    JSON line."""
+    env = dict(os.environ)
+    env['SAGECAL_BENCH_BACKEND'] = 'gloo'
+    cmd = [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+           '--nproc-per-node', '8', '--master-addr', '127.0.0.1',
+           '--master-port', '29549', os.path.join(ROOT, 'bench.py'),
+           '--gpus', '8', '--cpu', '--steps', '1', '--warmup', '0',
+           '--stations', '8', '--dirs', '2', '--srcs', '2',
+           '--tilesz', '2', '--chan', '2']
+    out = subprocess.run(cmd, capture_output=True, text=True, cwd=ROOT,
+                         env=env, timeout=900)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.strip().splitlines()
+            if l.startswith('{')][-1]
+    d = json.loads(line)
+    _check(d)
+    assert d['config']['parallelism'].endswith('8')
