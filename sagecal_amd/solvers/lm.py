@@ -60,7 +60,8 @@ def lm_solve(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
             J, info = _lm_solve_graphed(prob, J, n, tau, eps1, eps2)
             done += n
             c = float(info['final_cost'].sum())
-            if cost_prev is not None and                     abs(cost_prev - c) < 1e-4 * abs(cost_prev):
+            if (cost_prev is not None
+                    and abs(cost_prev - c) < 1e-4 * abs(cost_prev)):
                 break
             cost_prev = c
         return J, info
