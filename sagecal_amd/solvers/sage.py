@@ -379,15 +379,20 @@ def sagefit(state, cohs, tile, bb, opts, flags=None, admm_terms=None):
                            opts.max_iter,
                            int(1.5 * opts.max_iter)]
                 itermax = min(buckets, key=lambda b: abs(b - raw))
-            r_before = float((res[valid].abs() ** 2).sum())
+            if opts.randomize:
+                # per-group cost tracking syncs the device; only pay for
+                # it when -R weighted allocation is on
+                r_before = float((res[valid].abs() ** 2).sum())
             res = _solve_group(state, group, res, cohs, bb, T, Nbase, B,
                                opts, admm_terms, itermax=itermax)
-            red.append(max(r_before
-                           - float((res[valid].abs() ** 2).sum()), 0.0))
-        tot = sum(red)
-        if tot > 0:
-            nerr = [r / tot for r in red]
+            if opts.randomize:
+                red.append(max(r_before
+                               - float((res[valid].abs() ** 2).sum()),
+                               0.0))
         if opts.randomize:
+            tot = sum(red)
+            if tot > 0:
+                nerr = [r / tot for r in red]
             weighted = not weighted
         # divergence guard (fullbatch_mode.cpp:622-632 resets on blow-up):
         rn = resnorm(res)
