@@ -250,18 +250,11 @@ class ConsensusADMM:
         the image_basis normalization (4 pi^2/beta^2 prefactor, argument
         2 pi l / beta), so the raw-phi series scale is beta/(2 pi)
         (sagecal_master.cpp spatial model -> diffuse_predict.c hand-off)."""
-        if self.Zspat is None or self._Phi is None:
+        Zsp = self.spatial_coefficients()
+        if Zsp is None:
             return None, None
-        from . import fista as fista_mod
-        M = self.state.M
         N = self.Z.shape[2]
         Npoly = self.Npoly
-        # refit to recover the coefficient matrix (Zspat stores Phi @ Zsp)
-        lam, mu_l1, order, fiters, cadence = self.spatial
-        Zb = torch.view_as_real(self.Z).reshape(M, -1)
-        Zb = torch.complex(Zb[:, 0::2], Zb[:, 1::2])
-        Zsp = fista_mod.update_spatialreg_fista(
-            Zb.cpu(), self._Phi, lam=lam, mu=mu_l1, maxiter=fiters)
         G = Zsp.shape[1]
         Zsp = Zsp.reshape(Npoly, N, 2, 2, G)
         Bf = self.Bf.to(Zsp.dtype)
