@@ -396,3 +396,16 @@ def test_create_clusters(tmp_path):
     clusters = skymod.read_sky_cluster(str(tmp_path / 'sky.txt'), out,
                                        0.0, np.pi / 4, 150e6)
     assert sum(c.nsrc for c in clusters) == 3
+
+
+def test_change_freq(tmp_path):
+    from sagecal_amd.apps import change_freq
+    msf = str(tmp_path / 'a.npz')
+    msdata.make_synthetic_npz(msf, N=5, tilesz=2, Ntime=2, Nchan=2)
+    out = str(tmp_path / 'b.npz')
+    rc = change_freq.main(['-d', msf, '-f', '160e6', '-o', out])
+    assert rc == 0
+    z = np.load(out)
+    assert abs(float(np.mean(z['freqs'])) - 160e6) < 1.0
+    ms = msdata.NpzMS(out)
+    assert abs(ms.freq0 - 160e6) < 1.0
