@@ -173,7 +173,8 @@ def main(argv=None):
         if cohs.dtype != cdtype:
             cohs = cohs.to(cdtype)
         res0, res1 = adm.run(cohs, tile, bb, opts, n_admm=args.nadmm,
-                             flags=flags)
+                             flags=flags,
+                             verbose=args.verbose and rank == 0)
         if args.use_global:
             state.J = adm.global_solution()
         ccid = args.ccid if args.ccid != -99999 else None

@@ -179,7 +179,8 @@ class ConsensusADMM:
         self.Yhat_prev = Yhat.clone()
         self.J_prev = self.state.J.clone()
 
-    def run(self, cohs, tile, bb, opts, n_admm=10, flags=None):
+    def run(self, cohs, tile, bb, opts, n_admm=10, flags=None,
+            verbose=False):
         """The per-tile ADMM loop (sagecal_master.cpp:731-1060 semantics).
         Returns (res0, res1) of the final local solve."""
         res0 = res1 = None
@@ -194,6 +195,10 @@ class ConsensusADMM:
             if res0 is None:
                 res0 = r0
             res1 = r1
+            # pre-consensus divergence guard (sagecal_slave.cpp:798-803):
+            # a blown-up local solve must not contaminate the global Z
+            if not (r1 == r1) or (r0 > 0 and r1 > 5.0 * r0):
+                st.reset()
             self.z_update()
             BZ = self.bz()
             self.y_update(BZ)
@@ -203,6 +208,14 @@ class ConsensusADMM:
                 lam, mu_l1, order, fiters, cadence = self.spatial
                 if (it + 1) % max(cadence, 1) == 0:
                     self.spatial_update(lam, mu_l1, fiters)
+            if verbose:
+                # primal ||J - BZ|| and dual ||Z - Zold|| residual norms
+                # (sagecal_master.cpp:881-885 / sagecal_slave.cpp:911-919)
+                primal = float((st.J - BZ).abs().norm()) / max(st.Mt, 1)
+                dual = 0.0 if BZ_old is None else \
+                    float((BZ - BZ_old).abs().norm()) / max(st.Mt, 1)
+                print(f"ADMM {it}: res {r1:.6f} primal {primal:.3e} "
+                      f"dual {dual:.3e} rho[0] {float(self.rho[0]):.2f}")
         return res0, res1
 
     def spatial_update(self, lam, mu_l1, fiters):

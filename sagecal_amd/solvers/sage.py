@@ -366,7 +366,11 @@ def sagefit(state, cohs, tile, bb, opts, flags=None, admm_terms=None):
               for g in range(0, state.M, G)]
     nerr = [1.0 / len(groups)] * len(groups)
     weighted = False
+    import os as _os
+    import time as _time
+    trace = _os.environ.get('SAGECAL_TRACE') == '1'
     for em in range(opts.max_emiter):
+        t_em = _time.perf_counter() if trace else 0.0
         red = []
         for gi, group in enumerate(groups):
             itermax = None
@@ -400,6 +404,11 @@ def sagefit(state, cohs, tile, bb, opts, flags=None, admm_terms=None):
             state.reset()
             V = total_model(state, cohs, bb, T, Nbase)
             res = x - V
+        if trace:
+            if x.is_cuda:
+                torch.cuda.synchronize()
+            dt_ms = 1e3 * (_time.perf_counter() - t_em)
+            print(f"[trace] EM {em}: {dt_ms:.1f} ms, res {rn:.6f}")
 
     if opts.joint_iters > 0:
         weights = None
