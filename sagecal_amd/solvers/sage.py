@@ -337,7 +337,10 @@ def _solve_group(state, group, res, cohs, bb, T, Nbase, B, opts,
 
 def sagefit(state, cohs, tile, bb, opts, flags=None, admm_terms=None):
     """The SAGE EM loop (lmfit.c:777-1053). Returns (res_0, res_1): initial
-    and final residual norms (per-visibility RMS like lmfit.c res_0/res_1).
+    and final residual scales. NOTE on normalization: we report the
+    per-visibility RMS sqrt(sum|r|^2 / n_real); the reference prints
+    ||r||_2 / n_real (lmfit.c:869) — a factor sqrt(n_real) smaller. Both
+    are monotone in the same quantity; ratios res_1/res_0 agree exactly.
 
     cohs: [M, B, 2, 2] cluster coherencies (channel-averaged).
     """
