@@ -245,3 +245,33 @@ def test_predict_withbeam_element_modes():
     # scalar array weight per baseline (off-diagonals are 0/0)
     ratio = (w1[..., 0, 0] / s0[..., 0, 0])[..., None, None]
     assert torch.allclose(w2, ratio * w3, atol=1e-8)
+
+
+def test_create_beam_model_roundtrip(tmp_path):
+    """Fit sampled patterns generated FROM known coefficients; the tool
+    must recover them and element_beam must reproduce the samples
+    (create_header.py analog)."""
+    from sagecal_amd import beams
+    from sagecal_amd.apps import create_beam_model as cbm
+    n0 = 3
+    rng = np.random.default_rng(4)
+    nm = beams.n_modes(n0)
+    true = rng.standard_normal(nm) + 1j * rng.standard_normal(nm)
+    theta = np.linspace(0.0, np.pi / 2, 12)
+    phi = np.linspace(0.0, 2 * np.pi, 24, endpoint=False)
+    th, ph = np.meshgrid(theta, phi, indexing='ij')
+    B = beams.sharmonic_basis(th.ravel(), ph.ravel(), n0)
+    et = (B @ true).reshape(1, 12, 24)
+    ep = (B @ (0.3 * true)).reshape(1, 12, 24)
+    for name, arr in (('theta.npy', theta), ('phi.npy', phi),
+                      ('frequency.npy', np.array([150e6])),
+                      ('etheta.npy', et), ('ephi.npy', ep)):
+        np.save(tmp_path / name, arr)
+    out = str(tmp_path / 'co.npz')
+    rc = cbm.main(['-d', str(tmp_path), '--order', str(n0), '-o', out])
+    assert rc == 0
+    co = beams.ElementCoeffs.load(out)
+    assert np.allclose(co.ctheta_x[0], true, atol=1e-8)
+    # evaluated pattern matches the input samples at grid points
+    E = beams.element_beam(co, ph.ravel(), np.pi / 2 - th.ravel(), 150e6)
+    assert np.allclose(E[:, 0, 0].numpy(), et.ravel(), atol=1e-8)
