@@ -110,7 +110,17 @@ def read_sky_model(path, fmt=0, modes_dir=None):
     sources = {}
     for toks in _parse_sky_lines(path):
         name = toks[0]
-        vals = [float(t) for t in toks[1:]]
+        need = 18 if fmt == 1 else 16
+        try:
+            vals = [float(t) for t in toks[1:]]
+            if len(vals) < need:
+                raise ValueError(
+                    f"{len(vals)} numeric columns, need {need}")
+        except ValueError as e:
+            raise ValueError(
+                f"sky model {path}: bad line for source '{name}': {e} "
+                f"(format {fmt}: name + {need} columns per README §2c)"
+            ) from e
         if fmt == 1:
             (rahr, ramin, rasec, decd, decmin, decsec, sI, sQ, sU, sV,
              si0, si1, si2, RM, eX, eY, eP, f0) = vals[:18]
