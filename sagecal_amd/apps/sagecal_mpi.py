@@ -172,6 +172,8 @@ def main(argv=None):
         cohs = sage.precalc_coherencies(pack, tile)
         if cohs.dtype != cdtype:
             cohs = cohs.to(cdtype)
+        # TAG_FRATIO: weigh this band's rho by its unflagged fraction
+        adm.set_fratio(float((~flags).float().mean()))
         res0, res1 = adm.run(cohs, tile, bb, opts, n_admm=args.nadmm,
                              flags=flags,
                              verbose=args.verbose and rank == 0)

@@ -52,6 +52,7 @@ class ConsensusADMM:
         self.rho_upper = rho_upper
         self.use_bb = use_bb
         self.alpha = federated_alpha
+        self.fratio = torch.ones(world)      # per-band flag-ratio weights
         self._update_bii()
         # consensus state (per rank): Y dual, Z poly coefficients
         self.Y = torch.zeros(Mt, N, 2, 2, dtype=self.cdtype, device=dev)
@@ -77,10 +78,22 @@ class ConsensusADMM:
                                                 order, beta)
 
     def _update_bii(self):
-        # rho may differ per band after fratio scaling; here identical
-        # across bands (reference scales by flag ratio; we expose scale())
-        rho_mf = self.rho[:, None].expand(-1, self.world)
+        # per-band effective rho = rho * fratio_f (sagecal_master.cpp:720:
+        # bands with more flagged data weigh less in the Z fit)
+        rho_mf = self.rho[:, None] * self.fratio[None, :].double()
         self.Bii = poly.find_prod_inverse(self.B, rho_mf, self.alpha)
+
+    def set_fratio(self, unflagged_fraction):
+        """Exchange per-band unflagged-data fractions (TAG_FRATIO) and
+        rebuild the Bii inverses with fratio-scaled rho."""
+        fr = torch.zeros(self.world)
+        fr[self.rank] = float(unflagged_fraction)
+        if self.world > 1 and dist.is_initialized():
+            dist.all_reduce(fr, group=self.group)
+        else:
+            fr[:] = float(unflagged_fraction)
+        self.fratio = fr.clamp_min(1e-3)
+        self._update_bii()
 
     def _chunk_expand(self, Zj):
         """B_f Z -> per-chunk [Mt, N, 2, 2] (each chunk of cluster ci gets
@@ -110,7 +123,7 @@ class ConsensusADMM:
             dist.all_reduce(t, group=self.group)
         return t
 
-    def z_update(self):
+    def z_update(self, first=False):
         """Global Z from all bands: fused all-reduce of B_f (x) (Y + rho J)
         per cluster (collapsed TAG_YDATA + update_global_z_multi +
         TAG_CONSENSUS, sagecal_master.cpp:813-877)."""
@@ -126,8 +139,14 @@ class ConsensusADMM:
             Ym = torch.stack([self.Y[st.chunk_off[ci]:st.chunk_off[ci]
                                      + st.nchunks[ci]].mean(dim=0)
                               for ci in range(M)])
-        rho_m = self.rho.to(device=self.dev).to(Jm.real.dtype)
-        contrib = Ym + rho_m[:, None, None, None].to(self.cdtype) * Jm
+        rho_m = (self.rho * float(self.fratio[self.rank])).to(
+            device=self.dev).to(Jm.real.dtype)
+        if first:
+            # admm==0: Y already holds rho * (gauge-unified J); the Z fit
+            # uses it alone (sagecal_master.cpp:843-877 at admm==0)
+            contrib = Ym
+        else:
+            contrib = Ym + rho_m[:, None, None, None].to(self.cdtype) * Jm
         Bfd = self.Bf.to(device=self.dev, dtype=Jm.real.dtype)
         acc = (Bfd[None, :, None, None, None].to(self.cdtype)
                * contrib[:, None]).contiguous()
@@ -144,12 +163,17 @@ class ConsensusADMM:
         Zj = poly.eval_poly_jones(self.Z, self.Bf)
         return self._chunk_expand(Zj)
 
-    def y_update(self, BZ):
-        """Y <- Y + rho (J - BZ) (sagecal_slave.cpp:870-888)."""
+    def y_update(self, BZ, first=False):
+        """Y <- Y + rho (J - BZ) (sagecal_slave.cpp:870-888). At admm==0
+        Y already holds rho*Jbar (the unified gauge), so the update is
+        Y <- Y - rho BZ (slave :856-874: 'we already have Y + rho J')."""
         st = self.state
         rho_chunk = self._rho_chunk()
-        self.Y = self.Y + rho_chunk[:, None, None, None].to(self.cdtype) * \
-            (st.J - BZ)
+        rc = rho_chunk[:, None, None, None].to(self.cdtype)
+        if first:
+            self.Y = self.Y - rc * BZ
+        else:
+            self.Y = self.Y + rc * (st.J - BZ)
 
     def bb_update(self, BZ_old):
         """Barzilai-Borwein rho (sagecal_slave.cpp:899-904 +
@@ -199,9 +223,14 @@ class ConsensusADMM:
             # a blown-up local solve must not contaminate the global Z
             if not (r1 == r1) or (r0 > 0 and r1 > 5.0 * r0):
                 st.reset()
-            self.z_update()
+            if it == 0:
+                # gauge unification (sagecal_master.cpp:827 at admm==0):
+                # manifold-average J across bands and seed Y = rho Jbar so
+                # the first Z fit sees one common unitary frame
+                self._unify_gauge()
+            self.z_update(first=(it == 0))
             BZ = self.bz()
-            self.y_update(BZ)
+            self.y_update(BZ, first=(it == 0))
             if self.use_bb and BZ_old is not None:
                 self.bb_update(BZ_old)
             if self.spatial is not None and self._Phi is not None:
@@ -217,6 +246,27 @@ class ConsensusADMM:
                 print(f"ADMM {it}: res {r1:.6f} primal {primal:.3e} "
                       f"dual {dual:.3e} rho[0] {float(self.rho[0]):.2f}")
         return res0, res1
+
+    def _unify_gauge(self):
+        """Replicated analog of calculate_manifold_average at admm==0:
+        all-gather every band's J, average each cluster-chunk over bands
+        up to the common unitary, and seed Y = rho * Jbar (the slaves'
+        received 'unified Y', sagecal_slave.cpp:836-841)."""
+        from . import manifold as man
+        st = self.state
+        Jr = torch.view_as_real(st.J.cpu().to(torch.complex128))
+        if self.world > 1 and dist.is_initialized():
+            gath = [torch.zeros_like(Jr) for _ in range(self.world)]
+            dist.all_gather(gath, Jr, group=self.group)
+            Jall = torch.stack([torch.view_as_complex(g) for g in gath])
+        else:
+            Jall = torch.view_as_complex(Jr).unsqueeze(0)
+        Jbar = torch.empty_like(Jall[0])
+        for t in range(st.Mt):
+            Jbar[t] = man.manifold_average(Jall[:, t], niter=5)[0]
+        rho_chunk = self._rho_chunk().cpu().double()
+        self.Y = (rho_chunk[:, None, None, None] * Jbar).to(
+            device=self.dev, dtype=self.cdtype)
 
     def spatial_update(self, lam, mu_l1, fiters):
         """Fit the spatial (elastic-net shapelet) model to the per-cluster
