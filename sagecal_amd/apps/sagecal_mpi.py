@@ -72,6 +72,71 @@ def build_argparser():
     return ap
 
 
+def _run_multiplexed(args, names, rank, world, device, dtype, cdtype):
+    from .. import sky as skymod, msdata
+    from ..ops.reference import SourcePack
+    from ..solvers import sage
+    from ..consensus.admm import MultiplexedADMM
+    my_names = names[rank::world]
+    mss, bandsets, tilesets, my_ids, f0_local = [], [], [], [], []
+    for i, name in enumerate(my_names):
+        ms = msdata.open_ms(name, tilesz=args.tilesz, device=device,
+                            dtype=dtype)
+        mss.append(ms)
+        my_ids.append(rank + i * world)
+        f0_local.append(ms.freq0)
+    # gather every band's centre frequency
+    F = len(names)
+    f0s = torch.zeros(F)
+    for bi, f0 in zip(my_ids, f0_local):
+        f0s[bi] = f0
+    if world > 1:
+        dist.all_reduce(f0s)
+    freq0_global = float(f0s.mean())
+    clusters = skymod.read_sky_cluster(args.sky, args.cluster,
+                                       mss[0].ra0, mss[0].dec0,
+                                       freq0_global, fmt=args.format)
+    pack = SourcePack(clusters)
+    rho = torch.full((pack.M,), args.admm_rho)
+    opts = sage.SageSolveOptions(
+        max_emiter=args.max_emiter, max_iter=args.max_iter,
+        solver_mode=args.solver_mode, robust_nulow=args.nulow,
+        robust_nuhigh=args.nuhigh, lbfgs_iters=max(args.max_lbfgs, 0))
+    for ms in mss:
+        bandsets.append({'state': sage.CalState(pack, ms.N,
+                                                device=device,
+                                                dtype=cdtype),
+                         'freq0': ms.freq0})
+    adm = MultiplexedADMM(bandsets, my_ids, f0s.tolist(), freq0_global,
+                          rank, world, Npoly=args.npoly,
+                          poly_type=args.polytype, rho=rho)
+    for ti_all in range(min(ms.n_tiles() for ms in mss)):
+        tilesets = []
+        for ms in mss:
+            tile = ms.load_tile(ti_all)
+            cohs = sage.precalc_coherencies(pack, tile)
+            if cohs.dtype != cdtype:
+                cohs = cohs.to(cdtype)
+            tilesets.append({'cohs': cohs, 'tile': tile,
+                             'bb': ms.bb_tensor(device=device)})
+        res = adm.run(tilesets, opts, n_admm=args.nadmm)
+        for k, ms in enumerate(mss):
+            st = bandsets[k]['state']
+            xres = sage.calculate_residuals_multifreq(st, pack,
+                                                      tilesets[k]['tile'],
+                                                      tilesets[k]['bb'])
+            ms.write_column(args.outcol, ti_all, xres)
+        if rank == 0 or args.verbose:
+            rr = {bi: (round(a, 4), round(b, 4))
+                  for bi, (a, b) in res.items()}
+            print(f"rank {rank} tile {ti_all}: bands {rr}")
+    for ms in mss:
+        ms.save()
+    if world > 1:
+        dist.destroy_process_group()
+    return 0
+
+
 def main(argv=None):
     args = build_argparser().parse_args(argv)
     rank = int(os.environ.get('RANK', '0'))
@@ -101,8 +166,13 @@ def main(argv=None):
             print(f"no MS matches {args.mslist}", file=sys.stderr)
             return 1
     torch.set_num_threads(max(1, args.nthreads))
-    my_ms = names[rank % len(names)] if world <= len(names) else \
-        names[rank]
+    if len(names) > world:
+        # more MSs than ranks: this rank owns names[rank::world] and
+        # rotates through them per ADMM iteration
+        # (sagecal_master.cpp:1055 Scurrent multiplexing)
+        return _run_multiplexed(args, names, rank, world, device, dtype,
+                                cdtype)
+    my_ms = names[rank % len(names)]
     ms = msdata.open_ms(my_ms, tilesz=args.tilesz, device=device,
                         dtype=dtype)
     clusters = skymod.read_sky_cluster(args.sky, args.cluster, ms.ra0,

@@ -344,3 +344,117 @@ class ConsensusADMM:
     def global_solution(self):
         """J = B_f Z (use_global_solution path, sagecal_master:1064)."""
         return self._chunk_expand(poly.eval_poly_jones(self.Z, self.Bf))
+
+
+class MultiplexedADMM:
+    """Consensus ADMM with MORE frequency bands than ranks: each rank
+    owns a slice of the MS list and rotates through them, one local
+    solve per ADMM iteration per rank (sagecal_master.cpp Scurrent
+    rotation :1055-1060; slave mmid multiplexing). The Z fit always sums
+    over ALL bands — non-current bands contribute their cached
+    (Y + rho J) — mirroring the master's persistent per-MS Y buffers.
+
+    band_sets: list of per-band dicts owned by THIS rank, each with keys
+    'state' (CalState), 'freq0'; freqs_all covers every band globally;
+    my_band_ids are this rank's global band indices.
+    """
+
+    def __init__(self, band_sets, my_band_ids, freqs_all, freq0, rank,
+                 world, Npoly=2, poly_type=0, rho=None, group=None):
+        self.bands = band_sets
+        self.ids = list(my_band_ids)
+        self.rank, self.world = rank, world
+        self.group = group
+        self.F = len(freqs_all)
+        self.Npoly = min(Npoly, self.F)
+        st0 = band_sets[0]['state']
+        self.M, self.N, self.Mt = st0.M, st0.N, st0.Mt
+        self.dev = st0.J.device
+        self.cdtype = st0.J.dtype
+        self.B = poly.setup_polynomials(freqs_all, freq0, self.Npoly,
+                                        poly_type)
+        if rho is None:
+            rho = torch.full((self.M,), 5.0)
+        self.rho = rho.clone().double()
+        rho_mf = self.rho[:, None].expand(-1, self.F)
+        self.Bii = poly.find_prod_inverse(self.B, rho_mf)
+        for b in self.bands:
+            b['Y'] = torch.zeros(self.Mt, self.N, 2, 2,
+                                 dtype=self.cdtype, device=self.dev)
+        self.Z = torch.zeros(self.M, self.Npoly, self.N, 2, 2,
+                             dtype=self.cdtype, device=self.dev)
+        self.cur = 0
+
+    def _cluster_mean(self, T):
+        st = self.bands[0]['state']
+        if st.Mt == st.M:
+            return T
+        return torch.stack([T[st.chunk_off[ci]:st.chunk_off[ci]
+                              + st.nchunks[ci]].mean(dim=0)
+                            for ci in range(st.M)])
+
+    def _chunk_expand(self, Zj, st):
+        if st.Mt == st.M:
+            return Zj.to(self.cdtype)
+        out = torch.empty(st.Mt, self.N, 2, 2, dtype=self.cdtype,
+                          device=self.dev)
+        for ci in range(st.M):
+            o = st.chunk_off[ci]
+            out[o:o + st.nchunks[ci]] = Zj[ci]
+        return out
+
+    def _rho_chunk(self, st):
+        if st.Mt == st.M:
+            return self.rho.to(self.dev)
+        return torch.cat([
+            torch.full((st.nchunks[ci],), float(self.rho[ci]))
+            for ci in range(st.M)]).to(self.dev)
+
+    def bz(self, bi, st):
+        Zj = poly.eval_poly_jones(self.Z, self.B[bi])
+        return self._chunk_expand(Zj, st)
+
+    def z_update(self):
+        """Allreduce of sum over ALL owned bands of B_b (x) (Y_b +
+        rho J_b) — stale contributions for the bands not solved this
+        iteration, like the master's persistent Y (sagecal_master.cpp)."""
+        acc = torch.zeros(self.M, self.Npoly, self.N, 2, 2,
+                          dtype=self.cdtype, device=self.dev)
+        rho_m = self.rho.to(self.dev).to(torch.float64)
+        for bi, b in zip(self.ids, self.bands):
+            st = b['state']
+            Jm = self._cluster_mean(st.J)
+            Ym = self._cluster_mean(b['Y'])
+            contrib = Ym + rho_m[:, None, None, None].to(self.cdtype) * Jm
+            Bb = self.B[bi].to(torch.float64)
+            acc = acc + (Bb[None, :, None, None, None].to(self.cdtype)
+                         * contrib[:, None])
+        if self.world > 1 and dist.is_initialized():
+            dist.all_reduce(torch.view_as_real(acc), group=self.group)
+        self.Z = poly.update_global_z(acc, self.Bii)
+
+    def run(self, tiles, opts, n_admm=10):
+        """tiles: list aligned with band_sets of dicts holding 'cohs',
+        'tile', 'bb'. One rotating local solve per ADMM iteration."""
+        res = {}
+        for it in range(n_admm):
+            k = self.cur
+            b = self.bands[k]
+            bi = self.ids[k]
+            st = b['state']
+            t = tiles[k]
+            admm_terms = None
+            if it >= len(self.bands):    # every band solved once already
+                BZ = self.bz(bi, st)
+                admm_terms = (self.rho.to(self.dev), b['Y'], BZ)
+            r0, r1 = sage.sagefit(st, t['cohs'], t['tile'], t['bb'],
+                                  opts, admm_terms=admm_terms)
+            res.setdefault(bi, [r0, r1])[1] = r1
+            self.z_update()
+            BZ = self.bz(bi, st)
+            rc = self._rho_chunk(st)[:, None, None, None].to(self.cdtype)
+            b['Y'] = b['Y'] + rc * (st.J - BZ)
+            # rotate to the next owned band (Scurrent increment)
+            if len(self.bands) > 1:
+                self.cur = (self.cur + 1) % len(self.bands)
+        return res
