@@ -251,22 +251,34 @@ class ConsensusADMM:
         """Replicated analog of calculate_manifold_average at admm==0:
         all-gather every band's J, average each cluster-chunk over bands
         up to the common unitary, and seed Y = rho * Jbar (the slaves'
-        received 'unified Y', sagecal_slave.cpp:836-841)."""
-        from . import manifold as man
+        received 'unified Y', sagecal_slave.cpp:836-841). For world==1
+        the average is exactly J (polar of a PSD Gram is identity), so
+        seed directly — no gather, no sync-heavy averaging.
+
+        The multi-band average is fully batched over (band, chunk):
+        A_{f,t} = sum_s J_{f,t,s}^H Jbar_{t,s}; U = polar(A); aligned
+        mean — no per-station python loops (manifold_average.c:204
+        vectorized)."""
         st = self.state
+        rho_chunk = self._rho_chunk().to(self.dev)
+        if self.world == 1 or not dist.is_initialized():
+            self.Y = rho_chunk[:, None, None, None].to(self.cdtype) * st.J
+            return
+        from .manifold import polar_unitary
         Jr = torch.view_as_real(st.J.cpu().to(torch.complex128))
-        if self.world > 1 and dist.is_initialized():
-            gath = [torch.zeros_like(Jr) for _ in range(self.world)]
-            dist.all_gather(gath, Jr, group=self.group)
-            Jall = torch.stack([torch.view_as_complex(g) for g in gath])
-        else:
-            Jall = torch.view_as_complex(Jr).unsqueeze(0)
-        Jbar = torch.empty_like(Jall[0])
-        for t in range(st.Mt):
-            Jbar[t] = man.manifold_average(Jall[:, t], niter=5)[0]
-        rho_chunk = self._rho_chunk().cpu().double()
-        self.Y = (rho_chunk[:, None, None, None] * Jbar).to(
-            device=self.dev, dtype=self.cdtype)
+        gath = [torch.zeros_like(Jr) for _ in range(self.world)]
+        dist.all_gather(gath, Jr, group=self.group)
+        Jall = torch.stack([torch.view_as_complex(g)
+                            for g in gath])            # [F, T, N, 2, 2]
+        Jbar = Jall[0].clone()
+        for _ in range(5):
+            A = torch.einsum('ftsji,tsjk->ftik', Jall.conj(), Jbar)
+            U = polar_unitary(A.reshape(-1, 2, 2)).reshape(
+                self.world, st.Mt, 2, 2)
+            aligned = torch.einsum('ftsij,ftjk->ftsik', Jall, U)
+            Jbar = aligned.mean(dim=0)
+        self.Y = (rho_chunk.cpu().double()[:, None, None, None]
+                  * Jbar).to(device=self.dev, dtype=self.cdtype)
 
     def spatial_update(self, lam, mu_l1, fiters):
         """Fit the spatial (elastic-net shapelet) model to the per-cluster
