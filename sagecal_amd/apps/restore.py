@@ -106,7 +106,10 @@ def main(argv=None):
             off = 0
             for c in clusters:
                 Jc = J[off:off + c.nchunk]
-                gains.append(float((Jc.abs() ** 2).mean()))
+                # apparent unpolarized flux under J: I' = tr(J J^H)/2 I
+                # (restore.c withsol scaling)
+                gains.append(float((Jc.abs() ** 2)
+                                   .sum(dim=(-1, -2)).mean() / 2.0))
                 off += c.nchunk
     img = render(clusters, hdr, img0.shape, gains, args.beam)
     if args.add:

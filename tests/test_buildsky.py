@@ -171,3 +171,41 @@ def test_multifrequency_spectral_index(tmp_path):
     si = float(rows[0][11])
     assert abs(flux - 10.0) / 10.0 < 0.1
     assert abs(si - si_true) < 0.1, si
+
+
+def test_restore_with_solutions_scales_flux(tmp_path):
+    """restore -p: cluster fluxes scaled by the mean |J|^2 of that
+    cluster's solutions (restore.c withsol path)."""
+    from sagecal_amd.apps import buildsky, restore
+    from sagecal_amd import solutions, sky as skymod
+    from sagecal_amd.solvers import sage
+    from sagecal_amd.ops.reference import SourcePack
+    import torch
+    img = _gauss_img((48, 48), [(8.0, 24.0, 24.0, 2.0, 2.0)],
+                     noise=0.001)
+    p = str(tmp_path / 'f.fits')
+    fitsio.write_fits_image(p, img, crval=(0.0, 45.0),
+                            cdelt=(-0.01, 0.01))
+    skyf = str(tmp_path / 'sky.txt')
+    clf = str(tmp_path / 'cl.txt')
+    assert buildsky.main(['-f', p, '-s', skyf, '-c', clf, '-Q', '1']) == 0
+    clusters = skymod.read_sky_cluster(skyf, clf, 0.0, np.deg2rad(45.0),
+                                       150e6)
+    pack = SourcePack(clusters)
+    state = sage.CalState(pack, 6)
+    state.J *= 2.0                         # |J|^2 = 4 -> flux x4
+    sol = str(tmp_path / 'sol.txt')
+    w = solutions.SolutionWriter(sol, 150e6, 1e5, 1.0, 6, state.M,
+                                 state.Mt)
+    w.write_tile(state)
+    w.close()
+    out1 = str(tmp_path / 'o1.fits')
+    out2 = str(tmp_path / 'o2.fits')
+    assert restore.main(['-f', p, '-s', skyf, '-c', clf,
+                         '-o', out1]) == 0
+    assert restore.main(['-f', p, '-s', skyf, '-c', clf, '-p', sol,
+                         '-o', out2]) == 0
+    r1, _ = fitsio.read_fits_image(out1)
+    r2, _ = fitsio.read_fits_image(out2)
+    ratio = r2.sum() / r1.sum()
+    assert abs(ratio - 4.0) < 0.2, ratio
