@@ -128,3 +128,21 @@ def test_extract_phases_joint_diagonalization():
                 break
         else:
             raise AssertionError("phase differences not recovered")
+
+
+def test_ncp_taper_properties():
+    """whiten_data / ncp_weight (updatenu.c:310): short baselines are
+    down-weighted, weights are monotone in |uv|, and beyond ~400 lambda
+    the taper is exactly off."""
+    import torch
+    from sagecal_amd.utils import taper
+    ud = torch.tensor([0.0, 50.0, 200.0, 399.0, 401.0, 5000.0])
+    w = taper.ncp_weight(ud)
+    assert float(w[0]) < 0.4                 # strong suppression at 0
+    assert bool((w[1:] >= w[:-1] - 1e-12).all())
+    assert float(w[4]) == 1.0 and float(w[5]) == 1.0
+    x = torch.ones(6, 2, 2, dtype=torch.complex128)
+    u = ud / 150e6
+    v = torch.zeros(6)
+    xw, wts = taper.whiten_data(x, u, v, 150e6)
+    assert torch.allclose(xw[:, 0, 0].real, w)
