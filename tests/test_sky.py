@@ -145,4 +145,4 @@ def test_ncp_taper_properties():
     u = ud / 150e6
     v = torch.zeros(6)
     xw, wts = taper.whiten_data(x, u, v, 150e6)
-    assert torch.allclose(xw[:, 0, 0].real, w)
+    assert torch.allclose(xw[:, 0, 0].real, w.double())
