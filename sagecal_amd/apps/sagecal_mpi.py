@@ -291,6 +291,19 @@ def main(argv=None):
     if rank == 0 and args.solfile and args.spatial:
         Zsp = adm.spatial_coefficients()
         if Zsp is not None:
+            # render the spatial model amplitude as PPM
+            # (plot_spatial_model / pngoutput.c DEBUG dump made a
+            # first-class output)
+            from ..utils import image as img_mod
+            from ..consensus import fista as fista_mod
+            beta = float(max(np.max(np.abs(centroids[0])),
+                             np.max(np.abs(centroids[1])), 1e-3))
+            order = int(args.spatial.split(',')[2])
+            ext = 2.0 * beta
+            img_mod.plot_spatial_model(
+                f"{args.solfile}.spatial.ppm", Zsp,
+                lambda l, m: fista_mod.spatial_basis(l, m, order, beta),
+                grid=64, extent=ext)
             # spatial_<solfile>: rows p of the [P, 2G] real coefficient
             # matrix (sagecal_master.cpp spatial write)
             import os as _os

@@ -155,6 +155,7 @@ def test_sagecal_mpi_two_bands(tmp_path):
     assert ztiles, "global Z solution file missing/empty"
     spf = tmp_path / 'spatial_sol.txt'
     assert spf.exists() and len(spf.read_text().splitlines()) > 2
+    assert (tmp_path / 'sol.txt.spatial.ppm').exists()
     # Npoly=2: the Z file carries twice the J file's values per tile
     # (read_solutions chunks rows in 8N groups, so Npoly groups come back
     # as Npoly entries)
