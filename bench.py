@@ -197,7 +197,8 @@ def main():
         sync()
         elapsed = time.perf_counter() - t0
         if dist:
-            te = torch.tensor([elapsed])
+            te = torch.tensor([elapsed],
+                              device=device if use_gpu else 'cpu')
             torch.distributed.all_reduce(
                 te, op=torch.distributed.ReduceOp.MAX)
             elapsed = float(te)

@@ -87,11 +87,13 @@ def _run_multiplexed(args, names, rank, world, device, dtype, cdtype):
         f0_local.append(ms.freq0)
     # gather every band's centre frequency
     F = len(names)
-    f0s = torch.zeros(F)
+    use_gpu = device != 'cpu'
+    f0s = torch.zeros(F, device=device if use_gpu else 'cpu')
     for bi, f0 in zip(my_ids, f0_local):
         f0s[bi] = f0
     if world > 1:
         dist.all_reduce(f0s)
+    f0s = f0s.cpu()
     freq0_global = float(f0s.mean())
     clusters = skymod.read_sky_cluster(args.sky, args.cluster,
                                        mss[0].ra0, mss[0].dec0,
@@ -187,11 +189,13 @@ def main(argv=None):
         rho = torch.tensor(arho, dtype=torch.float64)
         rho[rho == 0] = args.admm_rho
 
-    # gather every band's centre frequency (TAG_MSAUX metadata exchange)
-    f0s = torch.zeros(world)
+    # gather every band's centre frequency (TAG_MSAUX metadata exchange);
+    # NCCL needs device tensors for collectives
+    f0s = torch.zeros(world, device=device if use_gpu else 'cpu')
     f0s[rank] = ms.freq0
     if world > 1:
         dist.all_reduce(f0s)
+    f0s = f0s.cpu()
     freq0_global = float(f0s.mean())
 
     spatial = None

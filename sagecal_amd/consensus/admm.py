@@ -86,13 +86,14 @@ class ConsensusADMM:
     def set_fratio(self, unflagged_fraction):
         """Exchange per-band unflagged-data fractions (TAG_FRATIO) and
         rebuild the Bii inverses with fratio-scaled rho."""
-        fr = torch.zeros(self.world)
+        # collectives must run on the backend's device (NCCL: cuda)
+        fr = torch.zeros(self.world, device=self.dev)
         fr[self.rank] = float(unflagged_fraction)
         if self.world > 1 and dist.is_initialized():
             dist.all_reduce(fr, group=self.group)
         else:
             fr[:] = float(unflagged_fraction)
-        self.fratio = fr.clamp_min(1e-3)
+        self.fratio = fr.cpu().clamp_min(1e-3)
         self._update_bii()
 
     def _chunk_expand(self, Zj):
@@ -194,10 +195,12 @@ class ConsensusADMM:
                 dY[ci] = torch.view_as_real(dy).reshape(-1).double()
                 dJ[ci] = torch.view_as_real(dj).reshape(-1).double()
             new_rho = poly.update_rho_bb(self.rho, self.rho_upper, dY, dJ)
-            # keep rho consistent across ranks (mean)
+            # keep rho consistent across ranks (mean); reduce on the
+            # backend's device (NCCL: cuda)
             if self.world > 1 and dist.is_initialized():
-                dist.all_reduce(new_rho, group=self.group)
-                new_rho /= self.world
+                nr = new_rho.to(self.dev)
+                dist.all_reduce(nr, group=self.group)
+                new_rho = (nr / self.world).cpu()
             self.rho = new_rho
             self._update_bii()
         self.Yhat_prev = Yhat.clone()
@@ -265,10 +268,12 @@ class ConsensusADMM:
             self.Y = rho_chunk[:, None, None, None].to(self.cdtype) * st.J
             return
         from .manifold import polar_unitary
-        Jr = torch.view_as_real(st.J.cpu().to(torch.complex128))
+        # gather on the backend's device (NCCL: cuda), average on CPU
+        Jr = torch.view_as_real(st.J).contiguous()
         gath = [torch.zeros_like(Jr) for _ in range(self.world)]
         dist.all_gather(gath, Jr, group=self.group)
-        Jall = torch.stack([torch.view_as_complex(g)
+        Jall = torch.stack([torch.view_as_complex(g).cpu()
+                            .to(torch.complex128)
                             for g in gath])            # [F, T, N, 2, 2]
         Jbar = Jall[0].clone()
         for _ in range(5):
