@@ -276,8 +276,9 @@ def main(argv=None):
                 Jall = Jb
             best, scores = mdl_mod.minimum_description_length(
                 torch.view_as_complex(
-                    Jall.reshape(world, state.M, -1, 2).contiguous()),
-                rho.double(), f0s.double(), freq0_global,
+                    Jall.reshape(world, state.M, -1, 2)
+                    .contiguous()).cpu(),
+                rho.double().cpu(), f0s.double().cpu(), freq0_global,
                 polytype=args.polytype, Kstart=1,
                 Kfinish=min(args.npoly, world))
             if rank == 0:
