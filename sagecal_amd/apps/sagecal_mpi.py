@@ -9,8 +9,12 @@ Launch:
   torchrun --nnodes 1 --nproc-per-node W --master-addr 127.0.0.1 \\
       sagecal_mpi.py -f mslist.txt -s sky.txt -c cluster.txt -A 10 -P 2
 
-mslist.txt: one NpzMS path per line; rank r takes line r (more MSs than
-ranks: round-robin multiplexing like sagecal_master.cpp:1055).
+-f: a file listing one MS path per line, OR a glob pattern. With as many
+MSs as ranks, rank r takes entry r. With MORE MSs than ranks, each rank
+owns names[rank::world] and rotates through them one ADMM iteration at a
+time (sagecal_master.cpp:1055 Scurrent multiplexing; this path currently
+solves with the core consensus options — -G/-X/-p extras apply to the
+one-band-per-rank mode).
 """
 import argparse
 import os
