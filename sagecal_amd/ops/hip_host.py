@@ -262,16 +262,6 @@ def apply_jones(coh, J, bb, chunk_rows=None, layout=None):
     return out.view(-1, 2, 2)
 
 
-def residual_total(x, cohs, J, chunk_tabs, layout):
-    """x - sum_ci J C J^H; chunk_tabs: [M, nseg*T] int32."""
-    M = cohs.shape[0]
-    Jc = J.to(torch.complex64).reshape(-1, 4).contiguous()
-    out = _ext().apply_jones(
-        _c64(x), cohs.to(torch.complex64).reshape(-1, 4).contiguous(), Jc,
-        layout.pairs, chunk_tabs.reshape(-1).contiguous(), layout.Nbase,
-        layout.T, J.shape[1], layout.nseg, M, 1)
-    return out.view(-1, 2, 2)
-
 
 def lbfgs_cost_grad(x, cohs, J_packed, chunk_off, nchunks, bb, T, Nbase,
                     robust_nu=None, weights=None):

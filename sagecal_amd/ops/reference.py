@@ -251,21 +251,6 @@ def apply_jones(coh, J, bb, chunk_rows=None):
     return Jp @ coh @ Jq.conj().transpose(-1, -2)
 
 
-def model_all_clusters(cohs, J_packed, chunk_off, chunk_map, bb, T, Nbase):
-    """Sum of J_p C J_q^H over all clusters.
-
-    cohs: [M,B,2,2]; J_packed: [Mt, N, 2, 2] complex (all clusters' chunks
-    packed); chunk_off: [M] first chunk row per cluster; chunk_map:
-    chunk index per (cluster, row) — computed from nchunk and T."""
-    B = cohs.shape[1]
-    out = torch.zeros(B, 2, 2, dtype=cohs.dtype, device=cohs.device)
-    M = cohs.shape[0]
-    for ci in range(M):
-        rows = chunk_rows_for(ci, chunk_map, T, Nbase, B, cohs.device)
-        J = J_packed[chunk_off[ci]:chunk_off[ci] + chunk_map[ci]]
-        out += apply_jones(cohs[ci], J, bb, rows)
-    return out
-
 
 def chunk_rows_for(ci, nchunks, T, Nbase, B, device):
     """Chunk index per baseline row for cluster ci given its nchunk count.

@@ -158,3 +158,31 @@ def test_student_t_weights_bounded(seed, nu):
     assert float(w.min()) > 0
     assert float(w.max()) <= (nu + 8.0) / nu + 1e-9
     assert float(w[0]) < float(w[1:].min())
+
+
+@given(jones())
+@settings(max_examples=15, deadline=None)
+def test_matC_inverts_vecR(J):
+    assert torch.equal(R.matC(R.vecR(J)), J)
+
+
+@given(st.integers(0, 2 ** 31 - 1), st.floats(2.0, 30.0))
+@settings(max_examples=10, deadline=None)
+def test_robust_cost_formula(seed, nu):
+    """robust_cost = sum ln(1 + ||e||^2/nu) over baselines."""
+    rng = np.random.default_rng(seed)
+    N, B = 4, 12
+    pairs = [(p, q) for p in range(N) for q in range(p + 1, N)]
+    bb = torch.tensor((pairs * 3)[:B])
+    coh = torch.tensor(rng.standard_normal((B, 2, 2))
+                       + 1j * rng.standard_normal((B, 2, 2)))
+    J = torch.tensor(np.eye(2)[None, None]
+                     + 0.1 * rng.standard_normal((1, N, 2, 2)) + 0j)
+    x = torch.tensor(rng.standard_normal((B, 2, 2))
+                     + 1j * rng.standard_normal((B, 2, 2)))
+    c, r = R.robust_cost(x, coh, J, bb, nu)
+    e2 = (r.abs() ** 2).sum(dim=(-1, -2))
+    want = float(torch.log1p(e2 / nu).sum())
+    assert abs(float(c) - want) < 1e-10
+    V = R.apply_jones(coh, J, bb)
+    assert torch.allclose(r, x - V)

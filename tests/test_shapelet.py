@@ -165,3 +165,18 @@ def test_diffuse_coherencies_identity_gain():
         * torch.eye(2, dtype=torch.complex128)
     err = float((got - want).abs().max() / want.abs().max())
     assert err < 5e-3, err
+
+
+def test_decompose_image_roundtrip():
+    """decompose_image recovers modes whose image rendering matches the
+    input fluxes (buildsky shapelet fitting role)."""
+    rng = np.random.default_rng(9)
+    n0, beta = 3, 1.5e-3
+    true = rng.standard_normal(n0 * n0)
+    ext = 6 * beta / (2 * np.pi) * 4
+    l = np.linspace(-ext, ext, 24)
+    ll, mm = np.meshgrid(l, l, indexing='ij')
+    img = shapelet.image_basis(ll.ravel(), mm.ravel(), n0, beta).numpy() \
+        @ true
+    got = shapelet.decompose_image(ll.ravel(), mm.ravel(), img, n0, beta)
+    np.testing.assert_allclose(got, true, atol=1e-8)

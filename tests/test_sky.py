@@ -146,3 +146,14 @@ def test_ncp_taper_properties():
     v = torch.zeros(6)
     xw, wts = taper.whiten_data(x, u, v, 150e6)
     assert torch.allclose(xw[:, 0, 0].real, w.double())
+
+
+def test_xyz_llh_roundtrip():
+    """ITRF xyz -> lon/lat/height (transforms.c xyz2llh) sanity: a point
+    on the equatorial WGS84 radius returns lat~0, h~0."""
+    from sagecal_amd import coords
+    a = 6378137.0
+    lon, lat, h = coords.xyz_to_llh(a, 0.0, 0.0)
+    assert abs(lon) < 1e-9 and abs(lat) < 1e-6 and abs(h) < 1e-3
+    lon2, lat2, h2 = coords.xyz_to_llh(0.0, a, 1000.0 * 0 + a * 0)
+    assert abs(lon2 - np.pi / 2) < 1e-9
