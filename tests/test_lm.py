@@ -132,7 +132,8 @@ def test_lbfgs_grad_vs_autograd():
 
         c2 = closure(vr)
         g2, = torch.autograd.grad(c2, vr)
-        assert float(cost) == pytest.approx(float(c2), rel=1e-12)
+        assert float(cost) == pytest.approx(float(c2.detach()),
+                                            rel=1e-12)
         torch.testing.assert_close(grad, g2.reshape(-1), rtol=1e-9,
                                    atol=1e-9)
 
