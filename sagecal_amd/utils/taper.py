@@ -1,7 +1,7 @@
 """uv-distance data tapers.
 
 Re-implements /root/reference/src/lib/Dirac/updatenu.c whiten_data
-(Dirac.h:841) and ncp_weight (updatenu.c:310-318): down-weight short
+(Dirac.h:841) and ncp_weight (updatenu.c:343-349): down-weight short
 baselines with 1/(1 + 1.8 exp(-0.05 |uv|_lambda)) (no effect beyond
 ~400 lambda), used to suppress the north-celestial-pole / large-scale
 contamination before calibration.
