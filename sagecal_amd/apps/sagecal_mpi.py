@@ -231,8 +231,8 @@ def main(argv=None):
             # global Z solution file (the master's write,
             # sagecal_master.cpp:1165): Z is replicated on every rank
             zwriter = solutions.GlobalZWriter(
-                f"{args.solfile}.Z", freq0_global, ms.fdelta,
-                ms.tilesz * ms.tdelta / 60.0, ms.N, state.M, adm.Npoly)
+                f"{args.solfile}.Z", freq0_global, ms.N, state.M,
+                state.Mt, adm.Npoly)
     bb = ms.bb_tensor(device=device)
     for ti, tile in enumerate(ms.tiles()):
         if ti < args.nskip:

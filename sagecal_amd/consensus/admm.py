@@ -54,9 +54,12 @@ class ConsensusADMM:
         self.alpha = federated_alpha
         self.fratio = torch.ones(world)      # per-band flag-ratio weights
         self._update_bii()
-        # consensus state (per rank): Y dual, Z poly coefficients
+        # consensus state (per rank): Y dual, Z poly coefficients.
+        # Z is indexed by EFFECTIVE cluster (Mt, hybrid chunks expanded),
+        # matching the reference master where iodata.M = worker Mt and
+        # the Z file has Mt columns (sagecal_master.cpp:250,517,1168).
         self.Y = torch.zeros(Mt, N, 2, 2, dtype=self.cdtype, device=dev)
-        self.Z = torch.zeros(M, Npoly, N, 2, 2, dtype=self.cdtype,
+        self.Z = torch.zeros(Mt, Npoly, N, 2, 2, dtype=self.cdtype,
                              device=dev)
         self.Zold = None
         self.Yhat_prev = None
@@ -72,15 +75,33 @@ class ConsensusADMM:
         if spatial is not None and centroids is not None:
             from . import fista as fista_mod
             lam, mu_l1, order, fiters, cadence = spatial
-            beta = float(max(np.max(np.abs(centroids[0])),
-                             np.max(np.abs(centroids[1])), 1e-3))
-            self._Phi = fista_mod.spatial_basis(centroids[0], centroids[1],
-                                                order, beta)
+            # centroids come per user cluster; the spatial basis rows are
+            # per EFFECTIVE cluster (ll/mm are length iodata.M = Mt in
+            # sagecal_master.cpp:523-529) — chunk-expand
+            nch = state.nchunks
+            ll = np.concatenate([[centroids[0][ci]] * int(nch[ci])
+                                 for ci in range(M)])
+            mm = np.concatenate([[centroids[1][ci]] * int(nch[ci])
+                                 for ci in range(M)])
+            self.centroids = (ll, mm)
+            beta = float(max(np.max(np.abs(ll)), np.max(np.abs(mm)), 1e-3))
+            self._Phi = fista_mod.spatial_basis(ll, mm, order, beta)
+
+    def _rho_eff(self):
+        """rho expanded to effective clusters [Mt] (double, CPU) — the
+        master's arho expansion via chunkvec (sagecal_master.cpp:598)."""
+        st = self.state
+        if st.Mt == st.M:
+            return self.rho
+        return torch.cat([
+            torch.full((int(st.nchunks[ci]),), float(self.rho[ci]),
+                       dtype=torch.float64) for ci in range(st.M)])
 
     def _update_bii(self):
         # per-band effective rho = rho * fratio_f (sagecal_master.cpp:720:
-        # bands with more flagged data weigh less in the Z fit)
-        rho_mf = self.rho[:, None] * self.fratio[None, :].double()
+        # bands with more flagged data weigh less in the Z fit);
+        # Bii per EFFECTIVE cluster [Mt, Npoly, Npoly]
+        rho_mf = self._rho_eff()[:, None] * self.fratio[None, :].double()
         self.Bii = poly.find_prod_inverse(self.B, rho_mf, self.alpha)
 
     def set_fratio(self, unflagged_fraction):
@@ -95,19 +116,6 @@ class ConsensusADMM:
             fr[:] = float(unflagged_fraction)
         self.fratio = fr.cpu().clamp_min(1e-3)
         self._update_bii()
-
-    def _chunk_expand(self, Zj):
-        """B_f Z -> per-chunk [Mt, N, 2, 2] (each chunk of cluster ci gets
-        the same consensus target)."""
-        st = self.state
-        if st.Mt == st.M:
-            return Zj.to(self.cdtype)
-        out = torch.empty(st.Mt, st.N, 2, 2, dtype=self.cdtype,
-                          device=self.dev)
-        for ci in range(st.M):
-            o = st.chunk_off[ci]
-            out[o:o + st.nchunks[ci]] = Zj[ci]
-        return out
 
     def _rho_chunk(self):
         st = self.state
@@ -129,18 +137,11 @@ class ConsensusADMM:
         per cluster (collapsed TAG_YDATA + update_global_z_multi +
         TAG_CONSENSUS, sagecal_master.cpp:813-877)."""
         st = self.state
-        M, N, P = st.M, st.N, self.Npoly
-        if st.Mt == M:           # common case: one chunk per cluster
-            Jm = st.J
-            Ym = self.Y
-        else:
-            Jm = torch.stack([st.J[st.chunk_off[ci]:st.chunk_off[ci]
-                                   + st.nchunks[ci]].mean(dim=0)
-                              for ci in range(M)])
-            Ym = torch.stack([self.Y[st.chunk_off[ci]:st.chunk_off[ci]
-                                     + st.nchunks[ci]].mean(dim=0)
-                              for ci in range(M)])
-        rho_m = (self.rho * float(self.fratio[self.rank])).to(
+        # per effective cluster [Mt]: no chunk-averaging — each hybrid
+        # chunk is its own consensus variable, like the reference master
+        Jm = st.J
+        Ym = self.Y
+        rho_m = (self._rho_eff() * float(self.fratio[self.rank])).to(
             device=self.dev).to(Jm.real.dtype)
         if first:
             # admm==0: Y already holds rho * (gauge-unified J); the Z fit
@@ -160,9 +161,9 @@ class ConsensusADMM:
         return self.Z
 
     def bz(self):
-        """This band's consensus target B_f Z as per-chunk Jones."""
-        Zj = poly.eval_poly_jones(self.Z, self.Bf)
-        return self._chunk_expand(Zj)
+        """This band's consensus target B_f Z, per effective cluster
+        [Mt, N, 2, 2]."""
+        return poly.eval_poly_jones(self.Z, self.Bf).to(self.cdtype)
 
     def y_update(self, BZ, first=False):
         """Y <- Y + rho (J - BZ) (sagecal_slave.cpp:870-888). At admm==0
@@ -189,9 +190,11 @@ class ConsensusADMM:
             dY = torch.zeros(M, 8 * st.N, dtype=torch.float64)
             dJ = torch.zeros(M, 8 * st.N, dtype=torch.float64)
             for ci in range(M):
-                o = st.chunk_off[ci]
-                dy = (Yhat[o] - self.Yhat_prev[o]).cpu()
-                dj = (st.J[o] - self.J_prev[o]).cpu()
+                o, nc = st.chunk_off[ci], int(st.nchunks[ci])
+                dy = (Yhat[o:o + nc] - self.Yhat_prev[o:o + nc]
+                      ).mean(dim=0).cpu()
+                dj = (st.J[o:o + nc] - self.J_prev[o:o + nc]
+                      ).mean(dim=0).cpu()
                 dY[ci] = torch.view_as_real(dy).reshape(-1).double()
                 dJ[ci] = torch.view_as_real(dj).reshape(-1).double()
             new_rho = poly.update_rho_bb(self.rho, self.rho_upper, dY, dJ)
@@ -290,7 +293,7 @@ class ConsensusADMM:
         Z and form the smoothed constraint Zspat + multiplier update
         (sagecal_master.cpp:887-985)."""
         from . import fista as fista_mod
-        M = self.state.M
+        M = self.state.Mt            # spatial rows per effective cluster
         Zb = torch.view_as_real(self.Z).reshape(M, -1)
         Zb = torch.complex(Zb[:, 0::2], Zb[:, 1::2])   # [M, K]
         Zsp = fista_mod.update_spatialreg_fista(
@@ -316,7 +319,7 @@ class ConsensusADMM:
         if self.Zspat is None or self._Phi is None:
             return None
         from . import fista as fista_mod
-        M = self.state.M
+        M = self.state.Mt
         lam, mu_l1, order, fiters, cadence = self.spatial
         Zb = torch.view_as_real(self.Z).reshape(M, -1)
         Zb = torch.complex(Zb[:, 0::2], Zb[:, 1::2])
@@ -360,7 +363,7 @@ class ConsensusADMM:
 
     def global_solution(self):
         """J = B_f Z (use_global_solution path, sagecal_master:1064)."""
-        return self._chunk_expand(poly.eval_poly_jones(self.Z, self.Bf))
+        return self.bz()
 
 
 class MultiplexedADMM:
@@ -393,55 +396,41 @@ class MultiplexedADMM:
         if rho is None:
             rho = torch.full((self.M,), 5.0)
         self.rho = rho.clone().double()
-        rho_mf = self.rho[:, None].expand(-1, self.F)
+        # Z/Bii per EFFECTIVE cluster (reference master iodata.M = Mt)
+        rho_mf = self._rho_eff()[:, None].expand(-1, self.F)
         self.Bii = poly.find_prod_inverse(self.B, rho_mf)
         for b in self.bands:
             b['Y'] = torch.zeros(self.Mt, self.N, 2, 2,
                                  dtype=self.cdtype, device=self.dev)
-        self.Z = torch.zeros(self.M, self.Npoly, self.N, 2, 2,
+        self.Z = torch.zeros(self.Mt, self.Npoly, self.N, 2, 2,
                              dtype=self.cdtype, device=self.dev)
         self.cur = 0
 
-    def _cluster_mean(self, T):
+    def _rho_eff(self):
         st = self.bands[0]['state']
         if st.Mt == st.M:
-            return T
-        return torch.stack([T[st.chunk_off[ci]:st.chunk_off[ci]
-                              + st.nchunks[ci]].mean(dim=0)
-                            for ci in range(st.M)])
-
-    def _chunk_expand(self, Zj, st):
-        if st.Mt == st.M:
-            return Zj.to(self.cdtype)
-        out = torch.empty(st.Mt, self.N, 2, 2, dtype=self.cdtype,
-                          device=self.dev)
-        for ci in range(st.M):
-            o = st.chunk_off[ci]
-            out[o:o + st.nchunks[ci]] = Zj[ci]
-        return out
+            return self.rho
+        return torch.cat([
+            torch.full((int(st.nchunks[ci]),), float(self.rho[ci]),
+                       dtype=torch.float64) for ci in range(st.M)])
 
     def _rho_chunk(self, st):
-        if st.Mt == st.M:
-            return self.rho.to(self.dev)
-        return torch.cat([
-            torch.full((st.nchunks[ci],), float(self.rho[ci]))
-            for ci in range(st.M)]).to(self.dev)
+        return self._rho_eff().to(self.dev)
 
     def bz(self, bi, st):
-        Zj = poly.eval_poly_jones(self.Z, self.B[bi])
-        return self._chunk_expand(Zj, st)
+        return poly.eval_poly_jones(self.Z, self.B[bi]).to(self.cdtype)
 
     def z_update(self):
         """Allreduce of sum over ALL owned bands of B_b (x) (Y_b +
         rho J_b) — stale contributions for the bands not solved this
         iteration, like the master's persistent Y (sagecal_master.cpp)."""
-        acc = torch.zeros(self.M, self.Npoly, self.N, 2, 2,
+        acc = torch.zeros(self.Mt, self.Npoly, self.N, 2, 2,
                           dtype=self.cdtype, device=self.dev)
-        rho_m = self.rho.to(self.dev).to(torch.float64)
+        rho_m = self._rho_eff().to(self.dev).to(torch.float64)
         for bi, b in zip(self.ids, self.bands):
             st = b['state']
-            Jm = self._cluster_mean(st.J)
-            Ym = self._cluster_mean(b['Y'])
+            Jm = st.J
+            Ym = b['Y']
             contrib = Ym + rho_m[:, None, None, None].to(self.cdtype) * Jm
             Bb = self.B[bi].to(torch.float64)
             acc = acc + (Bb[None, :, None, None, None].to(self.cdtype)

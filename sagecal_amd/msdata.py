@@ -442,13 +442,36 @@ class CasaMS:
         arr = xres.permute(1, 0, 2, 3).cpu().numpy().astype(np.complex64)
         self._out.setdefault(col, []).append((self._rows(ti), arr))
 
+    def _ensure_column(self, col):
+        """Add a missing output column shaped like the data column.
+
+        The reference errors out on a missing output column
+        (src/MS/data.cpp:1404 ArrayColumn on Data::OutField); we go one
+        step friendlier and create it (addImagingColumns analog), but we
+        NEVER silently redirect writes into the input data column.
+        """
+        if col in self.tab.colnames():
+            return
+        try:
+            desc = self.tab.getcoldesc(self.data_col)
+            desc['comment'] = f'added by sagecal_amd ({col})'
+            self.tab.addcols(self._ct.maketabdesc(
+                self._ct.makecoldesc(col, desc)))
+            # initialize cells (addImagingColumns convention: copy DATA)
+            self.tab.putcol(col, np.array(self.tab.getcol(self.data_col)))
+        except Exception as e:
+            raise RuntimeError(
+                f"output column {col!r} does not exist in {self.path} and "
+                f"could not be created ({e}); create it first (e.g. "
+                f"casacore addImagingColumns) or pass -O DATA to "
+                f"explicitly overwrite the input column") from e
+
     def save(self, path=None):
         for col, chunks in self._out.items():
+            self._ensure_column(col)
             for rows, arr in chunks:
-                full = np.asarray(self.tab.getcol(
-                    col if col in self.tab.colnames() else self.data_col))
+                full = np.asarray(self.tab.getcol(col))
                 full[rows] = arr.reshape(arr.shape[0], self.Nchan, 4)
-                self.tab.putcol(col if col in self.tab.colnames()
-                                else self.data_col, full)
+                self.tab.putcol(col, full)
         self._out.clear()
         self.tab.flush()

@@ -102,6 +102,43 @@ def read_solutions(path):
     return header, tiles
 
 
+def read_global_z(path):
+    """Parse a global-Z consensus solution file (GlobalZWriter format =
+    sagecal_master.cpp:513-517,1165-1174). Returns (header dict, list of
+    tiles; each tile [Mt, Npoly, N, 2, 2] complex, columns mapped from
+    file reverse order back to forward effective-cluster order)."""
+    header = None
+    rows = []
+    tiles = []
+    with open(path) as f:
+        for line in f:
+            line = line.strip()
+            if not line or line.startswith('#'):
+                continue
+            toks = line.split()
+            if header is None:
+                header = {
+                    'freq_mhz': float(toks[0]), 'Npoly': int(toks[1]),
+                    'N': int(toks[2]), 'M': int(toks[3]),
+                    'Mt': int(toks[4])}
+                continue
+            rows.append([float(t) for t in toks[1:]])
+            if len(rows) == 8 * header['N'] * header['Npoly']:
+                arr = np.asarray(rows)      # [8N*Npoly, Mt] reverse order
+                rows = []
+                Mt = arr.shape[1]
+                N, P = header['N'], header['Npoly']
+                Z = torch.empty(Mt, P, N, 2, 2, dtype=torch.complex128)
+                for k in range(Mt):
+                    ci = Mt - 1 - k         # undo reverse cluster order
+                    for p in range(P):
+                        Z[ci, p] = ref_vec_to_jones(
+                            torch.tensor(arr[p * 8 * N:(p + 1) * 8 * N, k]),
+                            N)
+                tiles.append(Z)
+    return header, tiles
+
+
 def reorder_read_tile(tile_J, nchunks):
     """Columns in the file are reverse-cluster-order; map a read tile
     [Mt, N, 2, 2] back to forward cluster order given per-cluster chunk
@@ -122,23 +159,24 @@ def reorder_read_tile(tile_J, nchunks):
 
 class GlobalZWriter:
     """Global consensus-polynomial solution file (the MPI master's Z
-    write, sagecal_master.cpp:1165-1174): same layout as the J solution
-    file but with Npoly times more rows — row index p in [0, 8N*Npoly),
-    one column per cluster in REVERSE order."""
+    write, sagecal_master.cpp:513-517 header + :1165-1174 body): same
+    layout as the J solution file but with Npoly times more rows — row
+    index p in [0, 8N*Npoly), one column per EFFECTIVE cluster (Mt,
+    hybrid chunks expanded — iodata.M on the master) in REVERSE order.
+    Header: freq0(MHz) Npoly N Mo Mt."""
 
-    def __init__(self, path, freq0, bandwidth, tile_minutes, N, M, Npoly):
+    def __init__(self, path, freq0, N, Mo, Mt, Npoly):
         self.f = open(path, 'w')
-        self.N, self.M, self.Npoly = N, M, Npoly
-        self.f.write("# solution file created by SAGECal (global Z)\n")
-        self.f.write("# freq(MHz) bandwidth(MHz) time_interval(min) "
-                     "stations clusters effective_clusters\n")
-        self.f.write("%lf %lf %lf %d %d %d\n" % (
-            freq0 * 1e-6, bandwidth * 1e-6, tile_minutes, N, M,
-            M * Npoly))
+        self.N, self.M, self.Npoly = N, Mt, Npoly
+        self.f.write("# solution file (Z) created by SAGECal\n")
+        self.f.write("# reference_freq(MHz) polynomial_order stations "
+                     "clusters effective_clusters\n")
+        self.f.write("%lf %d %d %d %d\n" % (
+            freq0 * 1e-6, Npoly, N, Mo, Mt))
 
     def write_tile(self, Z):
-        """Z: [M, Npoly, N, 2, 2] complex (consensus.admm.ConsensusADMM.Z
-        layout)."""
+        """Z: [Mt, Npoly, N, 2, 2] complex (consensus.admm.ConsensusADMM.Z
+        layout, per effective cluster)."""
         M, Npoly, N = Z.shape[:3]
         cols = []
         for ci in range(M - 1, -1, -1):
@@ -147,7 +185,7 @@ class GlobalZWriter:
                 col.append(jones_to_ref_vec(
                     Z[ci, p].cpu().to(torch.complex128)).numpy())
             cols.append(np.concatenate(col))
-        cols = np.stack(cols, axis=1)          # [8N*Npoly, M]
+        cols = np.stack(cols, axis=1)          # [8N*Npoly, Mt]
         for cj in range(cols.shape[0]):
             self.f.write("%d " % cj)
             self.f.write(''.join(" %e" % val for val in cols[cj]))

@@ -201,25 +201,54 @@ def _per_chunk_cost(x, coh, J, bb, prob):
 
 
 def os_lm_solve(prob, J0, maxiter=30, nsubsets=4, seed=0, **kw):
-    """Ordered-subsets LM (oslmfit.c / oslevmar_der_single_*): run a few LM
-    iterations on each random time-slot subset, then finish on full data.
+    """Ordered-subsets LM (oslmfit.c / oslevmar_der_single_*): a few LM
+    iterations on each subset — visited in random ORDER, like the
+    reference's random_permutation over subsets (lmfit.c:1084) — then a
+    final pass on the full data.
 
-    Subsets are contiguous time ranges (whole timeslots) so chunk mapping
-    stays valid."""
+    Subsets are contiguous WHOLE-TIMESLOT ranges, identical across the
+    seg axis, so the seg*(T*Nbase)+t*Nbase+b row structure survives and
+    the GPU kernels stay on the structured (BaselineLayout + hipGraph)
+    path instead of falling back to eager."""
     B = prob.x.shape[0]
     dev = prob.x.device
     g = torch.Generator(device='cpu').manual_seed(seed)
-    perm = torch.randperm(B, generator=g).to(dev)
     iters_per = max(2, maxiter // (nsubsets + 1))
+    order = torch.randperm(nsubsets, generator=g).tolist()
     J = J0
-    for si in range(nsubsets):
-        sel = perm[si * B // nsubsets:(si + 1) * B // nsubsets]
-        sub = LMProblem(
-            prob.x[sel], prob.coh[sel], prob.bb[sel], prob.N, prob.nchunk,
-            prob.chunk_rows[sel] if prob.chunk_rows is not None else None,
-            prob.weights[sel] if prob.weights is not None else None)
-        # random-row subsets break the (seg,t,pair) structure: CPU path
-        J, _ = lm_solve(sub, J, maxiter=iters_per, **kw)
+    lay = prob.layout
+    if lay is not None and lay.T >= nsubsets:
+        from ..ops.hip_host import BaselineLayout
+        T, Nbase, nseg = lay.T, lay.Nbase, lay.nseg
+        edges = [s * T // nsubsets for s in range(nsubsets + 1)]
+        base = torch.arange(nseg, device=dev) * (T * Nbase)
+        for s in order:
+            t0, t1 = edges[s], edges[s + 1]
+            offs = torch.arange(t0 * Nbase, t1 * Nbase, device=dev)
+            sel = (base[:, None] + offs[None, :]).reshape(-1)
+            sub_lay = BaselineLayout(prob.bb[sel], Nbase, t1 - t0, nseg,
+                                     prob.N, dev)
+            sub = LMProblem(
+                prob.x[sel], prob.coh[sel], prob.bb[sel], prob.N,
+                prob.nchunk,
+                prob.chunk_rows[sel] if prob.chunk_rows is not None
+                else None,
+                prob.weights[sel] if prob.weights is not None else None,
+                layout=sub_lay)
+            J, _ = lm_solve(sub, J, maxiter=iters_per, **kw)
+    else:
+        # CPU / tiny-T path: contiguous row blocks (rows are time-major,
+        # so blocks are still contiguous time ranges)
+        edges = [s * B // nsubsets for s in range(nsubsets + 1)]
+        for s in order:
+            sel = torch.arange(edges[s], edges[s + 1], device=dev)
+            sub = LMProblem(
+                prob.x[sel], prob.coh[sel], prob.bb[sel], prob.N,
+                prob.nchunk,
+                prob.chunk_rows[sel] if prob.chunk_rows is not None
+                else None,
+                prob.weights[sel] if prob.weights is not None else None)
+            J, _ = lm_solve(sub, J, maxiter=iters_per, **kw)
     J, info = lm_solve(prob, J, maxiter=iters_per, **kw)
     return J, info
 

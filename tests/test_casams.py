@@ -28,6 +28,14 @@ class _MockTable:
     def colnames(self):
         return list(self.cols)
 
+    def getcoldesc(self, name):
+        return {'valueType': 'complex', 'ndim': 2}
+
+    def addcols(self, desc):
+        # desc comes from _MockTables.maketabdesc -> {name: coldesc}
+        for name in desc:
+            self.cols[name] = None
+
     def close(self):
         pass
 
@@ -73,6 +81,14 @@ class _MockTables:
                 return t
         return self.main
 
+    @staticmethod
+    def makecoldesc(name, desc):
+        return {name: desc}
+
+    @staticmethod
+    def maketabdesc(coldesc):
+        return coldesc
+
 
 def test_casams_mapping():
     mock = _MockTables(N=4, T=4, F=2)
@@ -107,13 +123,32 @@ def test_casams_write_roundtrip():
     ms = msdata.CasaMS('fake.ms', tilesz=2, tables_mod=mock)
     tile = ms.load_tile(1)
     ms.write_column('residual', 1, tile.xo * 0.5)
+    data_before = mock.main.cols['DATA'].copy()
     ms.save()
     assert mock.main.flushed
-    # no CORRECTED_DATA column in the mock -> falls back to DATA
+    # missing CORRECTED_DATA is CREATED (addImagingColumns analog), the
+    # raw DATA column is never silently overwritten (ADVICE r1 high;
+    # reference errors on a missing output column, data.cpp:1404)
+    assert np.array_equal(mock.main.cols['DATA'], data_before)
     sel = mock.main.cols['ANTENNA1'] != mock.main.cols['ANTENNA2']
-    got = mock.main.cols['DATA'][sel][12:24]
+    got = mock.main.cols['CORRECTED_DATA'][sel][12:24]
     want = tile.xo.permute(1, 0, 2, 3).numpy().reshape(12, 2, 4) * 0.5
     assert np.allclose(got, want, atol=1e-6)
+    # untouched rows were initialized from DATA
+    got0 = mock.main.cols['CORRECTED_DATA'][sel][:12]
+    assert np.allclose(got0, data_before[sel][:12])
+
+
+def test_casams_save_raises_when_column_uncreatable():
+    mock = _MockTables(N=4, T=4, F=2)
+    mock.main.addcols = None  # table refuses addcols
+    ms = msdata.CasaMS('fake.ms', tilesz=2, tables_mod=mock)
+    tile = ms.load_tile(0)
+    ms.write_column('residual', 0, tile.xo * 0.5)
+    with pytest.raises(RuntimeError, match='could not be created'):
+        ms.save()
+    # DATA untouched even on failure
+    assert 'CORRECTED_DATA' not in mock.main.cols
 
 
 def test_casams_gated_without_casacore():

@@ -151,14 +151,16 @@ def test_sagecal_mpi_two_bands(tmp_path):
     from sagecal_amd import solutions
     hdr, tiles = solutions.read_solutions(str(tmp_path / 'sol.txt.rank0'))
     assert tiles and tiles[0].shape[-2:] == (2, 2)
-    hdrz, ztiles = solutions.read_solutions(str(tmp_path / 'sol.txt.Z'))
+    hdrz, ztiles = solutions.read_global_z(str(tmp_path / 'sol.txt.Z'))
     assert ztiles, "global Z solution file missing/empty"
+    # reference Z header (sagecal_master.cpp:517): freq Npoly N Mo Mt
+    assert hdrz['Npoly'] == 2 and hdrz['N'] == hdr['N']
+    assert hdrz['Mt'] >= hdrz['M']
     spf = tmp_path / 'spatial_sol.txt'
     assert spf.exists() and len(spf.read_text().splitlines()) > 2
     assert (tmp_path / 'sol.txt.spatial.ppm').exists()
-    # Npoly=2: the Z file carries twice the J file's values per tile
-    # (read_solutions chunks rows in 8N groups, so Npoly groups come back
-    # as Npoly entries)
+    # Npoly=2, Z per effective cluster: the Z file carries Npoly times
+    # the J file's values per tile
     assert sum(t.numel() for t in ztiles) == \
         2 * sum(t.numel() for t in tiles)
 
