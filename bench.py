@@ -28,6 +28,14 @@ def build_problem(args, device, dtype, rank=0, world=1):
 
     srcs, clist = sky.make_synthetic_sky(
         M=args.dirs, nsrc_per_cluster=args.srcs, seed=17)
+    # Batch P solution intervals per step via the hybrid time-chunk
+    # machinery (reference -t tilesz + cluster chunk column,
+    # lmfit.c:893-967): every cluster solves P independent Jones
+    # solutions, one per `tilesz`-slot interval — P x em_group problems
+    # per batched LM solve, sized to fill 256 CUs.
+    P = max(1, getattr(args, 'intervals', 1))
+    if P > 1:
+        clist = [(cid, P * nchunk, names) for cid, nchunk, names in clist]
     nshap = getattr(args, 'shapelet_dirs', 0)
     if nshap:
         # make the first sources of the first nshap clusters shapelets
@@ -47,13 +55,19 @@ def build_problem(args, device, dtype, rank=0, world=1):
     pack = SourcePack(clusters)
     # per-rank frequency sub-band (sagecal-mpi: one MS band per worker)
     freq0 = args.freq0 + rank * args.bandwidth
+    # synthetic truth is generated in fp64; for large arrays generate on
+    # the GPU (fp64 on device) — CPU generation at 512 stations takes
+    # minutes and starves the bench setup
+    gen_dev = device if (device != 'cpu'
+                         and args.stations * args.tilesz * P >= 2**21)         else 'cpu'
+    if gen_dev != 'cpu':
+        pack.to(gen_dev)
     ms = msdata.SyntheticMS(
-        N=args.stations, tilesz=args.tilesz, Ntime=args.tilesz,
+        N=args.stations, tilesz=args.tilesz * P, Ntime=args.tilesz * P,
         Nchan=args.chan, freq0=freq0, bandwidth=args.bandwidth,
         tdelta=10.0, pack=pack, seed=101 + rank,
         noise_sigma=5e-3, robust_noise=4.0,
-        device='cpu', dtype=torch.float64)
-    # generate the tile on CPU in fp64 (data provenance), move to device
+        device=gen_dev, dtype=torch.float64)
     tile = ms.load_tile(0)
     if device != 'cpu':
         cdt = torch.complex64 if dtype == torch.float32 else torch.complex128
@@ -135,14 +149,20 @@ def main():
     ap.add_argument('--dirs', type=int, default=10)
     ap.add_argument('--srcs', type=int, default=5)
     ap.add_argument('--tilesz', type=int, default=60)
+    ap.add_argument('--intervals', type=int, default=16,
+                    help='solution intervals batched per step (hybrid '
+                         'time-chunks; each interval is tilesz slots)')
     ap.add_argument('--chan', type=int, default=8)
     ap.add_argument('--freq0', type=float, default=150e6)
     ap.add_argument('--bandwidth', type=float, default=180e3)
-    ap.add_argument('--emiter', type=int, default=4)
+    ap.add_argument('--emiter', type=int, default=2)
     ap.add_argument('--maxiter', type=int, default=6)
     ap.add_argument('--robust-outer', type=int, default=1)
     ap.add_argument('--joint', type=int, default=0)
-    ap.add_argument('--em-group', type=int, default=5)
+    ap.add_argument('--lbfgs-polish', type=int, default=10,
+                    help='final joint LBFGS iterations per sagefit '
+                         '(lmfit.c:1019 polish; 0 = off)')
+    ap.add_argument('--em-group', type=int, default=10)
     ap.add_argument('--npoly', type=int, default=2)
     ap.add_argument('--solver', choices=['lm', 'rtr'], default='lm',
                     help='rtr: Riemannian trust-region (SKA config 5)')
@@ -241,7 +261,7 @@ def main():
         solver_mode=(SM_RTR_OSRLM_RLBFGS if args.solver == 'rtr'
                      else SM_RLM_RLBFGS),
         robust_outer=args.robust_outer, em_group=args.em_group,
-        joint_iters=args.joint)
+        joint_iters=args.joint, lbfgs_iters=args.lbfgs_polish)
 
     def sync():
         if use_gpu:
@@ -266,7 +286,7 @@ def main():
         elapsed = float(te)
 
     Nbase = args.stations * (args.stations - 1) // 2
-    vis_per_step = Nbase * args.tilesz * args.chan
+    vis_per_step = Nbase * args.tilesz * args.intervals * args.chan
     ms_per_step = elapsed / args.steps * 1e3
     value = vis_per_step * args.steps * world / elapsed
 
@@ -289,6 +309,7 @@ def main():
                          f'robust LM (SAGE)',
                 'stations': args.stations, 'directions': args.dirs,
                 'tilesz': args.tilesz, 'channels': args.chan,
+                'intervals_per_step': args.intervals,
                 'global_batch': vis_per_step,
                 'seq_len': args.tilesz,
                 'parallelism': f'consensus-admm freq-band dp{world}',

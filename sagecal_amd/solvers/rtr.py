@@ -172,7 +172,9 @@ def rtr_solve(prob, J0, maxiter=20, rsd_iters=2, delta0=None,
 
     prob: lm.LMProblem (weights -> robust RTR; admm -> RTR-ADMM)."""
     import os as _os
-    if (_os.environ.get('SAGECAL_RTR_GRAPH') == '1'
+    # graph-captured path is the CUDA default (GPU-validated round 2:
+    # test_rtr_graphed_matches_eager); opt out with SAGECAL_RTR_GRAPH=0
+    if (_os.environ.get('SAGECAL_RTR_GRAPH', '1') != '0'
             and _os.environ.get('SAGECAL_NO_GRAPH') != '1'
             and J0.is_cuda and prob.layout is not None):
         return rtr_solve_graphed(prob, J0, maxiter=maxiter,
@@ -270,8 +272,9 @@ def nsd_solve(prob, J0, maxiter=40, lr=None):
 # ---------------------------------------------------------------------------
 # hipGraph capture of the RTR loop (rtr_solve body with the host polls
 # removed): fixed RSD warmup + fixed outer iterations + fixed-length tCG.
-# Gated behind SAGECAL_RTR_GRAPH=1 (off by default until measured on
-# hardware); the poll-free body itself is validated on CPU in
+# Default ON for CUDA since round 2 (GPU-validated:
+# tests/test_gpu.py::test_rtr_graphed_matches_eager); disable with
+# SAGECAL_RTR_GRAPH=0. The poll-free body is also validated on CPU in
 # tests/test_rtr.py::test_rtr_body_matches_eager.
 # ---------------------------------------------------------------------------
 

@@ -282,7 +282,7 @@ def test_chol_mw_matches_single_kernel():
     assert float(err) < 1e-4, f"mw vs single rel err {float(err)}"
 
 
-@pytest.mark.skipif(os.environ.get('SAGECAL_RTR_GRAPH') != '1',
+@pytest.mark.skipif(os.environ.get('SAGECAL_RTR_GRAPH', '1') == '0',
                     reason='RTR graph path is opt-in '
                            '(SAGECAL_RTR_GRAPH=1); round-2 validation')
 def test_rtr_graphed_matches_eager():

@@ -22,6 +22,7 @@ def main():
     ap.add_argument('--dirs', type=int, default=10)
     ap.add_argument('--srcs', type=int, default=5)
     ap.add_argument('--tilesz', type=int, default=60)
+    ap.add_argument('--intervals', type=int, default=8)
     ap.add_argument('--chan', type=int, default=8)
     ap.add_argument('--emiter', type=int, default=3)
     ap.add_argument('--maxiter', type=int, default=8)
@@ -39,7 +40,7 @@ def main():
     a = A()
     a.__dict__.update(stations=args.stations, dirs=args.dirs, srcs=args.srcs,
                       tilesz=args.tilesz, chan=args.chan, freq0=150e6,
-                      bandwidth=180e3)
+                      bandwidth=180e3, intervals=args.intervals)
     dev = 'cuda:0'
     pack, ms, tile, bb = bench.build_problem(a, dev, torch.float32)
     state = sage.CalState(pack, args.stations, device=dev,
