@@ -11,9 +11,11 @@ Differences from the reference, by design:
     ITRF tables; the LOFAR_ANTENNA_FIELD metadata (element/tile offsets,
     pointing) maps onto ArrayConfig directly;
   - the LOFAR LBA/HBA/ALO element coefficient tables are DATA the
-    reference ships as headers (elementcoeff.h); we implement the same
-    evaluation machinery with a documented .npz coefficient format plus a
-    synthetic dipole-pattern generator, so real tables can be dropped in.
+    reference ships as headers (elementcoeff.h); they are ported into
+    sagecal_amd/data/lofar_element_*.npz (tools/port_elementcoeff.py)
+    and evaluated by LofarElementCoeffs in the reference's own
+    Laguerre-Gaussian basis; a synthetic generator + documented .npz
+    format (ElementCoeffs) remain for custom dipole models.
 
 All evaluation is vectorized torch over (source, time, station).
 """
@@ -189,7 +191,12 @@ def n_modes(n0):
 def element_beam(coeffs, az, el, freq):
     """E-Jones per direction from the coefficient tables
     (eval_elementcoeffs, elementbeam.c): rows (theta, phi) pattern of the
-    X and Y dipoles -> 2x2 complex [K, 2, 2]."""
+    X and Y dipoles -> 2x2 complex [K, 2, 2]. Accepts either the
+    synthetic spherical-harmonic ElementCoeffs or the real LOFAR tables
+    (LofarElementCoeffs)."""
+    if isinstance(coeffs, LofarElementCoeffs):
+        zen = np.pi / 2 - np.asarray(el)
+        return coeffs.eval(zen, np.asarray(az), freq)
     theta = np.pi / 2 - np.asarray(el)      # zenith angle
     phi = np.asarray(az)
     ct_x, cp_x, ct_y, cp_y = coeffs.at_freq(freq)
@@ -201,6 +208,118 @@ def element_beam(coeffs, az, el, freq):
     E = np.stack([np.stack([Ext, Exp], -1),
                   np.stack([Eyt, Eyp], -1)], -2)  # [K, 2, 2]
     return torch.from_numpy(E)
+
+
+class LofarElementCoeffs:
+    """Real LOFAR/ALO dipole element patterns in the reference's own
+    Laguerre-Gaussian polar basis (elementbeam.c set_elementcoeffs:40 +
+    eval_elementcoeffs:384; tables ported from elementcoeff.h /
+    elementcoeff_ALO.h by tools/port_elementcoeff.py).
+
+    Basis modes (n, m), n < M, m = -n..n step 2 (Nmodes = M(M+1)/2):
+      f_nm(r, t) = P_nm (pi/4 + r)^|m| L_{(n-|m|)/2}^{|m|}(r^2/beta^2)
+                   exp(-r^2/(2 beta^2)) exp(-j m t)
+      P_nm = sqrt(((n-|m|)/2)! / (pi ((n+|m|)/2)!)) (-1)^{(n-|m|)/2}
+             beta^{-1-|m|}
+    with r the zenith angle and t the dipole-frame azimuth; the X dipole
+    is evaluated at t = az - pi/4, the Y dipole at t = az + pi/4
+    (stationbeam.c:331-345)."""
+
+    def __init__(self, M, beta, freqs_ghz, theta, phi):
+        self.M = int(M)
+        self.beta = float(beta)
+        self.freqs_ghz = np.atleast_1d(freqs_ghz).astype(float)
+        self.theta = np.atleast_2d(theta)     # [Nf, Nmodes] complex
+        self.phi = np.atleast_2d(phi)
+        self.Nmodes = self.M * (self.M + 1) // 2
+        assert self.theta.shape[1] == self.Nmodes
+        # preamble per mode (elementbeam.c:160-175)
+        import math
+        pre = np.empty(self.Nmodes)
+        idx = 0
+        for n in range(self.M):
+            for m in range(-n, n + 1, 2):
+                am = abs(m)
+                v = math.sqrt(math.factorial((n - am) // 2)
+                              / (math.pi * math.factorial((n + am) // 2)))
+                if ((n - am) // 2) % 2:
+                    v = -v
+                pre[idx] = v * self.beta ** (-1.0 - am)
+                idx += 1
+        self.preamble = pre
+        self._nm = [(n, m) for n in range(self.M)
+                    for m in range(-n, n + 1, 2)]
+
+    @staticmethod
+    def load(kind_or_path):
+        """Load 'lba' / 'hba' / 'alo' (packaged tables) or an .npz path."""
+        import os
+        p = str(kind_or_path)
+        if p.lower() in ('lba', 'hba', 'alo'):
+            p = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                             'data', f'lofar_element_{p.lower()}.npz')
+        z = np.load(p)
+        return LofarElementCoeffs(z['M'], z['beta'], z['freqs_ghz'],
+                                  z['theta'], z['phi'])
+
+    def at_freq(self, freq_hz):
+        """(theta, phi) coefficient vectors at a frequency: clamp at the
+        table edges, linear interpolation inside (elementbeam.c:105-142)."""
+        f = freq_hz / 1e9
+        fr = self.freqs_ghz
+        idh = int(np.searchsorted(fr, f, side='left'))
+        if idh >= len(fr):
+            idl = idh = len(fr) - 1
+        elif idh == 0:
+            idl = 0
+        else:
+            idl = idh - 1
+        if idl == idh:
+            return self.theta[idl], self.phi[idl]
+        w1 = (f - fr[idl]) / (fr[idh] - fr[idl])
+        return ((1 - w1) * self.theta[idl] + w1 * self.theta[idh],
+                (1 - w1) * self.phi[idl] + w1 * self.phi[idh])
+
+    def basis(self, r, t):
+        """Mode stack f_nm at zenith angle r, basis azimuth t: [K, Nmodes]
+        complex (vectorized eval_elementcoeffs inner loop)."""
+        r = np.atleast_1d(np.asarray(r, dtype=float))
+        t = np.atleast_1d(np.asarray(t, dtype=float))
+        rb = (r / self.beta) ** 2
+        ex = np.exp(-0.5 * rb)
+        cols = np.empty((r.shape[0], self.Nmodes), dtype=np.complex128)
+        # Laguerre L_p^q(rb) by the reference's recurrence (L_g1)
+        lag = {}
+        maxp = (self.M - 1) // 2 + 1
+        for q in range(self.M):
+            Lp2 = np.ones_like(rb)
+            Lp1 = 1.0 - rb + q
+            lag[(0, q)] = Lp2
+            lag[(1, q)] = Lp1
+            for i in range(2, maxp + 1):
+                p1 = 1.0 / i
+                Lp = (2.0 + p1 * (q - 1.0 - rb)) * Lp1 \
+                    - (1.0 + p1 * (q - 1)) * Lp2
+                lag[(i, q)] = Lp
+                Lp2, Lp1 = Lp1, Lp
+        for idx, (n, m) in enumerate(self._nm):
+            am = abs(m)
+            Lg = lag[((n - am) // 2, am)]
+            rm = (np.pi / 4 + r) ** am
+            pr = rm * Lg * ex * self.preamble[idx]
+            cols[:, idx] = pr * np.exp(-1j * m * t)
+        return cols
+
+    def eval(self, zen, az, freq_hz):
+        """E-Jones rows per direction: [K, 2, 2] complex with
+        row 0 = X dipole (Etheta, Ephi) at t = az - pi/4,
+        row 1 = Y dipole at t = az + pi/4 (stationbeam.c:331-345)."""
+        ct, cp = self.at_freq(freq_hz)
+        bx = self.basis(zen, np.asarray(az) - np.pi / 4)
+        by = self.basis(zen, np.asarray(az) - np.pi / 4 + np.pi / 2)
+        E = np.stack([np.stack([bx @ ct, bx @ cp], -1),
+                      np.stack([by @ ct, by @ cp], -1)], -2)
+        return torch.from_numpy(E)
 
 
 def make_synthetic_element_coeffs(n0=4, freqs=(150e6,), seed=0,
@@ -263,6 +382,8 @@ def predict_coh_withbeam(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0,
         # array origin (kernel_element_beam; DOBEAM_ELEMENT/FULL modes)
         if coeffs is None:
             coeffs = make_synthetic_element_coeffs(freqs=(freq,))
+        elif isinstance(coeffs, str):
+            coeffs = LofarElementCoeffs.load(coeffs)
         gmst = coords.jd_to_gmst(np.asarray(np.atleast_1d(tmjd))
                                  + 2400000.5)
         Es = []
