@@ -36,7 +36,10 @@ class GPUPack:
                   'spec_idx', 'spec_idx1', 'spec_idx2', 'eX', 'eY', 'eP'):
             setattr(self, k, f32(getattr(pack, k)))
         self.f0 = pack.f0.to(device=device, dtype=torch.float64)
-        up = pack.use_proj.to(device=device).bool()
+        # gaussian honors use_projection; disk/ring ALWAYS project
+        # (disk_contrib/ring_contrib predict.c:60-90, oracle-verified)
+        up = pack.use_proj.to(device=device).bool() | \
+            (pack.stype.to(device) >= 2)
         one = torch.ones_like(f32(pack.cxi))
         zero = torch.zeros_like(one)
         self.cxi = torch.where(up, f32(pack.cxi), one).contiguous()

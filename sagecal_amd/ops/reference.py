@@ -119,7 +119,10 @@ def _extended_envelope(pack, sel, ul, vl, wl):
     cphi, sphi = pack.cphi[sel], pack.sphi[sel]
     up_p = ul * cxi - vl * cphi * sxi + wl * sphi * sxi
     vp_p = ul * sxi + vl * cphi * cxi - wl * sphi * cxi
-    use_p = pack.use_proj[sel].to(torch.bool)
+    # gaussian honors use_projection (gaussian_contrib, predict.c:36-44);
+    # disk/ring ALWAYS apply the rotation (disk_contrib/ring_contrib,
+    # predict.c:60-90 have no use_projection branch) — oracle-verified
+    use_p = pack.use_proj[sel].to(torch.bool) | (stype >= 2)
     up = torch.where(use_p, up_p, ul.expand_as(up_p))
     vp = torch.where(use_p, vp_p, vl.expand_as(vp_p))
     eX, eY, eP = pack.eX[sel], pack.eY[sel], pack.eP[sel]

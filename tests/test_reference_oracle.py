@@ -1,0 +1,149 @@
+"""Cross-validation against the REFERENCE implementation compiled from
+/root/reference (SURVEY.md §6 / VERDICT r1 item 6): the reference's own
+lbfgs.c and predict.c are built unmodified (tools/oracle/) and run as a
+correctness oracle. This converts "self-consistent" into "matches the
+reference" for the optimizer core and the coherency predict."""
+import os
+import subprocess
+import shutil
+
+import numpy as np
+import pytest
+import torch
+
+REF = '/root/reference'
+ORACLE_DIR = '/tmp/sagecal_oracle'
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture(scope='module')
+def oracle():
+    if shutil.which('gcc') is None or not os.path.isdir(REF):
+        pytest.skip('gcc or reference tree unavailable')
+    r = subprocess.run(
+        ['bash', os.path.join(REPO, 'tools', 'oracle', 'build_oracle.sh'),
+         ORACLE_DIR], capture_output=True, text=True)
+    if r.returncode != 0:
+        pytest.skip(f'oracle build failed: {r.stderr[-500:]}')
+    return ORACLE_DIR
+
+
+def test_lbfgs_matches_reference_rosenbrock(oracle):
+    """Reference lbfgs_fit (lbfgs.c:933, the test/Dirac/demo.c problem)
+    vs sagecal_amd.solvers.lbfgs.lbfgs_fit on extended Rosenbrock."""
+    out = subprocess.run([os.path.join(oracle, 'oracle_lbfgs'), '8',
+                          '200'], capture_output=True, text=True,
+                         check=True).stdout.split()
+    ref_p = np.array([float(t) for t in out[:-1]])
+    ref_cost = float(out[-1])
+    assert np.allclose(ref_p, 1.0, atol=1e-6)
+    assert ref_cost < 1e-12
+
+    from sagecal_amd.solvers.lbfgs import lbfgs_fit
+
+    def fg(p):
+        x0, x1 = p[0::2], p[1::2]
+        a = x1 - x0 ** 2
+        b = 1.0 - x0
+        f = (100.0 * a ** 2 + b ** 2).sum()
+        g = torch.zeros_like(p)
+        g[0::2] = -400.0 * x0 * a - 2.0 * b
+        g[1::2] = 200.0 * a
+        return float(f), g
+
+    p0 = torch.tensor([-1.2, 1.0] * 4, dtype=torch.float64)
+    p1, _, _ = lbfgs_fit(fg, p0, maxiter=200, m=7)
+    # both implementations must find the same minimum
+    assert torch.allclose(p1, torch.ones(8, dtype=torch.float64),
+                          atol=1e-5)
+    assert np.allclose(p1.numpy(), ref_p, atol=1e-5)
+
+
+def _write_layout(path, N, bb, u, v, w, pack, freq0, fdelta, tdelta,
+                  dec0):
+    Nbase = u.shape[0]
+    M = pack.M
+    with open(path, 'w') as f:
+        f.write(f"{N} {Nbase} {M} {freq0} {fdelta} {tdelta} {dec0}\n")
+        for b in range(Nbase):
+            f.write(f"{int(bb[b, 0])} {int(bb[b, 1])} "
+                    f"{float(u[b]):.17g} {float(v[b]):.17g} "
+                    f"{float(w[b]):.17g}\n")
+        for ci in range(M):
+            s0 = int(pack.cluster_off[ci])
+            s1 = int(pack.cluster_off[ci + 1])
+            f.write(f"{ci} {s1 - s0}\n")
+            for s in range(s0, s1):
+                f.write(f"{int(pack.stype[s])} "
+                        f"{float(pack.ll[s]):.17g} "
+                        f"{float(pack.mm[s]):.17g} "
+                        f"{float(pack.nn1[s]):.17g} "
+                        f"{float(pack.sI[s]):.17g} "
+                        f"{float(pack.sQ[s]):.17g} "
+                        f"{float(pack.sU[s]):.17g} "
+                        f"{float(pack.sV[s]):.17g} "
+                        f"{float(pack.eX[s]):.17g} "
+                        f"{float(pack.eY[s]):.17g} "
+                        f"{float(pack.eP[s]):.17g} "
+                        f"{float(pack.cxi[s]):.17g} "
+                        f"{float(pack.sxi[s]):.17g} "
+                        f"{float(pack.cphi[s]):.17g} "
+                        f"{float(pack.sphi[s]):.17g} "
+                        f"{int(pack.use_proj[s])}\n")
+
+
+def test_predict_matches_reference(oracle, tmp_path):
+    """Reference precalculate_coherencies (predict.c:503) vs
+    ops.reference.predict_coh on the same point+gaussian+disk+ring
+    layout — value-by-value."""
+    from sagecal_amd import sky
+    from sagecal_amd.ops.reference import SourcePack, predict_coh
+
+    rng = np.random.default_rng(11)
+    N, Nbase = 8, 28
+    freq0, fdelta, tdelta, dec0 = 150e6, 180e3, 10.0, np.pi / 4
+    pairs = np.array([(p, q) for p in range(N) for q in range(p + 1, N)])
+    u = rng.standard_normal(Nbase) * 300.0 / 3e8
+    v = rng.standard_normal(Nbase) * 300.0 / 3e8
+    w = rng.standard_normal(Nbase) * 30.0 / 3e8
+
+    srcs = {}
+    clist = []
+    stypes = [0, 1, 2, 3]      # point, gaussian, disk, ring
+    for ci, ty in enumerate(stypes):
+        name = f's{ci}'
+        s = sky.Source(
+            name=name, ra=0.02 * (ci + 1), dec=dec0 + 0.01 * ci,
+            sI=1.0 + 0.3 * ci, sQ=0.05, sU=0.02, sV=0.01,
+            spec_idx=0.0, spec_idx1=0.0, spec_idx2=0.0, f0=freq0,
+            stype=ty, eX=2e-3 if ty else 0.0, eY=1e-3 if ty else 0.0,
+            eP=0.3 if ty else 0.0)
+        srcs[name] = s
+        clist.append((ci, 1, [name]))
+    clusters = sky.build_clusters(srcs, clist, 0.0, dec0, freq0)
+    pack = SourcePack(clusters)
+
+    layout = tmp_path / 'layout.txt'
+    _write_layout(layout, N, pairs, u, v, w, pack, freq0, fdelta, tdelta,
+                  dec0)
+    lines = subprocess.run([os.path.join(oracle, 'oracle_predict'),
+                            str(layout)], capture_output=True, text=True,
+                           check=True).stdout.strip().splitlines()
+    vals = np.array([[float(t) for t in ln.split()] for ln in lines])
+    M = pack.M
+    ref = (vals[:, 0::2] + 1j * vals[:, 1::2]).reshape(Nbase, M, 2, 2)
+    ref = np.transpose(ref, (1, 0, 2, 3))            # [M, B, 2, 2]
+
+    ut = torch.tensor(u)
+    vt = torch.tensor(v)
+    wt = torch.tensor(w)
+    # the reference applies NO time smearing in precalculate_coherencies
+    # (time_smear's only call site, residual.c:434, is commented out, and
+    # the GPU kernels ignore deltat) — our predict implements it as a
+    # deliberate physics improvement; disable it here for the comparison
+    ours = predict_coh(pack, ut, vt, wt, freq0, freq0, fdelta, 0.0,
+                       dec0).numpy()
+    scale = np.abs(ref).max()
+    err = np.abs(ours - ref).max() / scale
+    # 5e-9: torch.special bessel j0/j1 vs libm j0/j1 precision
+    assert err < 5e-9, f"predict mismatch: rel err {err}"

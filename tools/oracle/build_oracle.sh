@@ -1,0 +1,20 @@
+#!/bin/bash
+# Build the reference-CPU oracle binaries (SURVEY.md §6 / VERDICT r1
+# item 6): compiles the reference's lbfgs.c, myblas.c and predict.c
+# unmodified against a minimal local BLAS (miniblas.c) and links two
+# tiny drivers. Usage: bash tools/oracle/build_oracle.sh [outdir]
+set -e
+REF=${REF:-/root/reference}
+HERE=$(cd "$(dirname "$0")" && pwd)
+OUT=${1:-/tmp/sagecal_oracle}
+mkdir -p "$OUT"
+CFLAGS="-O2 -fcommon -I$REF/src/lib/Dirac -I$REF/src/lib/Radio"
+gcc $CFLAGS -c "$REF/src/lib/Dirac/lbfgs.c" -o "$OUT/lbfgs.o"
+gcc $CFLAGS -c "$REF/src/lib/Dirac/myblas.c" -o "$OUT/myblas.o"
+gcc $CFLAGS -c "$REF/src/lib/Radio/predict.c" -o "$OUT/predict.o"
+gcc $CFLAGS -c "$HERE/miniblas.c" -o "$OUT/miniblas.o"
+gcc $CFLAGS "$HERE/oracle_lbfgs.c" "$OUT/lbfgs.o" "$OUT/myblas.o" \
+    "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_lbfgs"
+gcc $CFLAGS "$HERE/oracle_predict.c" "$OUT/predict.o" "$OUT/myblas.o" \
+    "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_predict"
+echo "built: $OUT/oracle_lbfgs $OUT/oracle_predict"
