@@ -23,24 +23,73 @@ FWHM = 2.0 * np.sqrt(2.0 * np.log(2.0))
 
 
 def find_islands(img, threshold=None, mask=None, nsigma=5.0):
-    """Connected components above threshold. Returns list of pixel-index
-    arrays (y, x)."""
+    """Island pixel lists (y, x).
+
+    With a Duchamp mask: pixels are grouped by their mask VALUE — each
+    distinct nonzero value is one Duchamp object — matching the
+    reference's id-keyed island table (buildsky.c read of the mask file);
+    connectivity relabeling would split disconnected objects and merge
+    touching ones. Without a mask: connected components above the
+    threshold (default 5-sigma via MAD)."""
     from scipy import ndimage
-    if mask is not None:
-        sel = mask > 0
-    else:
-        if threshold is None:
-            med = np.median(img)
-            mad = np.median(np.abs(img - med)) * 1.4826 + 1e-12
-            threshold = med + nsigma * mad
-        sel = img > threshold
-    lab, nl = ndimage.label(sel)
     islands = []
+    if mask is not None:
+        mi = np.rint(np.asarray(mask)).astype(np.int64)
+        for val in np.unique(mi):
+            if val == 0:
+                continue
+            ys, xs = np.nonzero(mi == val)
+            if len(ys) >= 1:
+                islands.append((ys, xs))
+        return islands
+    if threshold is None:
+        med = np.median(img)
+        mad = np.median(np.abs(img - med)) * 1.4826 + 1e-12
+        threshold = med + nsigma * mad
+    sel = img > threshold
+    lab, nl = ndimage.label(sel)
     for i in range(1, nl + 1):
         ys, xs = np.nonzero(lab == i)
         if len(ys) >= 1:
             islands.append((ys, xs))
     return islands
+
+
+def convex_hull(ys, xs):
+    """Convex hull of island pixels as an [Nh, 2] (y, x) vertex array
+    (the role of construct_hull, /root/reference/src/buildsky/hull.c:1-521
+    — computed with scipy instead of the reference's Graham stack)."""
+    pts = np.stack([ys, xs], axis=1).astype(float)
+    if len(pts) < 3 or np.ptp(pts[:, 0]) < 1e-9 or             np.ptp(pts[:, 1]) < 1e-9:
+        return None
+    from scipy.spatial import ConvexHull, QhullError
+    try:
+        h = ConvexHull(pts)
+    except QhullError:
+        return None
+    return pts[h.vertices]
+
+
+def outside_hull_distance(hull, y, x):
+    """0 inside the hull; distance beyond the nearest edge outside
+    (smooth penalty form of inside_hull, fitpixels.c:533-535)."""
+    if hull is None:
+        return 0.0
+    # signed distances to edges; orientation normalized so the hull
+    # centroid (always interior for a convex polygon) is "inside"
+    nh = len(hull)
+    cy, cx = hull[:, 0].mean(), hull[:, 1].mean()
+    d = -np.inf
+    for i in range(nh):
+        y0, x0 = hull[i]
+        y1, x1 = hull[(i + 1) % nh]
+        ex, ey = x1 - x0, y1 - y0
+        nrm = np.hypot(ex, ey) + 1e-30
+        cr_c = (ex * (cy - y0) - ey * (cx - x0))
+        sgn = 1.0 if cr_c > 0 else -1.0
+        cr = sgn * (ex * (y - y0) - ey * (x - x0)) / nrm
+        d = max(d, -cr)
+    return max(d, 0.0)
 
 
 def _moments(f, ys, xs):
@@ -71,9 +120,18 @@ def fit_island_multi(img, ys, xs, maxfits=10, criterion='aic'):
     """Fit 1..maxfits elliptical-Gaussian components to one island by LM
     (scipy least_squares plays the role of clmfit_nocuda.c) and select
     the model order by AIC (2p + n ln(rss/n)) or MDL ((p/2) ln n + ...)
-    — fitmultipixels.c model selection. Returns list of component dicts
-    in PIXEL units (flux, cy, cx, sy, sx, pa)."""
+    — fitmultipixels.c model selection. Component centres are constrained
+    to the island's convex hull by a smooth penalty (the reference's
+    inside_hull INFINITY_L penalty, fitpixels.c:533-535 + hull.c).
+    Returns list of component dicts in PIXEL units (flux, cy, cx, sy,
+    sx, pa)."""
     from scipy.optimize import least_squares
+    hull = convex_hull(ys, xs)
+
+    def _hull_pen(th, k):
+        return [100.0 * outside_hull_distance(hull, th[6 * i + 1],
+                                              th[6 * i + 2])
+                for i in range(k)]
     f = img[ys, xs].astype(float)
     ysf, xsf = ys.astype(float), xs.astype(float)
     n = len(f)
@@ -97,8 +155,9 @@ def fit_island_multi(img, ys, xs, maxfits=10, criterion='aic'):
         theta = list(theta) + [float(resid[j]), float(ysf[j]), float(xsf[j]),
                                s0, s0, 0.0]
         sol = least_squares(
-            lambda th: _gauss_eval(th, ysf, xsf, k) - f, theta,
-            method='lm', max_nfev=200 * k)
+            lambda th: np.concatenate(
+                [_gauss_eval(th, ysf, xsf, k) - f, _hull_pen(th, k)]),
+            theta, method='lm', max_nfev=200 * k)
         theta = list(sol.x)
         resid = f - _gauss_eval(sol.x, ysf, xsf, k)
         prev_rss = rss if k > 1 else None

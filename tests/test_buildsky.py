@@ -209,3 +209,77 @@ def test_restore_with_solutions_scales_flux(tmp_path):
     r2, _ = fitsio.read_fits_image(out2)
     ratio = r2.sum() / r1.sum()
     assert abs(ratio - 4.0) < 0.2, ratio
+
+
+def test_duchamp_mask_islands_by_value():
+    """A Duchamp mask groups pixels by object ID (mask value), not by
+    connectivity: one object with two disconnected parts stays ONE
+    island; two touching objects stay TWO islands (buildsky.c mask
+    semantics)."""
+    from sagecal_amd.apps.buildsky import find_islands
+    img = np.zeros((20, 20))
+    mask = np.zeros((20, 20))
+    # object 1: two disconnected patches
+    mask[2:4, 2:4] = 1
+    mask[10:12, 10:12] = 1
+    # objects 2 and 3: touching blocks with different ids
+    mask[15:17, 2:5] = 2
+    mask[15:17, 5:8] = 3
+    isl = find_islands(img, mask=mask)
+    assert len(isl) == 3
+    sizes = sorted(len(ys) for ys, xs in isl)
+    assert sizes == [6, 6, 8]
+
+
+def test_hull_penalty_helpers():
+    """Convex hull + outside-distance (hull.c / inside_hull penalty,
+    fitpixels.c:533)."""
+    from sagecal_amd.apps.buildsky import convex_hull, \
+        outside_hull_distance
+    ys, xs = np.mgrid[0:5, 0:5]
+    hull = convex_hull(ys.ravel(), xs.ravel())
+    assert hull is not None
+    # interior and vertex points: zero penalty
+    assert outside_hull_distance(hull, 2.0, 2.0) == 0.0
+    assert outside_hull_distance(hull, 0.0, 0.0) <= 1e-12
+    # a point 3 pixels beyond an edge
+    assert abs(outside_hull_distance(hull, 7.0, 2.0) - 3.0) < 1e-9
+    # degenerate (collinear) islands give None -> no penalty
+    assert convex_hull(np.zeros(4), np.arange(4.0)) is None
+    assert outside_hull_distance(None, 9.0, 9.0) == 0.0
+
+
+def test_restore_fft_shapelet_matches_direct(tmp_path):
+    """FFT-convolved shapelet rendering (fft.c path): with a narrow PSF
+    the FFT path converges to the direct basis evaluation; with a wide
+    PSF the peak smooths down but flux spreads, peak position intact."""
+    from sagecal_amd.apps.restore import render_shapelet_fft
+    from sagecal_amd import shapelet as shmod
+    ny = nx = 40
+    pscale = 2e-5
+    yy, xx = np.mgrid[0:ny, 0:nx]
+    lg = (xx - nx / 2) * pscale
+    mg = (yy - ny / 2) * pscale
+    n0, beta = 3, 8e-5
+    modes = np.zeros(9)
+    modes[0] = 1.0
+    modes[4] = 0.3
+    direct = shmod.image_basis(lg.ravel(), mg.ravel(), n0,
+                               beta).numpy() @ modes
+    direct = direct.reshape(ny, nx)
+    narrow = render_shapelet_fft(lg, mg, n0, beta, modes, 1.0, 1e-3,
+                                 1e-3)
+    assert np.allclose(narrow, direct, atol=1e-3 * np.abs(direct).max())
+    # wide PSF: peak position intact; shape is smoother (the unit-peak
+    # beam RAISES extended-source peaks — Jy/beam convention)
+    wide = render_shapelet_fft(lg, mg, n0, beta, modes, 1.0, 6.0, 6.0)
+    assert np.unravel_index(np.argmax(wide), wide.shape) == \
+        np.unravel_index(np.argmax(direct), direct.shape)
+    def width(a):
+        a = np.abs(a)
+        yy2, xx2 = np.mgrid[0:a.shape[0], 0:a.shape[1]]
+        cy = (a * yy2).sum() / a.sum()
+        cx = (a * xx2).sum() / a.sum()
+        return float(np.sqrt((a * ((yy2 - cy) ** 2
+                                   + (xx2 - cx) ** 2)).sum() / a.sum()))
+    assert width(wide) > width(direct) + 0.3   # convolution broadens
