@@ -175,12 +175,16 @@ class SyntheticMS:
             M = self.pack.M
             self.J_true = self.true_jones(M, tile_idx)
             fdelta_ch = self.fdelta / self.Nchan
+            # batched over (cluster, row): J_p C J_q^H summed over
+            # clusters in two bmm passes instead of M apply_jones calls
+            # (identical values; the per-cluster loop dominated large-N
+            # generation time)
+            J1 = self.J_true[:, bb[:, 0]]            # [M, rows, 2, 2]
+            J2h = self.J_true[:, bb[:, 1]].conj().transpose(-1, -2)
             for fi, f in enumerate(self.freqs):
                 coh = R.predict_coh(self.pack, u, v, w, float(f), self.freq0,
                                     fdelta_ch, self.tdelta, self.dec0)
-                for ci in range(M):
-                    xo[fi] += R.apply_jones(coh[ci],
-                                            self.J_true[ci:ci + 1], bb)
+                xo[fi] = ((J1 @ coh.to(J1.dtype)) @ J2h).sum(dim=0)
         # noise
         if self.noise_sigma > 0:
             sig = self.noise_sigma
