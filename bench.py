@@ -182,6 +182,10 @@ def main():
     local_rank = int(os.environ.get('LOCAL_RANK', str(rank)))
     dist = world > 1
     if dist:
+        # N concurrent ranks on one host each generating synthetic data
+        # on CPU: divide the cores instead of oversubscribing N-fold
+        torch.set_num_threads(max(1, (os.cpu_count() or 8) // world))
+    if dist:
         backend = os.environ.get('SAGECAL_BENCH_BACKEND')
         if backend is None:
             backend = 'nccl' if torch.cuda.is_available() else 'gloo'
