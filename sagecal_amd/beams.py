@@ -360,6 +360,14 @@ def predict_coh_withbeam(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0,
     (mode 1), or E_p C E_q^H with element beams (mode 2).
 
     Returns [M, B, 2, 2]. Times tmjd: [T] MJD days (one per timeslot).
+
+    The array factor is applied as its REAL magnitude |af| like the
+    reference (beamgain = |sum phasors|/K, stationbeam.c:318 and the GPU
+    beam buffer predict_model.cu:843-852), not as a complex gain.
+
+    On GPU with mode 1 the whole predict runs as ONE fused kernel call:
+    the torch-computed gains feed k_predict_coh's beam argument
+    (kernel_array_beam -> kernel_coherencies structure).
     """
     from .ops import reference as R
     B = u.shape[0]
@@ -367,7 +375,6 @@ def predict_coh_withbeam(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0,
     # per-source single predicts (phase/smear only), then scale by beam
     cdtype = torch.complex128 if u.dtype == torch.float64 \
         else torch.complex64
-    out = torch.zeros(M, B, 2, 2, dtype=cdtype, device=u.device)
     ra, dec = _pack_radec(pack, dec0)
     t_idx = torch.arange(B) // Nbase
     p = bb[:, 0]
@@ -376,7 +383,19 @@ def predict_coh_withbeam(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0,
     use_elem = mode in (2, 3)
     if use_array:
         af = array_beam(cfg, ra, dec, [freq], tmjd)  # [S, K, T, 1]
-        af = af[..., 0]                              # [S, K, T]
+        af = af[..., 0].abs()                        # [S, K, T] real
+    if mode == 1 and u.is_cuda:
+        # fused kernel path (-B 1): beam [T, K, N] + pairs into
+        # k_predict_coh; falls back to the torch loop if the extension
+        # is unavailable (dispatch raises on GPU by policy)
+        from .ops import hip_host
+        beam_t = af.permute(2, 1, 0).to(device=u.device,
+                                        dtype=torch.float32).contiguous()
+        return hip_host.predict_coh(
+            pack, u, v, w, freq, freq0, fdelta, tdelta, dec0,
+            beam=beam_t, pairs=bb[:Nbase].to(torch.int32),
+            Nbase=Nbase).to(cdtype)
+    out = torch.zeros(M, B, 2, 2, dtype=cdtype, device=u.device)
     if use_elem:
         # element E-Jones per (source, time): one dipole pattern per
         # array origin (kernel_element_beam; DOBEAM_ELEMENT/FULL modes)
@@ -402,9 +421,9 @@ def predict_coh_withbeam(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0,
                 Eb = Ej[gi, t_idx]                   # [B, 2, 2]
                 sub = Eb @ sub @ Eb.conj().transpose(-1, -2)
             if use_array:
-                gp = af[p, gi, t_idx]                # [B] complex
+                gp = af[p, gi, t_idx]                # [B] real |af|
                 gq = af[q, gi, t_idx]
-                sub = (gp * gq.conj()).to(sub.dtype)[:, None, None] * sub
+                sub = (gp * gq).to(sub.dtype)[:, None, None] * sub
             out[ci] += sub
     return out
 

@@ -6,10 +6,11 @@
 
 extern "C" {
 hipError_t launch_predict_coh(const double*, const double*, const double*,
-    const double*, const double*, const double*, const float*, const float*,
+    const double*, const double*, const double*, const float*,
     const float*, const float*, const float*, const float*, const float*,
     const float*, const float*, const float*, const float*, const float*,
-    const int*, const int*, int, int, double, double, double, float2*,
+    const float*, const int*, const int*, int, int, double, double,
+    double, const float*, const int*, int, int, int, float2*,
     hipStream_t);
 hipError_t launch_jtj_accum(const float2*, const float2*, const float2*,
     const int*, const int*, const float*, int, int, int, int, float2*,
@@ -49,11 +50,25 @@ torch::Tensor predict_coh(
     torch::Tensor eX, torch::Tensor eY, torch::Tensor eP,
     torch::Tensor cxi, torch::Tensor sxi, torch::Tensor cphi,
     torch::Tensor sphi, torch::Tensor r1, torch::Tensor stype,
-    torch::Tensor cluster_off, double freq, double fdelta2, double tdelta) {
+    torch::Tensor cluster_off, double freq, double fdelta2, double tdelta,
+    c10::optional<torch::Tensor> beam, c10::optional<torch::Tensor> pairs,
+    int64_t Nbase) {
   const int R = u.size(0);
   const int M = cluster_off.size(0) - 1;
   auto out = torch::empty({M, R, 4},
       torch::dtype(torch::kComplexFloat).device(u.device()));
+  const float* beam_p = nullptr;
+  const int* pairs_p = nullptr;
+  int Ktot = 0, Nsta = 0;
+  if (beam.has_value()) {
+    TORCH_CHECK(pairs.has_value(), "beam needs the station-pair table");
+    TORCH_CHECK(beam->is_contiguous() && beam->dim() == 3,
+                "beam must be contiguous [T, K, N] float32");
+    beam_p = beam->data_ptr<float>();
+    pairs_p = pairs->data_ptr<int>();
+    Ktot = beam->size(1);
+    Nsta = beam->size(2);
+  }
   CHECK_HIP(launch_predict_coh(
       u.data_ptr<double>(), v.data_ptr<double>(), w.data_ptr<double>(),
       ll.data_ptr<double>(), mm.data_ptr<double>(), nn1.data_ptr<double>(),
@@ -62,7 +77,8 @@ torch::Tensor predict_coh(
       eP.data_ptr<float>(), cxi.data_ptr<float>(), sxi.data_ptr<float>(),
       cphi.data_ptr<float>(), sphi.data_ptr<float>(), r1.data_ptr<float>(),
       stype.data_ptr<int>(), cluster_off.data_ptr<int>(), M, R, freq,
-      fdelta2, tdelta, cptr(out), cur_stream()));
+      fdelta2, tdelta, beam_p, pairs_p, (int)Nbase, Ktot, Nsta,
+      cptr(out), cur_stream()));
   return out;
 }
 
@@ -177,7 +193,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("chol_solve", &chol_solve, "batched damped Cholesky solve (gfx950)");
   m.def("chol_solve_mw", &chol_solve_mw,
         "multi-workgroup damped Cholesky solve (gfx950)");
-  m.def("predict_coh", &predict_coh, "coherency predict (gfx950)");
+  m.def("predict_coh", &predict_coh, "coherency predict (gfx950)",
+        py::arg("u"), py::arg("v"), py::arg("w"), py::arg("ll"),
+        py::arg("mm"), py::arg("nn1"), py::arg("sI"), py::arg("sQ"),
+        py::arg("sU"), py::arg("sV"), py::arg("eX"), py::arg("eY"),
+        py::arg("eP"), py::arg("cxi"), py::arg("sxi"), py::arg("cphi"),
+        py::arg("sphi"), py::arg("r1"), py::arg("stype"),
+        py::arg("cluster_off"), py::arg("freq"), py::arg("fdelta2"),
+        py::arg("tdelta"), py::arg("beam") = py::none(),
+        py::arg("pairs") = py::none(), py::arg("Nbase") = 0);
   m.def("jtj_jtr", &jtj_jtr, "fused JtJ/Jtr assembly (gfx950)");
   m.def("model_cost", &model_cost, "per-chunk model cost (gfx950)");
   m.def("apply_jones", &apply_jones, "model apply / residual (gfx950)");

@@ -14,12 +14,14 @@ hipError_t launch_predict_coh(
     const float* eP, const float* cxi, const float* sxi, const float* cphi,
     const float* sphi, const float* r1, const int* stype,
     const int* cluster_off, int M, int R, double freq, double fdelta2,
-    double tdelta, float2* out, hipStream_t stream) {
+    double tdelta, const float* beam, const int* pairs, int Nbase,
+    int Ktot, int Nsta, float2* out, hipStream_t stream) {
   const int tb = 128;
   const int nb = (R + tb - 1) / tb;
   hipLaunchKernelGGL(k_predict_coh, dim3(nb), dim3(tb), 0, stream,
       u, v, w, ll, mm, nn1, sI, sQ, sU, sV, eX, eY, eP, cxi, sxi, cphi,
-      sphi, r1, stype, cluster_off, M, R, freq, fdelta2, tdelta, out);
+      sphi, r1, stype, cluster_off, M, R, freq, fdelta2, tdelta,
+      beam, pairs, Nbase, Ktot, Nsta, out);
   return hipGetLastError();
 }
 

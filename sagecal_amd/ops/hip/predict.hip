@@ -8,6 +8,13 @@
 // math; shapelet envelopes are folded in by the host path (shapelet.py).
 //
 // Layout: rows r = t*Nbase + b (time-major); output coh [M, R, 4] complex64.
+//
+// Optional fused station-beam path (-B 1, the reference's DOBEAM_ARRAY:
+// predict_model.cu:843-852 'scalef *= beam1*beam2'): when `beam` is
+// non-null it holds the REAL scalar array-factor gain per
+// (timeslot, source, station) [T, K, N] (beamgain = |sum phasors|/K,
+// stationbeam.c:318 — the reference's array factor is real), and each
+// source's contribution is scaled by beam[t,g,sta1]*beam[t,g,sta2].
 #include "common.h"
 
 #define SRC_TILE 64
@@ -35,10 +42,23 @@ k_predict_coh(const double* __restrict__ u, const double* __restrict__ v,
               const int* __restrict__ stype,
               const int* __restrict__ cluster_off, int M, int R,
               double freq, double fdelta2,      // channel halfwidth [Hz]
-              double tdelta, float2* __restrict__ out) {
+              double tdelta,
+              const float* __restrict__ beam,   // [T, K, N] or null
+              const int* __restrict__ pairs,    // [Nbase, 2] (with beam)
+              int Nbase, int Ktot, int Nsta,
+              float2* __restrict__ out) {
   __shared__ SrcTile tile;
   const int r = blockIdx.x * blockDim.x + threadIdx.x;
   const bool live = r < R;
+  // beam indexing state (only touched when beam != null)
+  int b_sta1 = 0, b_sta2 = 0;
+  size_t b_toff = 0;
+  if (beam != nullptr && live) {
+    const int t = r / Nbase, bl = r - t * Nbase;
+    b_sta1 = pairs[2 * bl];
+    b_sta2 = pairs[2 * bl + 1];
+    b_toff = (size_t)t * Ktot;
+  }
   double ur = 0, vr = 0, wr = 0;
   float blf = 0.f;
   if (live) {
@@ -111,6 +131,11 @@ k_predict_coh(const double* __restrict__ u, const double* __restrict__ v,
             phc = cscale(phc, j0f(b));
           }
           // st==4 (shapelet): envelope folded in on the host path
+        }
+        if (beam != nullptr) {
+          const float* bg = beam +
+              (b_toff + (size_t)(base + k)) * (size_t)Nsta;
+          phc = cscale(phc, bg[b_sta1] * bg[b_sta2]);
         }
         aI = cadd(aI, cscale(phc, tile.sI[k]));
         aQ = cadd(aQ, cscale(phc, tile.sQ[k]));

@@ -113,18 +113,28 @@ def gpu_pack(pack, device):
     return _gpu_pack_cache[key]
 
 
-def predict_coh(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0, **kw):
+def predict_coh(pack, u, v, w, freq, freq0, fdelta, tdelta, dec0,
+                beam=None, pairs=None, Nbase=0, **kw):
+    """Kernel-backed coherency predict. Optional fused station beam
+    (-B 1, reference DOBEAM_ARRAY): `beam` [T, K, N] float32 real
+    array-factor gains on device + `pairs` [Nbase, 2] int32 — each
+    source's contribution is scaled by beam[t,k,p]*beam[t,k,q] inside
+    k_predict_coh (predict_model.cu:843-852 semantics)."""
     gp = pack if isinstance(pack, GPUPack) else gpu_pack(pack, u.device)
     sI, sQ, sU, sV = gp.fluxes_at(freq, freq0)
     r1 = gp.r1_for(dec0)
     u64 = u.to(torch.float64).contiguous()
     v64 = v.to(torch.float64).contiguous()
     w64 = w.to(torch.float64).contiguous()
+    if beam is not None:
+        beam = beam.to(device=u.device, dtype=torch.float32).contiguous()
+        pairs = pairs.to(device=u.device, dtype=torch.int32).contiguous()
+        Nbase = int(Nbase) or int(pairs.shape[0])
     out = _ext().predict_coh(
         u64, v64, w64, gp.ll, gp.mm, gp.nn1, sI, sQ, sU, sV,
         gp.eX, gp.eY, gp.eP, gp.cxi, gp.sxi, gp.cphi, gp.sphi, r1,
         gp.stype, gp.cluster_off, float(freq), float(fdelta) * 0.5,
-        float(tdelta))
+        float(tdelta), beam=beam, pairs=pairs, Nbase=int(Nbase))
     out = out.view(gp.M, -1, 2, 2)
     if gp.shapelets:
         out = out + _shapelet_coh(gp, u64, v64, w64, freq, freq0, fdelta,

@@ -318,3 +318,40 @@ def test_rtr_graphed_matches_eager():
     ce = float(info_e['final_cost'].max())
     cg = float(info_g['final_cost'].max())
     assert cg <= 1.2 * ce + 1e-6, (ce, cg)
+
+
+@pytest.mark.skipif(os.environ.get('SAGECAL_BEAM_KERNEL_TEST') != '1',
+                    reason='fused beam kernel written round 2, '
+                           'GPU-validated next round '
+                           '(SAGECAL_BEAM_KERNEL_TEST=1)')
+def test_fused_beam_predict_matches_torch():
+    """k_predict_coh with the beam argument (-B 1 fused path,
+    predict_model.cu:843-852 semantics) vs the torch loop on CPU."""
+    from sagecal_amd import sky, beams
+    from sagecal_amd.ops.reference import SourcePack
+    dev = 'cuda:0'
+    srcs, clist = sky.make_synthetic_sky(M=2, nsrc_per_cluster=3, seed=3)
+    clusters = sky.build_clusters(srcs, clist, 0.0, np.pi / 4, 150e6)
+    pack = SourcePack(clusters)
+    N, T = 8, 2
+    Nbase = N * (N - 1) // 2
+    rng = np.random.default_rng(4)
+    B = Nbase * T
+    u = torch.tensor(rng.uniform(-1e-5, 1e-5, B))
+    v = torch.tensor(rng.uniform(-1e-5, 1e-5, B))
+    w = torch.zeros(B, dtype=torch.float64)
+    bb = torch.tensor([(p_, q_) for p_ in range(N)
+                       for q_ in range(p_ + 1, N)] * T)
+    tmjd = np.array([57000.0, 57000.0001])
+    rng2 = np.random.default_rng(7)
+    elems = [rng2.uniform(-30, 30, (16, 3)) for _ in range(N)]
+    cfg = beams.ArrayConfig(elems, 0.1, 0.92, 0.0, np.pi / 4)
+    args = (pack, u, v, w, 150e6, 150e6, 1e4, 1.0, np.pi / 4, cfg, tmjd,
+            bb, Nbase, T)
+    cpu = beams.predict_coh_withbeam(*args, mode=1)
+    gargs = (pack, u.to(dev), v.to(dev), w.to(dev), 150e6, 150e6, 1e4,
+             1.0, np.pi / 4, cfg, tmjd, bb.to(dev), Nbase, T)
+    gpu = beams.predict_coh_withbeam(*gargs, mode=1)
+    err = float((gpu.cpu().to(cpu.dtype) - cpu).abs().max()
+                / cpu.abs().max())
+    assert err < 2e-4, err
