@@ -25,6 +25,11 @@
 #define PST 36   // padded LDS panel row stride (floats)
 #define NTH 512
 #define MAXN_CHOL 1024
+// multi-workgroup path row-chunking: panels taller than PANEL_CAP rows
+// stream through LDS in passes, lifting the mw size cap to MAXN_CHOL_MW
+// (the 512-station 8N=4096 normal equations, clmfit_cuda.c:1624-1674)
+#define PANEL_CAP 1024
+#define MAXN_CHOL_MW 4096
 
 extern "C" __global__ void __launch_bounds__(NTH)
 k_chol_solve(const float* __restrict__ JtJ, const float* __restrict__ Jtr,
@@ -331,13 +336,20 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
   float* LT = L + (size_t)n * n;
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* pan = smem;
+  // factored 32x32 diag block kept resident across row chunks so panels
+  // taller than the LDS budget (n > PANEL_CAP, up to MAXN_CHOL_MW) are
+  // processed in passes: the row-solve/LT-writeback/fwd-subst of a row
+  // depend only on that block + rdg/yv, never on other rows.
+  __shared__ float diag[NB * PST];
   __shared__ float yv[NB];
   __shared__ float rdg[NB];   // reciprocal diagonal (div -> mul downstream)
   __shared__ int bad;
   if (tid == 0) bad = 0;
   const int rows = n - k;
-  // stage panel rows k..n, cols k..k+NB from the (already-updated) L copy
-  for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
+  const int cap = rows < PANEL_CAP ? rows : PANEL_CAP;
+  // ---- pass 0: rows [0, cap) ----
+  // stage panel rows k..k+cap, cols k..k+NB from the (already-updated) L
+  for (int idx = tid; idx < cap * (NB / 4); idx += NTH) {
     const int r = idx >> 3, c4 = (idx & 7) << 2;
     *(float4*)(pan + r * PST + c4) =
         *(const float4*)(L + (size_t)(k + r) * n + k + c4);
@@ -375,9 +387,15 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
     pan[r * PST + r] = rdg[r];
   }
   __syncthreads();
-  // row-solve sub-panel rows NB..rows (multiply by reciprocal diag: the
+  // keep the factored diag block for later row chunks
+  for (int idx = tid; idx < NB * NB; idx += NTH) {
+    const int r = idx >> 5, c = idx & 31;
+    diag[r * PST + c] = pan[r * PST + c];
+  }
+  __syncthreads();
+  // row-solve sub-panel rows NB..cap (multiply by reciprocal diag: the
   // 32 serial divides per row-thread were the panel's longest chain)
-  for (int r = NB + tid; r < rows; r += NTH) {
+  for (int r = NB + tid; r < cap; r += NTH) {
     float rw[NB];
 #pragma unroll
     for (int c = 0; c < NB; ++c) rw[c] = pan[r * PST + c];
@@ -385,7 +403,7 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
     for (int c = 0; c < NB; ++c) {
       float s = rw[c];
 #pragma unroll
-      for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * pan[c * PST + c2];
+      for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * diag[c * PST + c2];
       rw[c] = s * rdg[c];
     }
 #pragma unroll
@@ -398,7 +416,7 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
   // all later panel stages ever read.
 #pragma unroll
   for (int c = 0; c < NB; ++c) {
-    for (int r = tid; r < rows; r += NTH) {
+    for (int r = tid; r < cap; r += NTH) {
       LT[(size_t)(k + c) * n + k + r] = pan[r * PST + c];
     }
   }
@@ -414,7 +432,7 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
     if (lane < 32) {
 #pragma unroll
       for (int c = 0; c < NB; ++c)
-        rv[c] = (c <= r) ? pan[r * PST + c] : 0.f;
+        rv[c] = (c <= r) ? diag[r * PST + c] : 0.f;
     }
 #pragma unroll
     for (int c = 0; c < NB; ++c) {
@@ -427,11 +445,48 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
     if (lane < 32) xo[k + r] = bv;
   }
   __syncthreads();
-  for (int r = NB + tid; r < rows; r += NTH) {
+  for (int r = NB + tid; r < cap; r += NTH) {
     float s = 0.f;
 #pragma unroll
     for (int c = 0; c < NB; ++c) s += pan[r * PST + c] * yv[c];
     xo[k + r] -= s;
+  }
+  // ---- passes 1..: remaining row chunks (uniform trip count) ----
+  for (int r0 = cap; r0 < rows; r0 += PANEL_CAP) {
+    const int nr = (rows - r0) < PANEL_CAP ? (rows - r0) : PANEL_CAP;
+    __syncthreads();
+    for (int idx = tid; idx < nr * (NB / 4); idx += NTH) {
+      const int r = idx >> 3, c4 = (idx & 7) << 2;
+      *(float4*)(pan + r * PST + c4) =
+          *(const float4*)(L + (size_t)(k + r0 + r) * n + k + c4);
+    }
+    __syncthreads();
+    for (int r = tid; r < nr; r += NTH) {
+      float rw[NB];
+#pragma unroll
+      for (int c = 0; c < NB; ++c) rw[c] = pan[r * PST + c];
+#pragma unroll
+      for (int c = 0; c < NB; ++c) {
+        float s = rw[c];
+#pragma unroll
+        for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * diag[c * PST + c2];
+        rw[c] = s * rdg[c];
+      }
+      float sxo = 0.f;
+#pragma unroll
+      for (int c = 0; c < NB; ++c) {
+        pan[r * PST + c] = rw[c];
+        sxo += rw[c] * yv[c];
+      }
+      xo[k + r0 + r] -= sxo;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      for (int r = tid; r < nr; r += NTH) {
+        LT[(size_t)(k + c) * n + k + r0 + r] = pan[r * PST + c];
+      }
+    }
   }
   if (bad && tid == 0) atomicOr(&info[bid], 1);
 }

@@ -85,7 +85,8 @@ hipError_t launch_chol_mw(
   hipLaunchKernelGGL(k_cholmw_init, dim3(batch, INIT_SLICES), dim3(512), 0,
       stream, JtJ, Jtr, mu, n, Lbuf, dp);
   if (stages < 2) return hipGetLastError();
-  const size_t shmem = (size_t)(n + 40) * PST * sizeof(float);
+  const int cap = n < PANEL_CAP ? n : PANEL_CAP;
+  const size_t shmem = (size_t)(cap + 8) * PST * sizeof(float);
   for (int k = 0; k < n; k += NB) {
     hipLaunchKernelGGL(k_cholmw_panel, dim3(batch), dim3(512), shmem,
         stream, n, k, Lbuf, dp, info);

@@ -355,3 +355,26 @@ def test_fused_beam_predict_matches_torch():
     err = float((gpu.cpu().to(cpu.dtype) - cpu).abs().max()
                 / cpu.abs().max())
     assert err < 2e-4, err
+
+
+@pytest.mark.skipif(os.environ.get('SAGECAL_CHOL_BIG') != '1',
+                    reason='large-n chunked-panel Cholesky written '
+                           'round 2, GPU-validated next round '
+                           '(SAGECAL_CHOL_BIG=1)')
+def test_chol_mw_large_n4096():
+    """Chunked-panel mw Cholesky at the 512-station LM shape
+    (8N=4096, clmfit_cuda.c:1624-1674 role) vs torch.cholesky_solve."""
+    from sagecal_amd.ops.hip_host import chol_solve_damped
+    dev = 'cuda:0'
+    torch.manual_seed(5)
+    n, batch = 4096, 2
+    G = torch.randn(batch, n, n, device=dev) / n ** 0.5
+    A = G @ G.transpose(-1, -2) + 0.5 * torch.eye(n, device=dev)
+    b = torch.randn(batch, n, device=dev)
+    mu = torch.full((batch,), 0.1, device=dev)
+    dp = chol_solve_damped(A, b, mu)
+    Ad = (A + mu[:, None, None] * torch.eye(n, device=dev)).double()
+    L = torch.linalg.cholesky(Ad)
+    ref = torch.cholesky_solve(b.double().unsqueeze(-1), L).squeeze(-1)
+    err = (dp.double() - ref).abs().max() / ref.abs().max()
+    assert float(err) < 5e-2, float(err)   # fp32 kernel vs fp64 at n=4096
