@@ -364,6 +364,8 @@ def chol_solve_damped(JtJ, Jtr, mu):
     mw = os.environ.get('SAGECAL_CHOL_MW')
     use_mw = (JtJ.shape[1] >= 384) if mw is None else mw == '1'
     if use_mw:
+        # chunked panels (round 2) support up to n=4096
+        assert JtJ.shape[1] <= 4096, "mw Cholesky supports n <= 4096"
         dp, info = _ext().chol_solve_mw(JtJ.contiguous(), Jtr.contiguous(),
                                         mu.to(torch.float32).contiguous(),
                                         sc, 4)
