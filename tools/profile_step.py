@@ -24,9 +24,9 @@ def main():
     ap.add_argument('--tilesz', type=int, default=60)
     ap.add_argument('--intervals', type=int, default=8)
     ap.add_argument('--chan', type=int, default=8)
-    ap.add_argument('--emiter', type=int, default=3)
+    ap.add_argument('--emiter', type=int, default=2)
     ap.add_argument('--maxiter', type=int, default=8)
-    ap.add_argument('--em-group', type=int, default=3)
+    ap.add_argument('--em-group', type=int, default=10)
     ap.add_argument('--reps', type=int, default=2)
     args = ap.parse_args()
     sys.argv = [sys.argv[0]]
