@@ -53,6 +53,32 @@ def _cost_grad_band(v, x, cohs, chunk_off, nchunks, bb, T, Nbase, Mt, N,
     return cost, grad
 
 
+def _cost_grad_band_multifreq(v, xs, cohs_f, chunk_off, nchunks, bb, T,
+                              Nbase, Mt, N, cdtype, robust_nu,
+                              consensus=None):
+    """Multi-channel LBFGS cost/grad: per-channel residuals summed over
+    the mini-band's channels instead of fitting the channel average —
+    the role of the reference's fused multifreq gradient kernel
+    (lbfgs_multifreq.cu:29 kernel_deriv_r_robust / cudakernel_lbfgs_
+    multifreq_r_robust, Dirac.h:523-525). xs: [F, B, 2, 2];
+    cohs_f: list of [M, B, 2, 2] per channel."""
+    J = torch.view_as_complex(v.reshape(Mt, N, 2, 2, 2).contiguous())
+    cost = None
+    grad = None
+    for fi in range(xs.shape[0]):
+        c, g = ops.lbfgs_cost_grad(xs[fi], cohs_f[fi], J, chunk_off,
+                                   nchunks, bb, T, Nbase,
+                                   robust_nu=robust_nu)
+        cost = c if cost is None else cost + c
+        grad = g if grad is None else grad + g
+    if consensus is not None:
+        y, bz, rho = consensus
+        d = v - bz
+        cost = cost + (y * d).sum() + 0.5 * (rho * d * d).sum()
+        grad = grad + y + rho * d
+    return cost, grad
+
+
 def bfgsfit_minibatch(band, x, cohs, bb, T, Nbase, chunk_off, nchunks,
                       lbfgs_iters=6, m=7, robust_nu=None, consensus=None):
     """One minibatch LBFGS fit with persistent memory. Returns final cost."""
@@ -94,6 +120,7 @@ class MinibatchConsensusCalibration:
     (find_prod_inverse_full_fed with alpha)."""
 
     def __init__(self, pack, N, freqs, nsolbw=2, Npoly=2, poly_type=0,
+                 multifreq=False,
                  rho=1.0, device='cpu', dtype=torch.complex128,
                  fed_alpha=0.0, dist_group=None, world=1, rank=0):
         self.pack = pack
@@ -130,6 +157,11 @@ class MinibatchConsensusCalibration:
         self.dist_group = dist_group
         self.world, self.rank = world, rank
         self.Xlag = torch.zeros_like(self.Z)    # federated Lagrange X
+        # multifreq: per-channel gradient accumulation across each
+        # mini-band (lbfgs_multifreq.cu semantics) instead of fitting
+        # the band's channel average — exact for steep bandpasses,
+        # costs one predict per channel
+        self.multifreq = multifreq
 
     def _consensus_vec(self, bi):
         """(y, bz) flattened for band bi from current Z and duals."""
@@ -150,6 +182,14 @@ class MinibatchConsensusCalibration:
             bbm = bb[rows]
             Tm = t1 - t0
             for bi, (a, b) in enumerate(self.bands):
+                band = self.states[bi]
+                bz = self._consensus_vec(bi)
+                cons = (band.Y, bz, self.rho_scalar)
+                if self.multifreq:
+                    self._fit_band_multifreq(band, tile, a, b, rows, bbm,
+                                             Tm, Nbase, fdelta_ch,
+                                             lbfgs_iters, robust_nu, cons)
+                    continue
                 f = float(self.band_freqs[bi])
                 cohs = ops.predict_coh(self.pack, u, v, w, f, tile.freq0,
                                        fdelta_ch * (b - a), tile.tdelta,
@@ -157,14 +197,36 @@ class MinibatchConsensusCalibration:
                 if cohs.dtype != self.dtype:
                     cohs = cohs.to(self.dtype)
                 xb = tile.xo[a:b, rows].mean(dim=0).to(self.dtype)
-                band = self.states[bi]
-                bz = self._consensus_vec(bi)
-                cons = (band.Y, bz, self.rho_scalar)
                 bfgsfit_minibatch(band, xb, cohs, bbm, Tm, Nbase,
                                   self.chunk_off, self.nchunks,
                                   lbfgs_iters=lbfgs_iters,
                                   robust_nu=robust_nu, consensus=cons)
             self.consensus_update()
+
+    def _fit_band_multifreq(self, band, tile, a, b, rows, bbm, Tm,
+                            Nbase, fdelta_ch, lbfgs_iters, robust_nu,
+                            cons):
+        from . import lbfgs as lbfgs_mod
+        u, v, w = tile.u[rows], tile.v[rows], tile.w[rows]
+        cohs_f = []
+        for fi in range(a, b):
+            c = ops.predict_coh(self.pack, u, v, w,
+                                float(tile.freqs[fi]), tile.freq0,
+                                fdelta_ch, tile.tdelta, tile.dec0)
+            cohs_f.append(c.to(self.dtype))
+        xs = tile.xo[a:b, rows].to(self.dtype)
+        Mt, N = band.J.shape[0], band.J.shape[1]
+
+        def fg(vv):
+            return _cost_grad_band_multifreq(
+                vv, xs, cohs_f, self.chunk_off, self.nchunks, bbm, Tm,
+                Nbase, Mt, N, self.dtype, robust_nu, consensus=cons)
+        v0 = torch.view_as_real(band.J).reshape(-1).clone()
+        v1, band.mem, info = lbfgs_mod.lbfgs_fit(
+            fg, v0, maxiter=lbfgs_iters, m=7, mem=band.mem,
+            stochastic=True)
+        band.J = torch.view_as_complex(
+            v1.reshape(Mt, N, 2, 2, 2).contiguous())
 
     def consensus_update(self):
         """Global Z from all mini-bands + dual updates
