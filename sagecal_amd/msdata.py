@@ -172,6 +172,7 @@ class SyntheticMS:
         xo = torch.zeros(self.Nchan, rows, 2, 2, dtype=cdtype,
                          device=self.device)
         if self.pack is not None:
+            import os as _os
             M = self.pack.M
             self.J_true = self.true_jones(M, tile_idx)
             fdelta_ch = self.fdelta / self.Nchan
@@ -181,10 +182,26 @@ class SyntheticMS:
             # generation time)
             J1 = self.J_true[:, bb[:, 0]]            # [M, rows, 2, 2]
             J2h = self.J_true[:, bb[:, 1]].conj().transpose(-1, -2)
+            # SAGECAL_GEN_KERNEL=1 on GPU: truth predict through the HIP
+            # kernel (fp64 phase inside, complex64 out) — A/B switch for
+            # large-array generation (ROUND3_NOTES P0)
+            use_kernel = (_os.environ.get('SAGECAL_GEN_KERNEL') == '1'
+                          and str(self.device).startswith('cuda'))
+            if use_kernel:
+                from .ops import dispatch as _disp
+                J1 = J1.to(torch.complex64)
+                J2h = J2h.to(torch.complex64)
             for fi, f in enumerate(self.freqs):
-                coh = R.predict_coh(self.pack, u, v, w, float(f), self.freq0,
-                                    fdelta_ch, self.tdelta, self.dec0)
-                xo[fi] = ((J1 @ coh.to(J1.dtype)) @ J2h).sum(dim=0)
+                if use_kernel:
+                    coh = _disp.predict_coh(self.pack, u, v, w, float(f),
+                                            self.freq0, fdelta_ch,
+                                            self.tdelta, self.dec0)
+                    xo[fi] = ((J1 @ coh) @ J2h).sum(dim=0).to(xo.dtype)
+                else:
+                    coh = R.predict_coh(self.pack, u, v, w, float(f),
+                                        self.freq0, fdelta_ch,
+                                        self.tdelta, self.dec0)
+                    xo[fi] = ((J1 @ coh.to(J1.dtype)) @ J2h).sum(dim=0)
         # noise
         if self.noise_sigma > 0:
             sig = self.noise_sigma
