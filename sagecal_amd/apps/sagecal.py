@@ -72,6 +72,10 @@ def build_argparser():
                     help='>0: stochastic calibration epochs')
     ap.add_argument('-M', dest='minibatches', type=int, default=1)
     ap.add_argument('-w', dest='minibands', type=int, default=1)
+    ap.add_argument('--multifreq', action='store_true',
+                    help='per-channel gradient across each mini-band '
+                         '(lbfgs_multifreq semantics) instead of the '
+                         'channel-average fit')
     ap.add_argument('-A', dest='nadmm', type=int, default=1,
                     help='consensus updates per epoch (stochastic -w>1)')
     ap.add_argument('-P', dest='npoly', type=int, default=2)
@@ -301,7 +305,7 @@ def run_stochastic(args):
     cal = MinibatchConsensusCalibration(
         pack, ms.N, ms.freqs, nsolbw=args.minibands, Npoly=args.npoly,
         poly_type=args.polytype, rho=args.admm_rho, device=device,
-        dtype=cdtype)
+        dtype=cdtype, multifreq=getattr(args, 'multifreq', False))
     bb = ms.bb_tensor(device=device)
     for ti, tile in enumerate(ms.tiles()):
         for ep in range(args.epochs):
