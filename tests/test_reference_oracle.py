@@ -147,3 +147,35 @@ def test_predict_matches_reference(oracle, tmp_path):
     err = np.abs(ours - ref).max() / scale
     # 5e-9: torch.special bessel j0/j1 vs libm j0/j1 precision
     assert err < 5e-9, f"predict mismatch: rel err {err}"
+
+
+def test_element_beam_matches_reference(oracle, tmp_path):
+    """Reference eval_elementcoeffs (elementbeam.c:384, with the REAL
+    LBA/HBA/ALO tables compiled in) vs beams.LofarElementCoeffs.basis
+    on the ported .npz tables — the port + Laguerre-Gaussian basis are
+    pinned to the reference value-by-value, including the frequency
+    interpolation."""
+    from sagecal_amd.beams import LofarElementCoeffs
+    rng = np.random.default_rng(21)
+    r = rng.uniform(0.0, np.pi / 2, 12)
+    t = rng.uniform(-np.pi, np.pi, 12)
+    pts = ''.join(f"{a:.17g} {b:.17g}\n" for a, b in zip(r, t))
+    for kind, freq in (('lba', 55e6), ('hba', 155e6), ('alo', 2.3e7),
+                       ('lba', 1e6), ('hba', 9e9)):   # incl. edge clamps
+        out = subprocess.run(
+            [os.path.join(oracle, 'oracle_element'), kind, str(freq)],
+            input=pts, capture_output=True, text=True, check=True).stdout
+        vals = np.array([[float(x) for x in ln.split()]
+                         for ln in out.strip().splitlines()])
+        ref_theta = vals[:, 0] + 1j * vals[:, 1]
+        ref_phi = vals[:, 2] + 1j * vals[:, 3]
+        ec = LofarElementCoeffs.load(kind)
+        ct, cp = ec.at_freq(freq)
+        B = ec.basis(r, t)
+        ours_theta = B @ ct
+        ours_phi = B @ cp
+        scale = max(np.abs(ref_theta).max(), np.abs(ref_phi).max())
+        assert np.allclose(ours_theta, ref_theta, atol=1e-12 * scale), \
+            (kind, freq)
+        assert np.allclose(ours_phi, ref_phi, atol=1e-12 * scale), \
+            (kind, freq)

@@ -17,4 +17,8 @@ gcc $CFLAGS "$HERE/oracle_lbfgs.c" "$OUT/lbfgs.o" "$OUT/myblas.o" \
     "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_lbfgs"
 gcc $CFLAGS "$HERE/oracle_predict.c" "$OUT/predict.o" "$OUT/myblas.o" \
     "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_predict"
-echo "built: $OUT/oracle_lbfgs $OUT/oracle_predict"
+gcc $CFLAGS -c "$REF/src/lib/Radio/elementbeam.c" -o "$OUT/elementbeam.o"
+gcc $CFLAGS "$HERE/oracle_element.c" "$OUT/elementbeam.o" \
+    "$OUT/myblas.o" "$OUT/miniblas.o" -lpthread -lm \
+    -o "$OUT/oracle_element"
+echo "built: $OUT/oracle_lbfgs $OUT/oracle_predict $OUT/oracle_element"
