@@ -18,7 +18,13 @@ gcc $CFLAGS "$HERE/oracle_lbfgs.c" "$OUT/lbfgs.o" "$OUT/myblas.o" \
 gcc $CFLAGS "$HERE/oracle_predict.c" "$OUT/predict.o" "$OUT/myblas.o" \
     "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_predict"
 gcc $CFLAGS -c "$REF/src/lib/Radio/elementbeam.c" -o "$OUT/elementbeam.o"
+gcc $CFLAGS -c "$REF/src/lib/Radio/transforms.c" -o "$OUT/transforms.o"
+gcc $CFLAGS -c "$REF/src/lib/Dirac/updatenu.c" -o "$OUT/updatenu.o"
+gcc $CFLAGS -c "$REF/src/lib/Radio/shapelet.c" -o "$OUT/shapelet.o"
 gcc $CFLAGS "$HERE/oracle_element.c" "$OUT/elementbeam.o" \
     "$OUT/myblas.o" "$OUT/miniblas.o" -lpthread -lm \
     -o "$OUT/oracle_element"
-echo "built: $OUT/oracle_lbfgs $OUT/oracle_predict $OUT/oracle_element"
+gcc $CFLAGS "$HERE/oracle_misc.c" "$OUT/transforms.o" "$OUT/updatenu.o" \
+    "$OUT/shapelet.o" "$OUT/elementbeam.o" "$OUT/myblas.o" \
+    "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_misc"
+echo "built: $OUT/oracle_lbfgs $OUT/oracle_predict $OUT/oracle_element $OUT/oracle_misc"

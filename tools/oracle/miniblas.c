@@ -134,7 +134,9 @@ void dgemm_(char *ta, char *tb, int *m, int *n, int *k, double *alpha,
       c[i + j * *ldc] = *alpha * s + *beta * c[i + j * *ldc];
     }
 }
-#define STUB(name) void name() { \
+/* weak: a real implementation linked alongside (e.g. shapelet.o for
+ * oracle_misc) overrides the stub */
+#define STUB(name) __attribute__((weak)) void name() { \
   fprintf(stderr, "miniblas: " #name " not implemented (oracle)\n"); \
   abort(); }
 STUB(cgels_) STUB(dgels_) STUB(dgeqrf_) STUB(dgesvd_) STUB(dorgqr_)
@@ -143,3 +145,5 @@ STUB(zgemm_) STUB(zgesvd_) STUB(zher_)
 /* predict.c references shapelet_contrib (shapelet.c) — oracle covers
  * point/gaussian/disk/ring only */
 STUB(shapelet_contrib)
+/* pngoutput.c dependency of shapelet.c's plot helper — unused in oracle */
+STUB(convert_tensor_to_image)
