@@ -34,6 +34,17 @@ def build_problem(args, device, dtype, rank=0, world=1):
     # solutions, one per `tilesz`-slot interval — P x em_group problems
     # per batched LM solve, sized to fill 256 CUs.
     P = max(1, getattr(args, 'intervals', 1))
+    # clamp the interval batch so the resident per-cluster coherencies
+    # stay within a sane fraction of 288 GB HBM (or CPU RAM): cohs are
+    # [dirs, rows, 2, 2] complex64 + data ~3x that
+    nb = args.stations * (args.stations - 1) // 2
+    bytes_per_iv = nb * args.tilesz * (args.dirs + 3) * 32
+    cap = max(1, int(96e9 // max(bytes_per_iv, 1)))
+    if P > cap:
+        print(f"# bench: clamping --intervals {P} -> {cap} "
+              f"(coherency residency budget)", file=sys.stderr)
+        P = cap
+        args.intervals = cap
     if P > 1:
         clist = [(cid, P * nchunk, names) for cid, nchunk, names in clist]
     nshap = getattr(args, 'shapelet_dirs', 0)
