@@ -187,17 +187,26 @@ class ConsensusADMM:
             (st.J - BZ_old)
         if self.Yhat_prev is not None:
             M = st.M
-            dY = torch.zeros(M, 8 * st.N, dtype=torch.float64)
-            dJ = torch.zeros(M, 8 * st.N, dtype=torch.float64)
+            # inner products over the CONCATENATED per-cluster chunk
+            # deltas (rho_bb_threadfn walks 8*N*nchunk contiguous values
+            # per cluster, consensus_poly.c:934-977)
+            dYt = torch.view_as_real(Yhat - self.Yhat_prev).reshape(
+                st.Mt, -1).double().cpu()
+            dJt = torch.view_as_real(st.J - self.J_prev).reshape(
+                st.Mt, -1).double().cpu()
+            ip11c = (dYt * dYt).sum(dim=1)
+            ip12c = (dYt * dJt).sum(dim=1)
+            ip22c = (dJt * dJt).sum(dim=1)
+            ip11 = torch.zeros(M, dtype=torch.float64)
+            ip12 = torch.zeros(M, dtype=torch.float64)
+            ip22 = torch.zeros(M, dtype=torch.float64)
             for ci in range(M):
                 o, nc = st.chunk_off[ci], int(st.nchunks[ci])
-                dy = (Yhat[o:o + nc] - self.Yhat_prev[o:o + nc]
-                      ).mean(dim=0).cpu()
-                dj = (st.J[o:o + nc] - self.J_prev[o:o + nc]
-                      ).mean(dim=0).cpu()
-                dY[ci] = torch.view_as_real(dy).reshape(-1).double()
-                dJ[ci] = torch.view_as_real(dj).reshape(-1).double()
-            new_rho = poly.update_rho_bb(self.rho, self.rho_upper, dY, dJ)
+                ip11[ci] = ip11c[o:o + nc].sum()
+                ip12[ci] = ip12c[o:o + nc].sum()
+                ip22[ci] = ip22c[o:o + nc].sum()
+            new_rho = poly.update_rho_bb_ip(self.rho, self.rho_upper,
+                                            ip11, ip12, ip22)
             # keep rho consistent across ranks (mean); reduce on the
             # backend's device (NCCL: cuda)
             if self.world > 1 and dist.is_initialized():

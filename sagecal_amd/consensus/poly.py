@@ -84,21 +84,30 @@ def eval_poly_jones(Z, Bf):
     return torch.einsum('p,mpnij->mnij', w.to(Z.dtype), Z)
 
 
-def update_rho_bb(rho, rho_upper, dY, dJ, eps=1e-12, alphacorrmin=0.2):
-    """Barzilai-Borwein penalty update per cluster
-    (consensus_poly.c:860-911): from dual delta dY = Yhat - Yhat_prev and
-    primal delta dJ = J - J_prev (flattened per cluster).
-
-    rho: [M]; dY/dJ: [M, K] reals. Returns updated rho."""
-    ip12 = (dY * dJ).sum(dim=1)
-    ip11 = (dY * dY).sum(dim=1)
-    ip22 = (dJ * dJ).sum(dim=1)
+def update_rho_bb_ip(rho, rho_upper, ip11, ip12, ip22, eps=1e-12,
+                     alphacorrmin=0.2):
+    """Barzilai-Borwein penalty update per cluster from the inner
+    products of the dual delta dY = Yhat - Yhat_prev and primal delta
+    dJ = J - J_prev (rho_bb_threadfn, consensus_poly.c:860-926,
+    oracle-verified): alphaMG if 2 alphaMG > alphaSD else
+    alphaSD - alphaMG/2; rho := alphahat when the deltas correlate
+    (corr > 0.2) and 1e-3 < alphahat < rho_upper."""
+    rho_upper = torch.as_tensor(rho_upper, dtype=rho.dtype).expand_as(rho)
     corr = ip12 / torch.sqrt((ip11 * ip22).clamp_min(eps))
     alphaSD = ip11 / ip12.clamp_min(eps)
     alphaMG = ip12 / ip22.clamp_min(eps)
     alphahat = torch.where(2.0 * alphaMG > alphaSD, alphaMG,
                            alphaSD - 0.5 * alphaMG)
-    rho_new = (1.0 / alphahat.clamp_min(eps)).clamp(max=rho_upper)
-    ok = (ip12 > eps) & (ip11 > eps) & (ip22 > eps) & (corr > alphacorrmin) \
-        & torch.isfinite(rho_new)
-    return torch.where(ok, rho_new, rho)
+    ok = (ip12 > eps) & (ip11 > eps) & (ip22 > eps) \
+        & (corr > alphacorrmin) & (alphahat > 1e-3) \
+        & (alphahat < rho_upper) & torch.isfinite(alphahat)
+    return torch.where(ok, alphahat, rho)
+
+
+def update_rho_bb(rho, rho_upper, dY, dJ, eps=1e-12, alphacorrmin=0.2):
+    """update_rho_bb_ip on flattened per-cluster deltas [M, K]."""
+    ip12 = (dY * dJ).sum(dim=1)
+    ip11 = (dY * dY).sum(dim=1)
+    ip22 = (dJ * dJ).sum(dim=1)
+    return update_rho_bb_ip(rho, rho_upper, ip11, ip12, ip22, eps,
+                            alphacorrmin)

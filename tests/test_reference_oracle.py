@@ -260,3 +260,70 @@ def test_shapelet_contrib_matches_reference(oracle, tmp_path):
                                   0.0, False, beta, n0, modes).numpy()
     assert np.allclose(ours, ref, atol=1e-10 * np.abs(ref).max()), \
         np.abs(ours - ref).max()
+
+
+def test_consensus_poly_basis_matches_reference(oracle):
+    """setup_polynomials types 0/1/2 (monomial / normalized / Bernstein,
+    consensus_poly.c) vs consensus.poly.setup_polynomials."""
+    from sagecal_amd.consensus import poly
+    freqs = np.array([115e6, 125e6, 135e6, 152e6, 170e6])
+    f0 = 140e6
+    for ty in (0, 1, 2):
+        pts = '\n'.join(f"{f:.17g}" for f in freqs)
+        out = subprocess.run(
+            [os.path.join(oracle, 'oracle_poly'), 'basis', '3',
+             str(len(freqs)), str(ty), str(f0)],
+            input=pts, capture_output=True, text=True, check=True).stdout
+        ref = np.array([[float(x) for x in ln.split()]
+                        for ln in out.strip().splitlines()])
+        ours = poly.setup_polynomials(freqs, f0, 3, ty).numpy()
+        assert np.allclose(ours, ref, atol=1e-12), (ty, ours - ref)
+
+
+def test_bb_rho_matches_reference(oracle):
+    """update_rho_bb (rho_bb_threadfn, consensus_poly.c:860-926) vs
+    consensus.poly.update_rho_bb_ip on hybrid-chunk clusters — the
+    alphaSD/alphaMG selection, correlation gate, accept window and the
+    CONCATENATED per-cluster chunk accumulation."""
+    from sagecal_amd.consensus import poly
+    rng = np.random.default_rng(6)
+    N, M = 4, 3
+    nchunks = [1, 2, 3]
+    Mt = sum(nchunks)
+    L = 8 * N * Mt
+    rho0 = np.array([5.0, 2.0, 7.0])
+    # correlated deltas for clusters 0/2, anti-correlated for 1
+    dJ = rng.standard_normal(L)
+    dY = 0.8 * dJ + 0.1 * rng.standard_normal(L)
+    off = 8 * N * nchunks[0]
+    ln1 = 8 * N * nchunks[1]
+    dY[off:off + ln1] = -dJ[off:off + ln1]
+    Yh0 = rng.standard_normal(L)
+    J0 = rng.standard_normal(L)
+    Yh = Yh0 + dY
+    J = J0 + dJ
+    inp = '\n'.join(f"{rho0[ci]:.17g} {nchunks[ci]}" for ci in range(M))
+    for buf in (Yh, Yh0, J, J0):
+        inp += '\n' + '\n'.join(f"{x:.17g}" for x in buf)
+    out = subprocess.run(
+        [os.path.join(oracle, 'oracle_poly'), 'rhobb', str(N), str(M),
+         '1000.0'], input=inp, capture_output=True, text=True,
+        check=True).stdout.split()
+    ref = np.array([float(x) for x in out])
+    # ours: segment-summed inner products per cluster
+    ip11 = np.zeros(M)
+    ip12 = np.zeros(M)
+    ip22 = np.zeros(M)
+    o = 0
+    for ci in range(M):
+        ln = 8 * N * nchunks[ci]
+        ip11[ci] = (dY[o:o + ln] ** 2).sum()
+        ip12[ci] = (dY[o:o + ln] * dJ[o:o + ln]).sum()
+        ip22[ci] = (dJ[o:o + ln] ** 2).sum()
+        o += ln
+    ours = poly.update_rho_bb_ip(
+        torch.tensor(rho0), 1000.0, torch.tensor(ip11),
+        torch.tensor(ip12), torch.tensor(ip22)).numpy()
+    assert np.allclose(ours, ref, atol=1e-10), (ours, ref)
+    # anti-correlated cluster kept its old rho
+    assert ours[1] == rho0[1]

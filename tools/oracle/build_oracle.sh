@@ -24,7 +24,10 @@ gcc $CFLAGS -c "$REF/src/lib/Radio/shapelet.c" -o "$OUT/shapelet.o"
 gcc $CFLAGS "$HERE/oracle_element.c" "$OUT/elementbeam.o" \
     "$OUT/myblas.o" "$OUT/miniblas.o" -lpthread -lm \
     -o "$OUT/oracle_element"
+gcc $CFLAGS -c "$REF/src/lib/Dirac/consensus_poly.c" -o "$OUT/consensus_poly.o"
+gcc $CFLAGS "$HERE/oracle_poly.c" "$OUT/consensus_poly.o" \
+    "$OUT/myblas.o" "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_poly"
 gcc $CFLAGS "$HERE/oracle_misc.c" "$OUT/transforms.o" "$OUT/updatenu.o" \
     "$OUT/shapelet.o" "$OUT/elementbeam.o" "$OUT/myblas.o" \
     "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_misc"
-echo "built: $OUT/oracle_lbfgs $OUT/oracle_predict $OUT/oracle_element $OUT/oracle_misc"
+echo "built: $OUT/oracle_lbfgs $OUT/oracle_predict $OUT/oracle_element $OUT/oracle_misc $OUT/oracle_poly"
