@@ -327,3 +327,33 @@ def test_bb_rho_matches_reference(oracle):
     assert np.allclose(ours, ref, atol=1e-10), (ours, ref)
     # anti-correlated cluster kept its old rho
     assert ours[1] == rho0[1]
+
+
+def test_find_prod_inverse_matches_reference(oracle):
+    """find_prod_inverse_full (per-cluster Npoly x Npoly pseudo-inverse
+    of sum_f rho_fk B_f B_f^T, consensus_poly.c:465 + prod_inv_threadfn)
+    vs consensus.poly.find_prod_inverse — including a rank-deficient
+    case (Npoly > Nf) where the pinv cutoff semantics matter."""
+    from sagecal_amd.consensus import poly
+    rng = np.random.default_rng(8)
+    for Npoly, Nf, M in ((3, 5, 2), (3, 2, 2)):   # full rank + deficient
+        freqs = 120e6 + 1e7 * np.arange(Nf)
+        f0 = float(np.mean(freqs))
+        rho_fk = rng.uniform(0.5, 4.0, size=(M, Nf))   # [M, Nf]
+        inp = '\n'.join(f"{f:.17g}" for f in freqs)
+        # reference layout rho[k + f*M]
+        for f in range(Nf):
+            for k in range(M):
+                inp += '\n' + f"{rho_fk[k, f]:.17g}"
+        out = subprocess.run(
+            [os.path.join(oracle, 'oracle_poly'), 'bii', str(Npoly),
+             str(Nf), str(M), '0', str(f0)],
+            input=inp, capture_output=True, text=True, check=True).stdout
+        vals = np.array([[float(x) for x in ln.split()]
+                         for ln in out.strip().splitlines()])
+        ref = vals.reshape(M, Npoly, Npoly)
+        B = poly.setup_polynomials(freqs, f0, Npoly, 0)
+        ours = poly.find_prod_inverse(
+            B, torch.tensor(rho_fk)).numpy()
+        assert np.allclose(ours, ref, atol=1e-8), \
+            (Npoly, Nf, np.abs(ours - ref).max())

@@ -136,10 +136,77 @@ void dgemm_(char *ta, char *tb, int *m, int *n, int *k, double *alpha,
 }
 /* weak: a real implementation linked alongside (e.g. shapelet.o for
  * oracle_misc) overrides the stub */
+/* small-matrix real SVD via one-sided Jacobi (column-major, m==n only;
+ * enough for the Npoly x Npoly pseudo-inverses of consensus_poly.c) */
+int dgesvd_(char *jobu, char *jobvt, int *m_, int *n_, double *a,
+            int *lda_, double *s, double *u, int *ldu_, double *vt,
+            int *ldvt_, double *work, int *lwork_, int *info) {
+  int m = *m_, n = *n_, lda = *lda_, ldu = *ldu_, ldvt = *ldvt_;
+  *info = 0;
+  if (*lwork_ == -1) { work[0] = 64; return 0; }
+  if (m != n || n > 32) { *info = -1; return 0; }
+  double V[32 * 32];
+  for (int i = 0; i < n; i++)
+    for (int j = 0; j < n; j++) V[i + j * n] = (i == j) ? 1.0 : 0.0;
+  for (int sweep = 0; sweep < 60; sweep++) {
+    double off = 0.0;
+    for (int p = 0; p < n - 1; p++)
+      for (int q = p + 1; q < n; q++) {
+        double app = 0, aqq = 0, apq = 0;
+        for (int i = 0; i < m; i++) {
+          app += a[i + p * lda] * a[i + p * lda];
+          aqq += a[i + q * lda] * a[i + q * lda];
+          apq += a[i + p * lda] * a[i + q * lda];
+        }
+        off += apq * apq;
+        if (fabs(apq) < 1e-300) continue;
+        double tau = (aqq - app) / (2.0 * apq);
+        double t = (tau >= 0 ? 1.0 : -1.0)
+                   / (fabs(tau) + sqrt(1.0 + tau * tau));
+        double c = 1.0 / sqrt(1.0 + t * t), sn = c * t;
+        for (int i = 0; i < m; i++) {
+          double x = a[i + p * lda], y = a[i + q * lda];
+          a[i + p * lda] = c * x - sn * y;
+          a[i + q * lda] = sn * x + c * y;
+        }
+        for (int i = 0; i < n; i++) {
+          double x = V[i + p * n], y = V[i + q * n];
+          V[i + p * n] = c * x - sn * y;
+          V[i + q * n] = sn * x + c * y;
+        }
+      }
+    if (off < 1e-30) break;
+  }
+  /* singular values + U columns; sort descending */
+  int idx[32];
+  for (int j = 0; j < n; j++) {
+    double nn = 0;
+    for (int i = 0; i < m; i++) nn += a[i + j * lda] * a[i + j * lda];
+    s[j] = sqrt(nn);
+    idx[j] = j;
+  }
+  for (int i = 0; i < n - 1; i++)
+    for (int j = i + 1; j < n; j++)
+      if (s[idx[j]] > s[idx[i]]) { int t2 = idx[i]; idx[i] = idx[j];
+                                   idx[j] = t2; }
+  double stmp[32];
+  for (int k = 0; k < n; k++) stmp[k] = s[idx[k]];
+  for (int k = 0; k < n; k++) {
+    int j = idx[k];
+    double inv = stmp[k] > 1e-300 ? 1.0 / stmp[k] : 0.0;
+    for (int i = 0; i < m; i++) u[i + k * ldu] = a[i + j * lda] * inv;
+    /* VT row k = V column j transposed */
+    for (int i = 0; i < n; i++) vt[k + i * ldvt] = V[i + j * n];
+  }
+  for (int k = 0; k < n; k++) s[k] = stmp[k];
+  (void)jobu; (void)jobvt;
+  return 0;
+}
+
 #define STUB(name) __attribute__((weak)) void name() { \
   fprintf(stderr, "miniblas: " #name " not implemented (oracle)\n"); \
   abort(); }
-STUB(cgels_) STUB(dgels_) STUB(dgeqrf_) STUB(dgesvd_) STUB(dorgqr_)
+STUB(cgels_) STUB(dgels_) STUB(dgeqrf_) STUB(dorgqr_)
 STUB(dpotrf_) STUB(dpotrs_) STUB(dsyevx_) STUB(dtrtrs_) STUB(zgels_)
 STUB(zgemm_) STUB(zgesvd_) STUB(zher_)
 /* predict.c references shapelet_contrib (shapelet.c) — oracle covers
