@@ -1,25 +1,29 @@
 """Solution-file I/O, byte-compatible with the reference format.
 
-Format (reference README.md §6 "Solution format"; writer
-fullbatch_mode.cpp:285-289,596-605):
+Format (writer fullbatch_mode.cpp:285-289,596-605):
   - '#' comment lines;
   - first non-comment line: freq(MHz) bandwidth(MHz) time_interval(min)
     stations clusters effective_clusters;
   - per solution interval, 8N rows: counter column then one column per
     effective cluster chunk, clusters written in REVERSE order
     (for ci=M-1..0), chunks in forward order (fullbatch_mode.cpp:598).
-  - per station, the 8 values S0..S7 encode the 2x2 Jones COLUMN-major:
-    J = [S0+jS1, S4+jS5; S2+jS3, S6+jS7].
+  - per station, the 8 values S0..S7 are the ROW-major 2x2 Jones with
+    interleaved re/im: J = [S0+jS1, S2+jS3; S4+jS5, S6+jS7].
 
-Internally sagecal_amd stores Jones row-major ([J00,J01,J10,J11]
-interleaved re/im); this module permutes on read/write.
+NOTE: README.md §6 describes the per-station order as COLUMN-major
+(J=[S0+jS1, S4+jS5; S2+jS3, S6+jS7]) — that contradicts the code. The
+writer dumps the in-memory p[] directly (fullbatch_mode.cpp:600), and
+every consumer of p[] builds the Jones ROW-major (residual.c:92-99 G1
+with the row-major amb() product :33). Oracle-verified
+(tests/test_reference_oracle.py::test_residuals_multifreq_matches_
+reference); we follow the implementation, not the README.
 """
 import numpy as np
 import torch
 
-# internal row-major interleaved index of each reference row s0..s7:
-# ref order [00r,00i,10r,10i,01r,01i,11r,11i] -> internal positions
-_REF_TO_INT = np.array([0, 1, 4, 5, 2, 3, 6, 7])
+# reference per-station order == our internal row-major interleaved
+# order (identity permutation; kept for documentation/symmetry)
+_REF_TO_INT = np.array([0, 1, 2, 3, 4, 5, 6, 7])
 
 
 def jones_to_ref_vec(J):

@@ -357,3 +357,109 @@ def test_find_prod_inverse_matches_reference(oracle):
             B, torch.tensor(rho_fk)).numpy()
         assert np.allclose(ours, ref, atol=1e-8), \
             (Npoly, Nf, np.abs(ours - ref).max())
+
+
+def test_residuals_multifreq_matches_reference(oracle, tmp_path):
+    """calculate_residuals_multifreq (residual.c:940): per-channel
+    re-predict with the SOLUTIONS applied and spectral-index flux
+    scaling, against our predict+apply path — this also pins the
+    reference's in-memory 8-reals-per-station parameter ordering to
+    solutions.jones_to_ref_vec (the solution-file layout)."""
+    from sagecal_amd import sky, solutions
+    from sagecal_amd.ops.reference import SourcePack, predict_coh, \
+        apply_jones
+
+    rng = np.random.default_rng(31)
+    N, tilesz, Nchan = 6, 2, 2
+    pairs = np.array([(p, q) for p in range(N) for q in range(p + 1, N)])
+    Nbase = len(pairs)
+    rows = Nbase * tilesz
+    freqs = np.array([140e6, 160e6])
+    freq0 = 150e6
+    fdelta, tdelta, dec0 = 10e6, 0.0, np.pi / 4
+
+    srcs = {}
+    clist = []
+    for ci in range(2):
+        names = []
+        for si in range(2):
+            nm = f's{ci}_{si}'
+            srcs[nm] = sky.Source(
+                name=nm, ra=0.02 * (ci + 1) + 0.005 * si,
+                dec=dec0 + 0.01 * ci - 0.004 * si,
+                sI=1.0 + 0.3 * ci + 0.1 * si, sQ=0.05, sU=0.02,
+                sV=0.01, spec_idx=-0.7 + 0.2 * ci, f0=148e6,
+                stype=1 if si else 0, eX=2e-3 if si else 0.0,
+                eY=1e-3 if si else 0.0, eP=0.3 if si else 0.0)
+            names.append(nm)
+        clist.append((ci, 1, names))
+    clusters = sky.build_clusters(srcs, clist, 0.0, dec0, freq0)
+    pack = SourcePack(clusters)
+    M = pack.M
+
+    u = torch.tensor(rng.standard_normal(rows) * 300.0 / 3e8)
+    v = torch.tensor(rng.standard_normal(rows) * 300.0 / 3e8)
+    w = torch.tensor(rng.standard_normal(rows) * 30.0 / 3e8)
+    bb = torch.tensor(np.tile(pairs, (tilesz, 1)))
+    J = torch.tensor(np.eye(2)[None, None] + 0.25 * (
+        rng.standard_normal((M, N, 2, 2))
+        + 1j * rng.standard_normal((M, N, 2, 2))))
+    x = torch.tensor(rng.standard_normal((Nchan, rows, 2, 2))
+                     + 1j * rng.standard_normal((Nchan, rows, 2, 2)))
+
+    # ---- layout file for the oracle ----
+    L = [f"{N} {Nbase} {tilesz} {M} {Nchan} {fdelta} {tdelta} {dec0} "
+         f"-1 0.0"]
+    L += [f"{f:.17g}" for f in freqs]
+    for b in range(rows):
+        L.append(f"{int(bb[b, 0])} {int(bb[b, 1])} {float(u[b]):.17g} "
+                 f"{float(v[b]):.17g} {float(w[b]):.17g}")
+    for ci in range(M):
+        s0, s1 = int(pack.cluster_off[ci]), int(pack.cluster_off[ci + 1])
+        L.append(f"{ci} {s1 - s0}")
+        for s in range(s0, s1):
+            L.append(
+                f"{int(pack.stype[s])} {float(pack.ll[s]):.17g} "
+                f"{float(pack.mm[s]):.17g} {float(pack.nn1[s]):.17g} "
+                f"{float(pack.sI[s]):.17g} {float(pack.sQ[s]):.17g} "
+                f"{float(pack.sU[s]):.17g} {float(pack.sV[s]):.17g} "
+                f"{float(pack.eX[s]):.17g} {float(pack.eY[s]):.17g} "
+                f"{float(pack.eP[s]):.17g} {float(pack.cxi[s]):.17g} "
+                f"{float(pack.sxi[s]):.17g} {float(pack.cphi[s]):.17g} "
+                f"{float(pack.sphi[s]):.17g} {int(pack.use_proj[s])} "
+                f"{float(pack.f0[s]):.17g} "
+                f"{float(pack.spec_idx[s]):.17g} "
+                f"{float(pack.spec_idx1[s]):.17g} "
+                f"{float(pack.spec_idx2[s]):.17g} "
+                f"{float(pack.sI0[s]):.17g} {float(pack.sQ0[s]):.17g} "
+                f"{float(pack.sU0[s]):.17g} {float(pack.sV0[s]):.17g}")
+    for ci in range(M):
+        pvec = solutions.jones_to_ref_vec(J[ci]).numpy()
+        L += [f"{val:.17g}" for val in pvec]
+    # x: XX(re,im),XY,YX,YY per baseline, per timeslot, per channel
+    for fi in range(Nchan):
+        xf = torch.view_as_real(x[fi]).reshape(rows, 8).numpy()
+        for b in range(rows):
+            L += [f"{val:.17g}" for val in xf[b]]
+    layout = tmp_path / 'res_layout.txt'
+    layout.write_text('\n'.join(L) + '\n')
+
+    out = subprocess.run([os.path.join(oracle, 'oracle_residual'),
+                          str(layout)], capture_output=True, text=True,
+                         check=True).stdout.split()
+    vals = np.array([float(t) for t in out]).reshape(Nchan, rows, 2, 2,
+                                                     2)
+    ref = torch.view_as_complex(torch.tensor(vals))
+
+    # ours: x - sum_ci J_ci C_ci(f) J_ci^H per channel (per-channel
+    # flux scaling handled inside predict_coh via freq != freq0)
+    fdelta_ch = fdelta / Nchan
+    ours = x.clone()
+    for fi in range(Nchan):
+        coh = predict_coh(pack, u, v, w, float(freqs[fi]), freq0,
+                          fdelta_ch, tdelta, dec0)
+        for ci in range(M):
+            ours[fi] -= apply_jones(coh[ci], J[ci:ci + 1], bb)
+    scale = float(ref.abs().max())
+    err = float((ours - ref).abs().max()) / scale
+    assert err < 1e-10, f"residual mismatch: rel err {err}"

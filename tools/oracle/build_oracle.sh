@@ -25,9 +25,19 @@ gcc $CFLAGS "$HERE/oracle_element.c" "$OUT/elementbeam.o" \
     "$OUT/myblas.o" "$OUT/miniblas.o" -lpthread -lm \
     -o "$OUT/oracle_element"
 gcc $CFLAGS -c "$REF/src/lib/Dirac/consensus_poly.c" -o "$OUT/consensus_poly.o"
+gcc $CFLAGS -c "$REF/src/lib/Radio/residual.c" -o "$OUT/residual.o"
+gcc $CFLAGS -c "$REF/src/lib/Dirac/manifold_average.c" -o "$OUT/manifold_average.o" 2>/dev/null || true
 gcc $CFLAGS "$HERE/oracle_poly.c" "$OUT/consensus_poly.o" \
     "$OUT/myblas.o" "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_poly"
+gcc $CFLAGS "$HERE/oracle_residual.c" "$OUT/residual.o" \
+    "$OUT/predict.o" "$OUT/shapelet.o" "$OUT/elementbeam.o" \
+    "$OUT/manifold_average.o" "$OUT/myblas.o" "$OUT/miniblas.o" \
+    -lpthread -lm -o "$OUT/oracle_residual" 2>/dev/null || \
+  gcc $CFLAGS "$HERE/oracle_residual.c" "$OUT/residual.o" \
+    "$OUT/predict.o" "$OUT/shapelet.o" "$OUT/elementbeam.o" \
+    "$OUT/myblas.o" "$OUT/miniblas.o" \
+    -lpthread -lm -o "$OUT/oracle_residual"
 gcc $CFLAGS "$HERE/oracle_misc.c" "$OUT/transforms.o" "$OUT/updatenu.o" \
     "$OUT/shapelet.o" "$OUT/elementbeam.o" "$OUT/myblas.o" \
     "$OUT/miniblas.o" -lpthread -lm -o "$OUT/oracle_misc"
-echo "built: $OUT/oracle_lbfgs $OUT/oracle_predict $OUT/oracle_element $OUT/oracle_misc $OUT/oracle_poly"
+echo "built: $OUT/oracle_lbfgs $OUT/oracle_predict $OUT/oracle_element $OUT/oracle_misc $OUT/oracle_poly $OUT/oracle_residual"
