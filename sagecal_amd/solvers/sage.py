@@ -475,12 +475,20 @@ def calculate_residuals_multifreq(state, pack, tile, bb, ccid=None, rho=0.0,
 
 
 def _mmse_inv(J, rho):
-    """MMSE-regularized left-inverse: (J^H J + rho I)^-1 J^H
-    (residual.c correction with robust rho; plain inverse at rho=0)."""
-    JhJ = J.conj().transpose(-1, -2) @ J
+    """Robust correction inverse, matching the reference's mat_invert
+    (residual.c:163-196) exactly: the direct 2x2 inverse of (J + rho I)
+    with a determinant guard (det += rho when sqrt(|det|) <= rho) —
+    oracle-verified; NOT a true MMSE left-inverse."""
     eye = torch.eye(2, dtype=J.dtype, device=J.device)
-    A = JhJ + rho * eye
-    return torch.linalg.solve(A, J.conj().transpose(-1, -2))
+    A = J + rho * eye
+    det = A[..., 0, 0] * A[..., 1, 1] - A[..., 0, 1] * A[..., 1, 0]
+    det = torch.where(det.abs().sqrt() <= rho, det + rho, det)
+    inv = torch.empty_like(A)
+    inv[..., 0, 0] = A[..., 1, 1]
+    inv[..., 0, 1] = -A[..., 0, 1]
+    inv[..., 1, 0] = -A[..., 1, 0]
+    inv[..., 1, 1] = A[..., 0, 0]
+    return inv / det[..., None, None]
 
 
 def correct_residuals(xres, Jp, Jq, rho=0.0):

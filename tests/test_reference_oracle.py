@@ -463,3 +463,73 @@ def test_residuals_multifreq_matches_reference(oracle, tmp_path):
     scale = float(ref.abs().max())
     err = float((ours - ref).abs().max()) / scale
     assert err < 1e-10, f"residual mismatch: rel err {err}"
+
+
+def test_residual_correction_matches_reference(oracle, tmp_path):
+    """The ccid MMSE-style correction (residual.c:125-196 mat_invert +
+    G1 x G2^H application, incl. the det guard) vs
+    sage.correct_residuals."""
+    from sagecal_amd import sky, solutions
+    from sagecal_amd.ops.reference import SourcePack, predict_coh, \
+        apply_jones
+    from sagecal_amd.solvers import sage as sage_mod
+
+    rng = np.random.default_rng(41)
+    N, tilesz, Nchan = 5, 1, 1
+    pairs = np.array([(p, q) for p in range(N) for q in range(p + 1, N)])
+    Nbase = len(pairs)
+    rows = Nbase * tilesz
+    freqs = np.array([150e6])
+    freq0, fdelta, tdelta, dec0 = 150e6, 5e6, 0.0, np.pi / 4
+    rho_c = 0.3
+
+    srcs = {'a': sky.Source(name='a', ra=0.02, dec=dec0 + 0.01, sI=1.2,
+                            sQ=0.0, sU=0.0, sV=0.0, f0=freq0, stype=0)}
+    clusters = sky.build_clusters(srcs, [(7, 1, ['a'])], 0.0, dec0,
+                                  freq0)
+    pack = SourcePack(clusters)
+
+    u = torch.tensor(rng.standard_normal(rows) * 300.0 / 3e8)
+    v = torch.tensor(rng.standard_normal(rows) * 300.0 / 3e8)
+    w = torch.tensor(rng.standard_normal(rows) * 30.0 / 3e8)
+    bb = torch.tensor(np.tile(pairs, (tilesz, 1)))
+    J = torch.tensor(np.eye(2)[None, None] + 0.3 * (
+        rng.standard_normal((1, N, 2, 2))
+        + 1j * rng.standard_normal((1, N, 2, 2))))
+    x = torch.tensor(rng.standard_normal((Nchan, rows, 2, 2))
+                     + 1j * rng.standard_normal((Nchan, rows, 2, 2)))
+
+    L = [f"{N} {Nbase} {tilesz} 1 {Nchan} {fdelta} {tdelta} {dec0} "
+         f"7 {rho_c}"]
+    L += [f"{f:.17g}" for f in freqs]
+    for b in range(rows):
+        L.append(f"{int(bb[b, 0])} {int(bb[b, 1])} {float(u[b]):.17g} "
+                 f"{float(v[b]):.17g} {float(w[b]):.17g}")
+    L.append("7 1")
+    s = 0
+    L.append(
+        f"{int(pack.stype[s])} {float(pack.ll[s]):.17g} "
+        f"{float(pack.mm[s]):.17g} {float(pack.nn1[s]):.17g} "
+        f"{float(pack.sI[s]):.17g} 0 0 0 0 0 0 1 0 1 0 0 "
+        f"{float(pack.f0[s]):.17g} 0 0 0 {float(pack.sI0[s]):.17g} "
+        f"0 0 0")
+    L += [f"{val:.17g}"
+          for val in solutions.jones_to_ref_vec(J[0]).numpy()]
+    xf = torch.view_as_real(x[0]).reshape(rows, 8).numpy()
+    for b in range(rows):
+        L += [f"{val:.17g}" for val in xf[b]]
+    layout = tmp_path / 'corr_layout.txt'
+    layout.write_text('\n'.join(L) + '\n')
+    out = subprocess.run([os.path.join(oracle, 'oracle_residual'),
+                          str(layout)], capture_output=True, text=True,
+                         check=True).stdout.split()
+    ref = torch.view_as_complex(torch.tensor(
+        np.array([float(t) for t in out]).reshape(Nchan, rows, 2, 2,
+                                                  2)))
+
+    coh = predict_coh(pack, u, v, w, freq0, freq0, fdelta, tdelta, dec0)
+    res = x[0] - apply_jones(coh[0], J, bb)
+    corr = sage_mod.correct_residuals(
+        res.unsqueeze(0), J[0, bb[:, 0]], J[0, bb[:, 1]], rho=rho_c)
+    err = float((corr[0] - ref[0]).abs().max() / ref.abs().max())
+    assert err < 1e-10, f"correction mismatch: rel err {err}"
