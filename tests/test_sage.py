@@ -235,3 +235,37 @@ def test_interval_batching_exact_equivalence():
             Jb = state.J[state.chunk_off[ci] + k]
             Js = st.J[st.chunk_off[ci]]
             assert torch.allclose(Jb, Js, atol=1e-10), (ci, k)
+
+
+def test_sagefit_with_flagged_rows_and_all_flagged():
+    """Flag handling (preset_flags_and_data semantics): flagged rows
+    contribute nothing; a fully-flagged tile must not crash or produce
+    NaN (the reference zeroes flagged data+coh, baseline_utils.c)."""
+    import bench as bench_mod
+    from sagecal_amd.solvers import sage as sage_mod
+    from sagecal_amd.constants import SM_RLM_RLBFGS
+
+    class A:
+        pass
+    a = A()
+    a.__dict__.update(stations=8, dirs=2, srcs=2, tilesz=4, chan=2,
+                      freq0=150e6, bandwidth=180e3, intervals=1,
+                      shapelet_dirs=0)
+    pack, ms, tile, bb = bench_mod.build_problem(a, 'cpu', torch.float64)
+    opts = sage_mod.SageSolveOptions(max_emiter=2, max_iter=10,
+                                     solver_mode=SM_RLM_RLBFGS,
+                                     robust_outer=1, em_group=2,
+                                     lbfgs_iters=10)
+    cohs = sage_mod.precalc_coherencies(pack, tile)
+    # half the rows flagged: solve converges on the remaining half
+    flags = torch.zeros(tile.x.shape[0], dtype=torch.bool)
+    flags[::2] = True
+    st = sage_mod.CalState(pack, a.stations)
+    r0, r1 = sage_mod.sagefit(st, cohs, tile, bb, opts, flags=flags)
+    assert np.isfinite(r1) and r1 < 0.5 * r0
+    # all rows flagged: no data — must return finite (zero-ish) residual
+    st2 = sage_mod.CalState(pack, a.stations)
+    r0a, r1a = sage_mod.sagefit(st2, cohs, tile, bb, opts,
+                                flags=torch.ones_like(flags))
+    assert np.isfinite(r0a) and np.isfinite(r1a)
+    assert torch.isfinite(torch.view_as_real(st2.J)).all()
