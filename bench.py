@@ -69,8 +69,8 @@ def build_problem(args, device, dtype, rank=0, world=1):
     # synthetic truth is generated in fp64; for large arrays generate on
     # the GPU (fp64 on device) — CPU generation at 512 stations takes
     # minutes and starves the bench setup
-    gen_dev = device if (device != 'cpu'
-                         and args.stations * args.tilesz * P >= 2**21)         else 'cpu'
+    nb_rows = nb * args.tilesz * P
+    gen_dev = device if (device != 'cpu' and nb_rows >= 2**21) else 'cpu'
     if gen_dev != 'cpu':
         pack.to(gen_dev)
     ms = msdata.SyntheticMS(
@@ -145,7 +145,8 @@ def run_bandpass(args, device, dtype, rank, world):
     cal = MinibatchConsensusCalibration(
         pack, args.stations, ms.freqs, nsolbw=args.nsolbw, Npoly=2,
         rho=1.0, device=device, dtype=cdt,
-        fed_alpha=0.1 if world > 1 else 0.0, world=world, rank=rank)
+        fed_alpha=0.1 if world > 1 else 0.0, world=world, rank=rank,
+        multifreq=not args.band_average)
     def step():
         cal.epoch(tile, bb, nmb=2, lbfgs_iters=6, robust_nu=10.0)
     return step, tile
@@ -181,6 +182,11 @@ def main():
                     help='bandpass: stochastic-LBFGS 256-chan consensus '
                          '(BASELINE config 4)')
     ap.add_argument('--nsolbw', type=int, default=8)
+    ap.add_argument('--band-average', action='store_true',
+                    help='bandpass mode: fit each mini-band to its '
+                         'channel AVERAGE instead of the reference-'
+                         'faithful per-channel gradient '
+                         '(robust_batchmode_lbfgs.c:942 loops channels)')
     ap.add_argument('--shapelet-dirs', type=int, default=0,
                     help='make this many clusters shapelet (config 5)')
     ap.add_argument('--admm-rho', type=float, default=5.0)
@@ -253,6 +259,9 @@ def main():
                 'config': {'model': f'{args.stations}-station stochastic-'
                            f'LBFGS bandpass, {args.chan} chan, '
                            f'{args.nsolbw} mini-bands',
+                           'gradient': ('band-average'
+                                        if args.band_average
+                                        else 'per-channel'),
                            'global_batch': vis_per_step,
                            'seq_len': args.tilesz,
                            'parallelism': f'federated dp{world}'}}))
