@@ -26,3 +26,8 @@ timeout 200 rocprofv3 --kernel-trace --stats --output-format csv \
     python /root/repo/bench.py --steps 1 --warmup 1 >/tmp/p.log 2>&1
 tail -2 /tmp/p.log
 ls /root/repo/gpurun_out/prof 2>/dev/null | head
+# 4) config-5 bench (generation now on GPU after the threshold fix)
+cd /root/repo
+timeout 300 python bench.py --stations 512 --dirs 20 --solver rtr \
+    --shapelet-dirs 2 --intervals 1 --emiter 2 --steps 2 --warmup 1 \
+    2>/dev/null | tee gpurun_out/bench_config5.json | tail -1
