@@ -49,6 +49,15 @@ def build_argparser():
                          '3 element; 4/5/6 same but per channel '
                          '(MS must carry element_enu; predict_withbeam.c '
                          '/ DOBEAM_* + _WB modes)')
+    ap.add_argument('--elem-type', dest='elem_type', default='auto',
+                    choices=['auto', 'synthetic', 'lba', 'hba', 'alo'],
+                    help='element dipole model for -B 2/3/5/6: the real '
+                         'LOFAR LBA/HBA or ALO coefficient tables '
+                         '(elementcoeff.h port) or the synthetic '
+                         'pattern; auto reads the MS key `elem_type`, '
+                         'falling back to synthetic (the reference '
+                         'reads this from LOFAR_ANTENNA_FIELD, '
+                         'data.cpp:268-288)')
     ap.add_argument('-W', dest='whiten', type=int, default=0,
                     help='1: pre-whiten data with the NCP uv taper '
                          '(whiten_data, updatenu.c)')
@@ -129,11 +138,15 @@ def _predict_with_beam(ms, pack, tile, ti, args):
     t0 = float(z['tmjd0']) if 'tmjd0' in z else 56789.0
     tmjd = t0 + (ti * T + np.arange(T) + 0.5) * ms.tdelta / 86400.0
     mode = args.dobeam
+    et = getattr(args, 'elem_type', 'auto')
+    if et == 'auto':
+        et = str(z['elem_type']) if 'elem_type' in z else 'synthetic'
+    coeffs = None if et == 'synthetic' else et   # name -> LOFAR tables
     if mode <= 3:
         return beams.predict_coh_withbeam(
             pack, tile.u, tile.v, tile.w, tile.freq0, tile.freq0,
             tile.fdelta, tile.tdelta, tile.dec0, cfg, tmjd,
-            ms.bb_tensor(), ms.Nbase, T, mode=mode)
+            ms.bb_tensor(), ms.Nbase, T, mode=mode, coeffs=coeffs)
     # -B 4/5/6: per-channel beam (DOBEAM_*_WB): evaluate the beam at each
     # channel frequency and average, like the discrete channel model
     mode -= 3
@@ -143,9 +156,35 @@ def _predict_with_beam(ms, pack, tile, ti, args):
         c = beams.predict_coh_withbeam(
             pack, tile.u, tile.v, tile.w, float(f), tile.freq0,
             fdelta_ch, tile.tdelta, tile.dec0, cfg, tmjd,
-            ms.bb_tensor(), ms.Nbase, T, mode=mode)
+            ms.bb_tensor(), ms.Nbase, T, mode=mode, coeffs=coeffs)
         acc = c if acc is None else acc + c
     return acc / len(tile.freqs)
+
+
+def _predict_channel_with_beam(ms, pack, tile, ti, args, freq,
+                               fdelta_ch):
+    """One channel's beamed coherencies for the residual path (-B with
+    multifreq residuals)."""
+    from .. import beams
+    z = getattr(ms, '_z', {})
+    lon = float(z['lon']) if 'lon' in z else 0.0
+    lat = float(z['lat']) if 'lat' in z else 0.92
+    cfg = beams.ArrayConfig(list(np.asarray(z['element_enu'])), lon, lat,
+                            ms.ra0, ms.dec0)
+    T = ms.tilesz
+    t0 = float(z['tmjd0']) if 'tmjd0' in z else 56789.0
+    tmjd = t0 + (ti * T + np.arange(T) + 0.5) * ms.tdelta / 86400.0
+    mode = args.dobeam
+    if mode > 3:
+        mode -= 3
+    et = getattr(args, 'elem_type', 'auto')
+    if et == 'auto':
+        et = str(z['elem_type']) if 'elem_type' in z else 'synthetic'
+    coeffs = None if et == 'synthetic' else et
+    return beams.predict_coh_withbeam(
+        pack, tile.u, tile.v, tile.w, freq, tile.freq0, fdelta_ch,
+        tile.tdelta, tile.dec0, cfg, tmjd, ms.bb_tensor(), ms.Nbase, T,
+        mode=mode, coeffs=coeffs)
 
 
 def uv_flags(tile, args):
@@ -203,8 +242,18 @@ def run_calibration(args):
                 nc = state.nchunks[match[0]]
                 Jc = state.J[o:o + nc]
                 state.J[o:o + nc] = Jc / Jc.abs().clamp_min(1e-12)
+        coh_fn = None
+        if args.dobeam:
+            # beamed residuals (calculate_residuals_multifreq_withbeam
+            # role): per-channel coherencies WITH the station beam, so
+            # the subtracted model matches what was calibrated against
+            fdch = tile.fdelta / len(tile.freqs)
+            coh_fn = (lambda f, _t=tile, _ti=ti:
+                      _predict_channel_with_beam(ms, pack, _t, _ti, args,
+                                                 f, fdch))
         xres = sage.calculate_residuals_multifreq(
-            state, pack, tile, bb, ccid=ccid, rho=args.rho_corr)
+            state, pack, tile, bb, ccid=ccid, rho=args.rho_corr,
+            coh_fn=coh_fn)
         if args.dochan:
             # per-channel refinement (-b 1, fullbatch_mode.cpp:453-499):
             # polish each channel's solutions with a short joint LBFGS and
@@ -278,9 +327,17 @@ def run_simulation(args):
         V = torch.zeros_like(tile.xo)
         fdelta_ch = tile.fdelta / len(tile.freqs)
         for fi, f in enumerate(tile.freqs):
-            cohs = ops.predict_coh(pack, tile.u, tile.v, tile.w, float(f),
-                                   tile.freq0, fdelta_ch, tile.tdelta,
-                                   tile.dec0)
+            if args.dobeam:
+                # predict WITH the station beam
+                # (predict_visibilities_multifreq_withbeam,
+                # fullbatch_mode.cpp:538-559)
+                cohs = _predict_channel_with_beam(ms, pack, tile, ti,
+                                                  args, float(f),
+                                                  fdelta_ch)
+            else:
+                cohs = ops.predict_coh(pack, tile.u, tile.v, tile.w,
+                                       float(f), tile.freq0, fdelta_ch,
+                                       tile.tdelta, tile.dec0)
             if cohs.dtype != cdtype:
                 cohs = cohs.to(cdtype)
             V[fi] = sage.total_model(state, cohs, bb, tile.tilesz,

@@ -436,14 +436,16 @@ def sagefit(state, cohs, tile, bb, opts, flags=None, admm_terms=None):
 
 
 def calculate_residuals_multifreq(state, pack, tile, bb, ccid=None, rho=0.0,
-                                  device=None):
+                                  device=None, coh_fn=None):
     """Per-channel residuals with the solved gains
     (residual.c calculate_residuals_multifreq:940): re-predict each channel's
     coherencies at its own frequency (spectral-index flux scaling + per
     channel smearing) and subtract J_p C J_q^H.
 
     Optionally correct residuals by the inverted solution of cluster `ccid`
-    (MMSE with robust rho: J (J^H J + rho I)^-1 ..., residual.c:570 notes).
+    (mat_invert(J + rho I), residual.c:125-196). coh_fn(freq) overrides
+    the per-channel coherency predict — the beamed-residual hook
+    (calculate_residuals_multifreq_withbeam, Dirac_radio.h:495-497).
     Returns xres [F, B, 2, 2]."""
     T, Nbase = tile.tilesz, tile.Nbase
     B = tile.x.shape[0]
@@ -454,8 +456,12 @@ def calculate_residuals_multifreq(state, pack, tile, bb, ccid=None, rho=0.0,
     ids = getattr(pack, 'cluster_ids', list(range(state.M)))
     skip = {i for i, c in enumerate(ids) if c < 0}
     for fi, f in enumerate(tile.freqs):
-        cohs = ops.predict_coh(pack, tile.u, tile.v, tile.w, float(f),
-                               tile.freq0, fdelta_ch, tile.tdelta, tile.dec0)
+        if coh_fn is not None:
+            cohs = coh_fn(float(f)).to(tile.xo.dtype)
+        else:
+            cohs = ops.predict_coh(pack, tile.u, tile.v, tile.w, float(f),
+                                   tile.freq0, fdelta_ch, tile.tdelta,
+                                   tile.dec0)
         V = total_model(state, cohs, bb, T, Nbase, skip=skip)
         out[fi] = tile.xo[fi] - V
     if ccid is not None:
