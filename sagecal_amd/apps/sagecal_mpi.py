@@ -67,6 +67,12 @@ def build_argparser():
     ap.add_argument('-T', dest='nend', type=int, default=0,
                     help='>0: stop after this solution interval')
     ap.add_argument('-W', dest='whiten', type=int, default=0)
+    ap.add_argument('-B', dest='dobeam', type=int, default=0,
+                    help='beam in predict (as in sagecal: 1 array, '
+                         '2 array+element, 3 element; the reference '
+                         'sagecal-mpi clamps >3 to 1, MPI/main.cpp:147)')
+    ap.add_argument('--elem-type', dest='elem_type', default='auto',
+                    choices=['auto', 'synthetic', 'lba', 'hba', 'alo'])
     ap.add_argument('-M', dest='mdl', action='store_true',
                     help='evaluate AIC/MDL over polynomial orders 1..-P '
                          'after the first tile and print the suggestion '
@@ -247,7 +253,13 @@ def main(argv=None):
         uvlen = torch.sqrt(tile.u ** 2 + tile.v ** 2) * tile.freq0
         flags = tile.flags | (uvlen < args.min_uvcut) | \
             (uvlen > args.max_uvcut)
-        cohs = sage.precalc_coherencies(pack, tile)
+        if getattr(args, 'dobeam', 0):
+            from .sagecal import _predict_with_beam
+            if args.dobeam > 3:       # MPI/main.cpp:147 clamp
+                args.dobeam = 1
+            cohs = _predict_with_beam(ms, pack, tile, ti, args)
+        else:
+            cohs = sage.precalc_coherencies(pack, tile)
         if cohs.dtype != cdtype:
             cohs = cohs.to(cdtype)
         # TAG_FRATIO: weigh this band's rho by its unflagged fraction
@@ -258,9 +270,17 @@ def main(argv=None):
         if args.use_global:
             state.J = adm.global_solution()
         ccid = args.ccid if args.ccid != -99999 else None
+        coh_fn = None
+        if getattr(args, 'dobeam', 0):
+            from .sagecal import _predict_channel_with_beam
+            fdch = tile.fdelta / len(tile.freqs)
+            coh_fn = (lambda f, _t=tile, _ti=ti:
+                      _predict_channel_with_beam(ms, pack, _t, _ti,
+                                                 args, f, fdch))
         xres = sage.calculate_residuals_multifreq(state, pack, tile, bb,
                                                   ccid=ccid,
-                                                  rho=args.rho_corr)
+                                                  rho=args.rho_corr,
+                                                  coh_fn=coh_fn)
         ms.write_column(args.outcol, ti, xres)
         if writer:
             writer.write_tile(state)
