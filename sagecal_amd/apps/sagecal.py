@@ -262,9 +262,13 @@ def run_calibration(args):
             from ..ops import dispatch as dops
             fdelta_ch = tile.fdelta / len(tile.freqs)
             for fi, f in enumerate(tile.freqs):
-                cohs_f = dops.predict_coh(pack, tile.u, tile.v, tile.w,
-                                          float(f), tile.freq0, fdelta_ch,
-                                          tile.tdelta, tile.dec0)
+                if coh_fn is not None:   # beamed per-channel (-B + -b)
+                    cohs_f = coh_fn(float(f))
+                else:
+                    cohs_f = dops.predict_coh(pack, tile.u, tile.v,
+                                              tile.w, float(f),
+                                              tile.freq0, fdelta_ch,
+                                              tile.tdelta, tile.dec0)
                 if cohs_f.dtype != state.J.dtype:
                     cohs_f = cohs_f.to(state.J.dtype)
 
