@@ -360,6 +360,11 @@ def run_simulation(args):
 def run_stochastic(args):
     """-N epochs: minibatch (bandpass-consensus when -w > 1) calibration
     (minibatch_mode.cpp / minibatch_consensus_mode.cpp)."""
+    if getattr(args, 'dobeam', 0):
+        raise SystemExit(
+            'sagecal: -B with stochastic mode (-N) is not supported yet; '
+            'failing loudly instead of silently calibrating beam-free '
+            '(ROUND3_NOTES)')
     from ..solvers.stochastic import MinibatchConsensusCalibration
     ms, pack, clusters, device, dtype = load_context(args)
     cdtype = torch.complex64 if dtype == torch.float32 else torch.complex128
