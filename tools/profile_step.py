@@ -22,7 +22,7 @@ def main():
     ap.add_argument('--dirs', type=int, default=10)
     ap.add_argument('--srcs', type=int, default=5)
     ap.add_argument('--tilesz', type=int, default=60)
-    ap.add_argument('--intervals', type=int, default=8)
+    ap.add_argument('--intervals', type=int, default=16)
     ap.add_argument('--chan', type=int, default=8)
     ap.add_argument('--emiter', type=int, default=2)
     ap.add_argument('--maxiter', type=int, default=8)
