@@ -73,6 +73,13 @@ def build_argparser():
                          'sagecal-mpi clamps >3 to 1, MPI/main.cpp:147)')
     ap.add_argument('--elem-type', dest='elem_type', default='auto',
                     choices=['auto', 'synthetic', 'lba', 'hba', 'alo'])
+    ap.add_argument('-D', dest='diffuse',
+                    help='cluster_id,gamma: treat this cluster as the '
+                         'diffuse foreground model when -X is on — its '
+                         'coherencies are re-predicted with the FISTA '
+                         'spatial model each cadence '
+                         '(recalculate_diffuse_coherencies, '
+                         'sagecal_slave.cpp:669-694; MPI/main.cpp:76)')
     ap.add_argument('-M', dest='mdl', action='store_true',
                     help='evaluate AIC/MDL over polynomial orders 1..-P '
                          'after the first tile and print the suggestion '
@@ -147,6 +154,43 @@ def _run_multiplexed(args, names, rank, world, device, dtype, cdtype):
     if world > 1:
         dist.destroy_process_group()
     return 0
+
+
+def _diffuse_hook(args, pack, tile, bb):
+    """Build the diffuse-model recalculation hook for -D cluster_id
+    (reference sp_diffuse_id): replaces the designated cluster's
+    coherencies with the spatial-model shapelet predict each cadence."""
+    spec = getattr(args, 'diffuse', None)
+    if not spec or not getattr(args, 'spatial', None):
+        return None
+    ddid = int(str(spec).split(',')[0])
+    import torch as _t
+    ids = getattr(pack, 'cluster_ids', [])
+    match = [i for i, c in enumerate(ids) if c == ddid]
+    if not match:
+        print(f"sagecal-mpi: -D cluster id {ddid} not found; ignored")
+        return None
+    ci = match[0]
+    s0, s1 = int(pack.cluster_off[ci]), int(pack.cluster_off[ci + 1])
+    gi = next((g for g in range(s0, s1)
+               if g in getattr(pack, 'shapelets', {})), None)
+    if gi is None:
+        print(f"sagecal-mpi: -D cluster {ddid} has no shapelet source; "
+              f"ignored")
+        return None
+    n0, beta_c, modes = pack.shapelets[gi]
+    Cm = _t.zeros(n0 * n0, 2, 2, dtype=_t.complex128)
+    Cm[:, 0, 0] = _t.as_tensor(np.asarray(modes), dtype=_t.float64)
+    Cm[:, 1, 1] = Cm[:, 0, 0]
+    lmn = (float(pack.ll[gi]), float(pack.mm[gi]), float(pack.nn1[gi]))
+
+    def hook(adm, cohs):
+        out = adm.diffuse_coherencies(tile.u, tile.v, tile.w, bb, Cm,
+                                      beta_c, lmn, tile.freq0,
+                                      tile.fdelta)
+        if out is not None:
+            cohs[ci] = out.to(device=cohs.device, dtype=cohs.dtype)
+    return hook
 
 
 def main(argv=None):
@@ -264,9 +308,11 @@ def main(argv=None):
             cohs = cohs.to(cdtype)
         # TAG_FRATIO: weigh this band's rho by its unflagged fraction
         adm.set_fratio(float((~flags).float().mean()))
+        dhook = _diffuse_hook(args, pack, tile, bb)
         res0, res1 = adm.run(cohs, tile, bb, opts, n_admm=args.nadmm,
                              flags=flags,
-                             verbose=args.verbose and rank == 0)
+                             verbose=args.verbose and rank == 0,
+                             diffuse_hook=dhook)
         if args.use_global:
             state.J = adm.global_solution()
         ccid = args.ccid if args.ccid != -99999 else None

@@ -219,9 +219,14 @@ class ConsensusADMM:
         self.J_prev = self.state.J.clone()
 
     def run(self, cohs, tile, bb, opts, n_admm=10, flags=None,
-            verbose=False):
+            verbose=False, diffuse_hook=None):
         """The per-tile ADMM loop (sagecal_master.cpp:731-1060 semantics).
-        Returns (res0, res1) of the final local solve."""
+        Returns (res0, res1) of the final local solve.
+
+        diffuse_hook(adm, cohs): called after each spatial-model update
+        so the designated diffuse cluster's coherencies are re-predicted
+        with the spatial model applied (recalculate_diffuse_coherencies
+        at the admm cadence, sagecal_slave.cpp:669-694)."""
         res0 = res1 = None
         st = self.state
         for it in range(n_admm):
@@ -252,6 +257,8 @@ class ConsensusADMM:
                 lam, mu_l1, order, fiters, cadence = self.spatial
                 if (it + 1) % max(cadence, 1) == 0:
                     self.spatial_update(lam, mu_l1, fiters)
+                    if diffuse_hook is not None:
+                        diffuse_hook(self, cohs)
             if verbose:
                 # primal ||J - BZ|| and dual ||Z - Zold|| residual norms
                 # (sagecal_master.cpp:881-885 / sagecal_slave.cpp:911-919)

@@ -499,3 +499,51 @@ def test_cli_beam_with_real_lofar_tables(obs):
     # right tables: near-noise residual; wrong tables: markedly worse
     assert res['lba'] < 0.05 * data_mean, res
     assert res['hba'] > 3.0 * res['lba'], res
+
+
+def test_sagecal_mpi_diffuse_spatial_model(tmp_path, monkeypatch):
+    """-D cluster_id,gamma with -X: the designated shapelet cluster's
+    coherencies are re-predicted from the FISTA spatial model at the
+    admm cadence (recalculate_diffuse_coherencies hook,
+    sagecal_slave.cpp:669-694) — world=1 run, residual still reduced,
+    hook actually fired."""
+    from sagecal_amd.apps import sagecal_mpi as app
+    from sagecal_amd import shapelet as shmod
+    skyf = tmp_path / 'sky.txt'
+    skyf.write_text(SKY + "SDIF 0 1 0 45 5 0 2.0 0 0 0 0 0 "
+                          "1e-3 1e-3 0 150e6\n")
+    clf = tmp_path / 'cluster.txt'
+    clf.write_text(CLUSTER + "3 1 SDIF\n")
+    # shapelet modes file next to the sky file (readsky.c convention)
+    shmod.write_modes_file(str(tmp_path / 'SDIF.fits.modes'),
+                           0.25, 0.79, 2, 5e-4,
+                           np.array([1.0, 0.2, 0.1, 0.05]))
+    clusters = sky.read_sky_cluster(str(skyf), str(clf), 0.0, np.pi / 4,
+                                    150e6)
+    pack = SourcePack(clusters)
+    assert pack.shapelets, "shapelet source did not load"
+    msf = tmp_path / 'band0.npz'
+    msdata.make_synthetic_npz(str(msf), N=8, tilesz=4, Ntime=4, Nchan=2,
+                              pack=pack, bandwidth=50e3,
+                              noise_sigma=1e-3, seed=9, ra0=0.0,
+                              dec0=np.pi / 4)
+    (tmp_path / 'mslist.txt').write_text(str(msf))
+    fired = {'n': 0}
+    from sagecal_amd.consensus.admm import ConsensusADMM
+    orig = ConsensusADMM.spatial_update
+
+    def spy(self, *a, **k):
+        fired['n'] += 1
+        return orig(self, *a, **k)
+    monkeypatch.setattr(ConsensusADMM, 'spatial_update', spy)
+    rc = app.main(['-f', str(tmp_path / 'mslist.txt'), '-s', str(skyf),
+                   '-c', str(clf), '-t', '4', '-A', '4', '-P', '1',
+                   '-j', '3', '-e', '2', '-g', '8', '-r', '2.0',
+                   '-X', '0.01,1e-4,1,10,2', '-u', '0.0', '-D', '3,0.1',
+                   '-F', '0'])
+    assert rc == 0
+    assert fired['n'] >= 1
+    z = np.load(str(msf))
+    assert 'residual' in z.files
+    assert np.isfinite(z['residual']).all()
+    assert np.abs(z['residual']).mean() < 0.6 * np.abs(z['data']).mean()
