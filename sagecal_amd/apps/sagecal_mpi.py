@@ -46,8 +46,9 @@ def build_argparser():
     ap.add_argument('-r', dest='admm_rho', type=float, default=5.0)
     ap.add_argument('-G', dest='rhofile',
                     help='per-cluster regularization file')
-    ap.add_argument('-a', dest='use_bb', type=int, default=0,
-                    help='1: Barzilai-Borwein adaptive rho')
+    ap.add_argument('-a', '-C', dest='use_bb', type=int, default=0,
+                    help='>0: Barzilai-Borwein adaptive rho (the '
+                         'reference flag is -C, MPI/main.cpp)')
     ap.add_argument('-U', dest='use_global', type=int, default=0,
                     help='1: residuals from the global solution B Z')
     ap.add_argument('-X', dest='spatial',
@@ -84,6 +85,29 @@ def build_argparser():
                     help='evaluate AIC/MDL over polynomial orders 1..-P '
                          'after the first tile and print the suggestion '
                          '(reference -M, sagecal_master.cpp:992)')
+    ap.add_argument('-I', dest='incol', default=None,
+                    help='input data column (reference -I)')
+    ap.add_argument('-J', dest='phase_only', type=int, default=0,
+                    help='1: phase-only correction (with -k)')
+    ap.add_argument('-q', dest='initsol', default=None,
+                    help='warm-start solutions file (per rank)')
+    ap.add_argument('-R', dest='randomize', type=int, default=0,
+                    help='randomized/weighted EM iteration allocation')
+    ap.add_argument('-N', dest='epochs', type=int, default=0,
+                    help='>0: FEDERATED stochastic calibration — every '
+                         'rank runs minibatch bandpass consensus and the '
+                         'per-rank Z are manifold-averaged across ranks '
+                         '(sagecal_stochastic_master/slave role)')
+    ap.add_argument('-M2', dest='minibatches', type=int, default=2,
+                    help='time minibatches per epoch (stochastic mode; '
+                         'the reference letter -M is taken by our MDL '
+                         'advisory flag)')
+    ap.add_argument('-w', dest='minibands', type=int, default=1,
+                    help='mini-bands for stochastic bandpass mode')
+    ap.add_argument('-S', dest='fed_alpha', type=float, default=0.1,
+                    help='federated averaging strength alpha '
+                         '(find_prod_inverse_full_fed, '
+                         'sagecal_stochastic_slave.cpp:563)')
     ap.add_argument('-O', dest='outcol', default='residual')
     ap.add_argument('-V', dest='verbose', action='store_true')
     return ap
@@ -152,6 +176,44 @@ def _run_multiplexed(args, names, rank, world, device, dtype, cdtype):
     for ms in mss:
         ms.save()
     if world > 1:
+        dist.destroy_process_group()
+    return 0
+
+
+def _run_federated_stochastic(args, ms, pack, device, cdtype, rank,
+                              world):
+    """-N epochs in sagecal-mpi: each rank runs stochastic bandpass
+    consensus on its own band; per-rank Z is pulled toward the
+    manifold-averaged global Z with strength -S alpha
+    (sagecal_stochastic_slave.cpp:563 find_prod_inverse_full_fed +
+    :799-878 X update; master averaging
+    sagecal_stochastic_master.cpp:340-350)."""
+    from ..solvers.stochastic import MinibatchConsensusCalibration
+    cal = MinibatchConsensusCalibration(
+        pack, ms.N, ms.freqs, nsolbw=args.minibands, Npoly=args.npoly,
+        poly_type=args.polytype, rho=args.admm_rho, device=device,
+        dtype=cdtype, fed_alpha=args.fed_alpha if world > 1 else 0.0,
+        world=world, rank=rank)
+    bb = ms.bb_tensor(device=device)
+    for ti, tile in enumerate(ms.tiles()):
+        if ti < args.nskip:
+            continue
+        if args.nend > 0 and ti >= args.nend:
+            break
+        for ep in range(args.epochs):
+            for _ in range(max(1, args.nadmm)):
+                cal.epoch(tile, bb, nmb=args.minibatches,
+                          lbfgs_iters=max(args.max_lbfgs, 8),
+                          robust_nu=(args.nulow + args.nuhigh) / 2)
+        xres = cal.residuals(tile, bb,
+                             use_global=bool(args.use_global))
+        ms.write_column(args.outcol, ti, xres)
+        if rank == 0 or args.verbose:
+            print(f"rank {rank} tile {ti}: federated stochastic "
+                  f"epochs {args.epochs}, "
+                  f"res {float(xres.abs().pow(2).mean()):.6f}")
+    ms.save()
+    if world > 1 and dist.is_initialized():
         dist.destroy_process_group()
     return 0
 
@@ -229,13 +291,28 @@ def main(argv=None):
         return _run_multiplexed(args, names, rank, world, device, dtype,
                                 cdtype)
     my_ms = names[rank % len(names)]
+    mskw = {}
+    if getattr(args, 'incol', None):
+        mskw['data_col'] = args.incol
     ms = msdata.open_ms(my_ms, tilesz=args.tilesz, device=device,
-                        dtype=dtype)
+                        dtype=dtype, **mskw)
     clusters = skymod.read_sky_cluster(args.sky, args.cluster, ms.ra0,
                                        ms.dec0, ms.freq0, fmt=args.format,
                                        jd=getattr(ms, 'jd0', None))
     pack = SourcePack(clusters)
     state = sage.CalState(pack, ms.N, device=device, dtype=cdtype)
+    if getattr(args, 'initsol', None):
+        hdr_i, tiles_i = solutions.read_solutions(args.initsol)
+        if tiles_i:
+            state.J = solutions.reorder_read_tile(
+                tiles_i[0], state.nchunks).to(device=device, dtype=cdtype)
+    if getattr(args, 'epochs', 0) > 0:
+        # federated stochastic mode (-N>0): per-rank minibatch bandpass
+        # consensus + manifold-averaged global Z across ranks (the
+        # reference's sagecal_stochastic_master/slave dispatch,
+        # MPI/main.cpp:336-370)
+        return _run_federated_stochastic(args, ms, pack, device, cdtype,
+                                         rank, world)
 
     rho = torch.full((pack.M,), args.admm_rho)
     if args.rhofile:
@@ -269,7 +346,8 @@ def main(argv=None):
         max_emiter=args.max_emiter, max_iter=args.max_iter,
         solver_mode=args.solver_mode, robust_nulow=args.nulow,
         robust_nuhigh=args.nuhigh,
-        lbfgs_iters=max(args.max_lbfgs, 0))
+        lbfgs_iters=max(args.max_lbfgs, 0),
+        randomize=bool(getattr(args, 'randomize', 0)))
     opts.lbfgs_m = args.lbfgs_m
     writer = None
     zwriter = None
@@ -316,6 +394,15 @@ def main(argv=None):
         if args.use_global:
             state.J = adm.global_solution()
         ccid = args.ccid if args.ccid != -99999 else None
+        if getattr(args, 'phase_only', 0) and ccid is not None:
+            # phase-only correction (-J, residual.c phase-only path)
+            ids = getattr(pack, 'cluster_ids', list(range(state.M)))
+            match = [i for i, c in enumerate(ids) if c == ccid]
+            if match:
+                o = state.chunk_off[match[0]]
+                nc = state.nchunks[match[0]]
+                Jc = state.J[o:o + nc]
+                state.J[o:o + nc] = Jc / Jc.abs().clamp_min(1e-12)
         coh_fn = None
         if getattr(args, 'dobeam', 0):
             from .sagecal import _predict_channel_with_beam

@@ -547,3 +547,51 @@ def test_sagecal_mpi_diffuse_spatial_model(tmp_path, monkeypatch):
     assert 'residual' in z.files
     assert np.isfinite(z['residual']).all()
     assert np.abs(z['residual']).mean() < 0.6 * np.abs(z['data']).mean()
+
+
+def _fed_worker(rank, world, tmpdir):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = '29561'
+    os.environ['RANK'] = str(rank)
+    os.environ['WORLD_SIZE'] = str(world)
+    import torch.distributed as dist
+    dist.init_process_group('gloo', rank=rank, world_size=world)
+    try:
+        from sagecal_amd.apps import sagecal_mpi as app
+        rc = app.main(['-f', os.path.join(tmpdir, 'mslist.txt'),
+                       '-s', os.path.join(tmpdir, 'sky.txt'),
+                       '-c', os.path.join(tmpdir, 'cluster.txt'),
+                       '-t', '4', '-N', '2', '-M2', '2', '-w', '2',
+                       '-l', '8', '-A', '2', '-P', '2', '-S', '0.1'])
+        assert rc == 0
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_sagecal_mpi_federated_stochastic(tmp_path):
+    """-N in sagecal-mpi: the federated stochastic mode (the reference's
+    sagecal_stochastic_master/slave dispatch) runs 2 ranks over gloo,
+    writes reduced residuals on both bands."""
+    import torch.multiprocessing as mp
+    (tmp_path / 'sky.txt').write_text(SKY)
+    (tmp_path / 'cluster.txt').write_text(CLUSTER)
+    names = []
+    for r, f0 in enumerate((150e6, 152e6)):
+        clusters = sky.read_sky_cluster(str(tmp_path / 'sky.txt'),
+                                        str(tmp_path / 'cluster.txt'),
+                                        0.0, np.pi / 4, f0)
+        pack = SourcePack(clusters)
+        msf = tmp_path / f'fband{r}.npz'
+        msdata.make_synthetic_npz(str(msf), N=8, tilesz=4, Ntime=4,
+                                  Nchan=4, pack=pack, freq0=f0,
+                                  bandwidth=50e3, noise_sigma=1e-3,
+                                  seed=11, ra0=0.0, dec0=np.pi / 4)
+        names.append(str(msf))
+    (tmp_path / 'mslist.txt').write_text('\n'.join(names))
+    mp.spawn(_fed_worker, args=(2, str(tmp_path)), nprocs=2, join=True)
+    for r in range(2):
+        z = np.load(names[r])
+        assert 'residual' in z.files
+        assert np.abs(z['residual']).mean() < \
+            0.6 * np.abs(z['data']).mean(), f"band {r}"
