@@ -336,6 +336,119 @@ k_cholmw_panel(int n, int k, float* __restrict__ Lbuf,
   float* LT = L + (size_t)n * n;
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* pan = smem;
+  __shared__ float yv[NB];
+  __shared__ float rdg[NB];   // reciprocal diagonal (div -> mul downstream)
+  __shared__ int bad;
+  if (tid == 0) bad = 0;
+  const int rows = n - k;
+  // stage panel rows k..n, cols k..k+NB from the (already-updated) L copy
+  for (int idx = tid; idx < rows * (NB / 4); idx += NTH) {
+    const int r = idx >> 3, c4 = (idx & 7) << 2;
+    *(float4*)(pan + r * PST + c4) =
+        *(const float4*)(L + (size_t)(k + r) * n + k + c4);
+  }
+  __syncthreads();
+  // wave-synchronous 32x32 factor (same scheme as k_chol_solve)
+  if (tid < 64 && lane < 32) {
+    const int r = lane;
+    float row[NB];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) row[c] = pan[r * PST + c];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      float pivraw = __shfl(row[c], c, 64);
+      if (pivraw <= 1e-30f) {
+        if (lane == 0) bad = 1;
+        pivraw = 1e-30f;
+      }
+      const float pv = sqrtf(pivraw);
+      const float rpv = 1.0f / pv;
+      if (r >= c) row[c] *= rpv;
+      yv[r] = row[c];
+      if (lane == c) rdg[c] = rpv;
+      const float lrc = row[c];
+#pragma unroll
+      for (int cc = c + 1; cc < NB; ++cc) {
+        if (r >= cc) row[cc] -= lrc * yv[cc];
+      }
+    }
+#pragma unroll
+    for (int c = 0; c < NB; ++c) pan[r * PST + c] = row[c];
+    // publish 1/pv on the panel diagonal: the LT mirror's diagonal is read
+    // ONLY by the backward substitution's divide (syrk/fwd never touch it),
+    // so storing the reciprocal there turns that divide into a multiply
+    pan[r * PST + r] = rdg[r];
+  }
+  __syncthreads();
+  // row-solve sub-panel rows NB..rows (multiply by reciprocal diag: the
+  // 32 serial divides per row-thread were the panel's longest chain)
+  for (int r = NB + tid; r < rows; r += NTH) {
+    float rw[NB];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) rw[c] = pan[r * PST + c];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      float s = rw[c];
+#pragma unroll
+      for (int c2 = 0; c2 < c; ++c2) s -= rw[c2] * pan[c * PST + c2];
+      rw[c] = s * rdg[c];
+    }
+#pragma unroll
+    for (int c = 0; c < NB; ++c) pan[r * PST + c] = rw[c];
+  }
+  __syncthreads();
+  // write back the LT mirror ONLY: every consumer of the factored panel
+  // (syrk fragments, fused forward subst via LDS, backward subst) reads
+  // L^T; the L working copy keeps unfactored trailing values, which is
+  // all later panel stages ever read.
+#pragma unroll
+  for (int c = 0; c < NB; ++c) {
+    for (int r = tid; r < rows; r += NTH) {
+      LT[(size_t)(k + c) * n + k + r] = pan[r * PST + c];
+    }
+  }
+  __syncthreads();
+  // fused forward-substitution step: y_k = L_kk^-1 xo[k..k+NB] on wave 0,
+  // then xo[k+NB..n] -= panel . y_k straight from LDS (the separate subst
+  // kernel then only runs the backward pass)
+  float* xo = dp + (size_t)bid * n;
+  if (tid < 64) {
+    const int r = lane & 31;
+    float rv[NB];
+    float bv = (lane < 32) ? xo[k + r] : 0.f;
+    if (lane < 32) {
+#pragma unroll
+      for (int c = 0; c < NB; ++c)
+        rv[c] = (c <= r) ? pan[r * PST + c] : 0.f;
+    }
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      float yc;
+      if (lane == c) bv *= rdg[c];
+      yc = __shfl(bv, c, 64);
+      if (lane < 32 && r > c) bv -= rv[c] * yc;
+      if (lane == c) yv[c] = bv;
+    }
+    if (lane < 32) xo[k + r] = bv;
+  }
+  __syncthreads();
+  for (int r = NB + tid; r < rows; r += NTH) {
+    float s = 0.f;
+#pragma unroll
+    for (int c = 0; c < NB; ++c) s += pan[r * PST + c] * yv[c];
+    xo[k + r] -= s;
+  }
+  if (bad && tid == 0) atomicOr(&info[bid], 1);
+}
+
+extern "C" __global__ void __launch_bounds__(NTH)
+k_cholmw_panel_big(int n, int k, float* __restrict__ Lbuf,
+               float* __restrict__ dp, int* __restrict__ info) {
+  const int bid = blockIdx.x, tid = threadIdx.x, lane = tid & 63;
+  float* L = Lbuf + (size_t)bid * 2 * n * n;
+  float* LT = L + (size_t)n * n;
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* pan = smem;
   // factored 32x32 diag block kept resident across row chunks so panels
   // taller than the LDS budget (n > PANEL_CAP, up to MAXN_CHOL_MW) are
   // processed in passes: the row-solve/LT-writeback/fwd-subst of a row

@@ -85,11 +85,18 @@ hipError_t launch_chol_mw(
   hipLaunchKernelGGL(k_cholmw_init, dim3(batch, INIT_SLICES), dim3(512), 0,
       stream, JtJ, Jtr, mu, n, Lbuf, dp);
   if (stages < 2) return hipGetLastError();
+  // n <= PANEL_CAP: the hardware-validated single-pass panel kernel;
+  // larger n: the chunked-panel variant (row passes through LDS)
   const int cap = n < PANEL_CAP ? n : PANEL_CAP;
   const size_t shmem = (size_t)(cap + 8) * PST * sizeof(float);
   for (int k = 0; k < n; k += NB) {
-    hipLaunchKernelGGL(k_cholmw_panel, dim3(batch), dim3(512), shmem,
-        stream, n, k, Lbuf, dp, info);
+    if (n <= PANEL_CAP) {
+      hipLaunchKernelGGL(k_cholmw_panel, dim3(batch), dim3(512), shmem,
+          stream, n, k, Lbuf, dp, info);
+    } else {
+      hipLaunchKernelGGL(k_cholmw_panel_big, dim3(batch), dim3(512),
+          shmem, stream, n, k, Lbuf, dp, info);
+    }
     const int tcnt = (n - k - NB) / NB;
     const int ntiles = tcnt * (tcnt + 1) / 2;
     if (ntiles > 0 && stages >= 3) {
