@@ -108,6 +108,9 @@ def build_argparser():
                     help='federated averaging strength alpha '
                          '(find_prod_inverse_full_fed, '
                          'sagecal_stochastic_slave.cpp:563)')
+    ap.add_argument('-i', dest='dodiag', type=int, default=0,
+                    help='1: replace output with influence diagnostics '
+                         '(reference MPI -i)')
     ap.add_argument('-O', dest='outcol', default='residual')
     ap.add_argument('-V', dest='verbose', action='store_true')
     return ap
@@ -414,6 +417,12 @@ def main(argv=None):
                                                   ccid=ccid,
                                                   rho=args.rho_corr,
                                                   coh_fn=coh_fn)
+        if getattr(args, 'dodiag', 0):
+            from ..solvers import diagnostics as diagmod
+            cohs0 = cohs if cohs.dtype == state.J.dtype \
+                else cohs.to(state.J.dtype)
+            lev = diagmod.influence_map(state, cohs0, tile, bb)
+            xres = lev[None, :, None, None].expand_as(xres).to(xres.dtype)
         ms.write_column(args.outcol, ti, xres)
         if writer:
             writer.write_tile(state)

@@ -540,13 +540,14 @@ def test_sagecal_mpi_diffuse_spatial_model(tmp_path, monkeypatch):
                    '-c', str(clf), '-t', '4', '-A', '4', '-P', '1',
                    '-j', '3', '-e', '2', '-g', '8', '-r', '2.0',
                    '-X', '0.01,1e-4,1,10,2', '-u', '0.0', '-D', '3,0.1',
-                   '-F', '0'])
+                   '-F', '0', '-i', '1'])
     assert rc == 0
     assert fired['n'] >= 1
     z = np.load(str(msf))
+    # -i 1: the output column carries influence-function leverage
     assert 'residual' in z.files
     assert np.isfinite(z['residual']).all()
-    assert np.abs(z['residual']).mean() < 0.6 * np.abs(z['data']).mean()
+    assert z['residual'].real.max() > 0
 
 
 def _fed_worker(rank, world, tmpdir):
