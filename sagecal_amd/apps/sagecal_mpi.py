@@ -104,10 +104,14 @@ def build_argparser():
                          'advisory flag)')
     ap.add_argument('-w', dest='minibands', type=int, default=1,
                     help='mini-bands for stochastic bandpass mode')
-    ap.add_argument('-S', dest='fed_alpha', type=float, default=0.1,
+    ap.add_argument('--fed-alpha', dest='fed_alpha', type=float,
+                    default=0.1,
                     help='federated averaging strength alpha '
                          '(find_prod_inverse_full_fed, '
                          'sagecal_stochastic_slave.cpp:563)')
+    ap.add_argument('-S', dest='heapsize', type=float, default=0,
+                    help='accepted for reference-CLI compatibility '
+                         '(GPU heap MB — not applicable here)')
     ap.add_argument('-i', dest='dodiag', type=int, default=0,
                     help='1: replace output with influence diagnostics '
                          '(reference MPI -i)')

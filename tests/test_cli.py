@@ -563,7 +563,7 @@ def _fed_worker(rank, world, tmpdir):
                        '-s', os.path.join(tmpdir, 'sky.txt'),
                        '-c', os.path.join(tmpdir, 'cluster.txt'),
                        '-t', '4', '-N', '2', '-M2', '2', '-w', '2',
-                       '-l', '8', '-A', '2', '-P', '2', '-S', '0.1'])
+                       '-l', '8', '-A', '2', '-P', '2', '--fed-alpha', '0.1'])
         assert rc == 0
     finally:
         if dist.is_initialized():
