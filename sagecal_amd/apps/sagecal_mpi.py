@@ -109,6 +109,10 @@ def build_argparser():
                     help='federated averaging strength alpha '
                          '(find_prod_inverse_full_fed, '
                          'sagecal_stochastic_slave.cpp:563)')
+    ap.add_argument('-E', dest='gpupredict', type=int, default=1,
+                    help='accepted for reference-CLI compatibility '
+                         '(model prediction runs on the GPU whenever '
+                         'one is present)')
     ap.add_argument('-S', dest='heapsize', type=float, default=0,
                     help='accepted for reference-CLI compatibility '
                          '(GPU heap MB — not applicable here)')
