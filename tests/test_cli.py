@@ -116,6 +116,16 @@ def _mpi_worker(rank, world, tmpdir):
                        '-X', '0.01,1e-4,1,10,2', '-u', '0.0',
                        '-p', os.path.join(tmpdir, 'sol.txt')])
         assert rc == 0
+        # warm start (-q) from the solutions just written (new round-2
+        # MPI path): one quick ADMM pass must run clean
+        rc = app.main(['-f', os.path.join(tmpdir, 'mslist.txt'),
+                       '-s', os.path.join(tmpdir, 'sky.txt'),
+                       '-c', os.path.join(tmpdir, 'cluster.txt'),
+                       '-t', '4', '-A', '2', '-P', '2', '-j', '3',
+                       '-e', '1', '-g', '6', '-r', '2.0',
+                       '-q', os.path.join(tmpdir, 'sol.txt.rank')
+                       + str(rank), '-O', 'res_q'])
+        assert rc == 0
     finally:
         if dist.is_initialized():
             dist.destroy_process_group()
@@ -148,6 +158,9 @@ def test_sagecal_mpi_two_bands(tmp_path):
         assert 'residual' in z.files
         assert np.abs(z['residual']).mean() < \
             0.4 * np.abs(z['data']).mean(), f"band {r}"
+        # warm-started run also reduced residuals
+        assert np.abs(z['res_q']).mean() < \
+            0.4 * np.abs(z['data']).mean(), f"band {r} warm"
     from sagecal_amd import solutions
     hdr, tiles = solutions.read_solutions(str(tmp_path / 'sol.txt.rank0'))
     assert tiles and tiles[0].shape[-2:] == (2, 2)
