@@ -99,9 +99,10 @@ def _lm_solve_eager(prob, J0, maxiter=30, tau=1e-3, eps1=1e-9, eps2=1e-9,
     diag_max = JtJ.diagonal(dim1=-2, dim2=-1).max(dim=-1).values
     mu = tau * diag_max
     niter = 0
-    # mw Cholesky handles up to n=4096 since round 2 (chunked panels);
-    # >1024 is gated off until GPU-validated (SAGECAL_CHOL_BIG=1)
-    chol_cap = 4096 if os.environ.get('SAGECAL_CHOL_BIG') == '1' else 1024
+    # mw Cholesky handles up to n=4096 (chunked panels,
+    # k_cholmw_panel_big) — hardware-validated end of round 2
+    # (profiles/standalone_validate_r2.txt); SAGECAL_CHOL_BIG=0 opts out
+    chol_cap = 1024 if os.environ.get('SAGECAL_CHOL_BIG') == '0' else 4096
     use_hip_chol = (prob.layout is not None and x.is_cuda
                     and 8 * N <= chol_cap)
     for it in range(maxiter):

@@ -357,10 +357,8 @@ def test_fused_beam_predict_matches_torch():
     assert err < 2e-4, err
 
 
-@pytest.mark.skipif(os.environ.get('SAGECAL_CHOL_BIG') != '1',
-                    reason='large-n chunked-panel Cholesky written '
-                           'round 2, GPU-validated next round '
-                           '(SAGECAL_CHOL_BIG=1)')
+@pytest.mark.skipif(os.environ.get('SAGECAL_CHOL_BIG') == '0',
+                    reason='opted out via SAGECAL_CHOL_BIG=0')
 def test_chol_mw_large_n4096():
     """Chunked-panel mw Cholesky at the 512-station LM shape
     (8N=4096, clmfit_cuda.c:1624-1674 role) vs torch.cholesky_solve."""
